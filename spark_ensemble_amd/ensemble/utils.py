@@ -1,0 +1,64 @@
+"""Ensemble utilities.
+
+``weighted_median`` reproduces reference ensemble/Utils.scala:26-40 (sort by
+value, first index where cumulative weight >= half of total); ``subspace`` /
+``slice_features`` reproduce the sub-bagging primitives of reference
+HasSubBag.scala:73-84 (Bernoulli(ratio) filter over feature indices with a
+per-learner seed; sorted indices; dense gather).
+"""
+
+from __future__ import annotations
+
+import torch
+
+
+def weighted_median(values: torch.Tensor, weights: torch.Tensor) -> torch.Tensor:
+    """Row-wise weighted median.
+
+    values, weights: [N, M] (M models per row) or [M].  Returns [N] (or
+    scalar).  Rule (reference Utils.scala:26-40): sort by value, take the
+    first value whose cumulative weight >= 0.5 * total weight.
+    """
+    single = values.dim() == 1
+    if single:
+        values = values.unsqueeze(0)
+        weights = weights.unsqueeze(0).expand_as(values)
+    if weights.dim() == 1:
+        weights = weights.unsqueeze(0).expand_as(values)
+    order = values.argsort(dim=1)
+    v_sorted = values.gather(1, order)
+    w_sorted = weights.gather(1, order)
+    cum = w_sorted.cumsum(dim=1)
+    half = 0.5 * w_sorted.sum(dim=1, keepdim=True)
+    # first index with cumulative weight >= half
+    idx = (cum >= half).float().argmax(dim=1, keepdim=True)
+    out = v_sorted.gather(1, idx).squeeze(1)
+    return out[0] if single else out
+
+
+def subspace(ratio: float, num_features: int, seed: int) -> torch.Tensor:
+    """Bernoulli(ratio) feature-index filter, deterministic in ``seed``.
+
+    Mirrors reference HasSubBag.scala:73-79 (XORShiftRandom(seed) Bernoulli
+    over 0..numFeatures-1; indices come out sorted).  Guarantees at least one
+    feature.  Ratio 1 is the identity (tested property, reference
+    HasSubBagSuite.scala:60-105).
+    """
+    if ratio >= 1.0:
+        return torch.arange(num_features, dtype=torch.long)
+    g = torch.Generator().manual_seed(int(seed) & 0x7FFFFFFFFFFFFFFF)
+    mask = torch.rand(num_features, generator=g) < ratio
+    idx = mask.nonzero(as_tuple=True)[0]
+    if idx.numel() == 0:
+        idx = torch.randint(0, num_features, (1,), generator=g)
+    return idx.sort().values
+
+
+def slice_features(x: torch.Tensor, indices: torch.Tensor) -> torch.Tensor:
+    """Dense feature gather [N, F] -> [N, |indices|]
+    (reference HasSubBag.scala:81-84)."""
+    if indices.numel() == x.shape[1]:
+        # identity subspace — avoid the gather
+        if bool((indices == torch.arange(x.shape[1], device=indices.device)).all()):
+            return x
+    return x.index_select(1, indices.to(x.device))

@@ -1,0 +1,221 @@
+"""Decision-tree base learners (the default base learners of every
+meta-estimator, standing in for MLlib's DecisionTreeClassifier/Regressor
+that the reference's tests use, e.g. reference BaggingClassifierSuite
+.scala:48-78).
+
+Param names/defaults follow Spark ML DecisionTree: maxDepth=5, maxBins=32,
+minInstancesPerNode=1, minInfoGain=0.0, seed.
+
+Both classes run on the unified histogram trainer (tree_grower.py):
+  * regressor: D=1 targets  (variance-reduction splits, leaf = mean)
+  * classifier: D=K one-hot targets (identically gini splits, leaf = class
+    probability vector)
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Optional
+
+import torch
+
+from .. import persistence
+from ..estimator import (
+    ProbabilisticClassificationModel,
+    ProbabilisticClassifier,
+    RegressionModel,
+    Regressor,
+)
+from ..frame import TensorFrame
+from ..ops import dispatch as ops
+from ..params import Params, ParamValidators
+from ..parallel import get_comm
+from .tree_grower import GrowParams, ensure_binned, grow_tree
+
+
+class _TreeParams(Params):
+    def _declare_params(self):
+        super()._declare_params()
+        self.maxDepth = self._int_param(
+            "maxDepth", "maximum tree depth", ParamValidators.inRange(0, 64)
+        )
+        self.maxBins = self._int_param(
+            "maxBins", "number of histogram bins", ParamValidators.inRange(2, 256)
+        )
+        self.minInstancesPerNode = self._int_param(
+            "minInstancesPerNode",
+            "minimum rows per child",
+            ParamValidators.gtEq(1),
+        )
+        self.minInfoGain = self._float_param(
+            "minInfoGain", "minimum gain to split", ParamValidators.gtEq(0.0)
+        )
+        self.minWeightFractionPerNode = self._float_param(
+            "minWeightFractionPerNode",
+            "minimum fraction of total weight per child",
+            ParamValidators.inRange(0.0, 0.5, upper_inclusive=False),
+        )
+        self.regLambda = self._float_param(
+            "regLambda", "L2 regularization on leaf values", ParamValidators.gtEq(0.0)
+        )
+        self.seed = self._int_param("seed", "random seed")
+        self._setDefault(
+            maxDepth=5,
+            maxBins=32,
+            minInstancesPerNode=1,
+            minInfoGain=0.0,
+            minWeightFractionPerNode=0.0,
+            regLambda=1e-6,
+            seed=0,
+        )
+
+    def setMaxDepth(self, v):
+        return self.set("maxDepth", v)
+
+    def setMaxBins(self, v):
+        return self.set("maxBins", v)
+
+    def setMinInstancesPerNode(self, v):
+        return self.set("minInstancesPerNode", v)
+
+    def setMinInfoGain(self, v):
+        return self.set("minInfoGain", v)
+
+    def setSeed(self, v):
+        return self.set("seed", v)
+
+    def _grow_params(self, total_weight: float) -> GrowParams:
+        return GrowParams(
+            max_depth=self.getOrDefault("maxDepth"),
+            max_bins=self.getOrDefault("maxBins"),
+            min_instances_per_node=self.getOrDefault("minInstancesPerNode"),
+            min_info_gain=self.getOrDefault("minInfoGain"),
+            min_child_weight=self.getOrDefault("minWeightFractionPerNode")
+            * total_weight,
+            lam=self.getOrDefault("regLambda"),
+        )
+
+
+class _TreeModelMixin:
+    """Shared storage/persistence for fitted tree models."""
+
+    def _set_tree(self, tree: dict, num_features: int):
+        self._tree = tree
+        self._num_features = num_features
+
+    def _predict_values(self, x: torch.Tensor) -> torch.Tensor:
+        t = self._tree
+        return ops.tree_predict(
+            x,
+            t["feature"],
+            t["threshold"],
+            t["left_child"],
+            t["leaf_value"],
+            max_depth=64,
+        )
+
+    @property
+    def depth(self) -> int:
+        # derived: longest root-to-leaf path
+        feature = self._tree["feature"].cpu()
+        left = self._tree["left_child"].cpu()
+
+        def d(i):
+            if feature[i] < 0:
+                return 0
+            li = int(left[i])
+            return 1 + max(d(li), d(li + 1))
+
+        return d(0)
+
+    def _save_impl(self, path: str):
+        persistence.save_metadata(
+            self, path, extra={"numFeatures": self._num_features, **self._extra_meta()}
+        )
+        persistence.save_tensors(os.path.join(path, "data"), self._tree)
+
+    def _extra_meta(self):
+        return {}
+
+    def _load_extra(self, path: str, meta: dict):
+        self._tree = persistence.load_tensors(os.path.join(path, "data"))
+        self._num_features = meta.get("numFeatures", -1)
+
+
+class DecisionTreeRegressor(Regressor, _TreeParams):
+    def _fit(self, dataset: TensorFrame) -> "DecisionTreeRegressionModel":
+        x, y, w = self._extract_xyw(dataset)
+        edges, bins = ensure_binned(dataset, x, self.getOrDefault("maxBins"))
+        comm = get_comm()
+        grad = (w * y).unsqueeze(1)
+        total_w = comm.all_reduce_scalar(float(w.sum()))
+        gp = self._grow_params(total_w)
+        mask = w > 0 if bool((w == 0).any()) else None
+        tree = grow_tree(bins, edges, grad, w, gp, comm, row_mask=mask)
+        model = DecisionTreeRegressionModel()
+        model._set_tree(tree, x.shape[1])
+        model._copy_cols_from(self)
+        return model
+
+    def _copy_cols_to(self, model):
+        pass
+
+
+class DecisionTreeRegressionModel(RegressionModel, _TreeParams, _TreeModelMixin):
+    def _copy_cols_from(self, est):
+        for p in ("featuresCol", "labelCol", "predictionCol"):
+            self.set(p, est.getOrDefault(p))
+
+    def predict(self, features: torch.Tensor) -> torch.Tensor:
+        return self._predict_values(features.float()).squeeze(1)
+
+
+class DecisionTreeClassifier(ProbabilisticClassifier, _TreeParams):
+    def _fit(self, dataset: TensorFrame) -> "DecisionTreeClassificationModel":
+        x, y, w = self._extract_xyw(dataset)
+        k = self._get_num_classes(dataset)
+        comm = get_comm()
+        k = int(comm.all_reduce_scalar(k, "max"))
+        edges, bins = ensure_binned(dataset, x, self.getOrDefault("maxBins"))
+        onehot = torch.zeros(x.shape[0], k, dtype=torch.float32, device=x.device)
+        onehot.scatter_(1, y.long().unsqueeze(1), 1.0)
+        grad = onehot * w.unsqueeze(1)
+        total_w = comm.all_reduce_scalar(float(w.sum()))
+        gp = self._grow_params(total_w)
+        mask = w > 0 if bool((w == 0).any()) else None
+        tree = grow_tree(bins, edges, grad, w, gp, comm, row_mask=mask)
+        model = DecisionTreeClassificationModel()
+        model._set_tree(tree, x.shape[1])
+        model._num_classes = k
+        model._copy_cols_from(self)
+        return model
+
+
+class DecisionTreeClassificationModel(
+    ProbabilisticClassificationModel, _TreeParams, _TreeModelMixin
+):
+    def _copy_cols_from(self, est):
+        for p in (
+            "featuresCol",
+            "labelCol",
+            "predictionCol",
+            "rawPredictionCol",
+            "probabilityCol",
+        ):
+            self.set(p, est.getOrDefault(p))
+
+    def predictRaw(self, features: torch.Tensor) -> torch.Tensor:
+        # leaf values are class-probability vectors; raw = probabilities
+        return self._predict_values(features.float())
+
+    def raw2probabilityInPlace(self, raw: torch.Tensor) -> torch.Tensor:
+        s = raw.sum(dim=1, keepdim=True).clamp_min(1e-12)
+        raw /= s
+        return raw
+
+    def _extra_meta(self):
+        return {"numClasses": self._num_classes}
+
+    def _load_extra(self, path: str, meta: dict):
+        super()._load_extra(path, meta)
+        self._num_classes = meta.get("numClasses", -1)

@@ -1,0 +1,190 @@
+"""Op dispatch: HIP/CDNA4 kernels on GPU, torch reference on CPU.
+
+Policy: on a GPU box the hand-written gfx950 extension (built in-tree from
+csrc/ into spark_ensemble_amd/_hip_ops.so) is REQUIRED for the designated
+native ops — a missing extension raises instead of silently falling back to
+eager torch, so a passing GPU test run always means the HIP path ran.  Set
+SEA_ALLOW_EAGER=1 to override for debugging only.
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Optional
+
+import torch
+
+from . import reference
+
+_hip = None
+_hip_error: Optional[str] = None
+
+
+def _load_hip():
+    global _hip, _hip_error
+    if _hip is not None or _hip_error is not None:
+        return _hip
+    try:
+        from . import hip_ext
+
+        _hip = hip_ext.load()
+    except Exception as e:  # noqa: BLE001
+        _hip_error = f"{type(e).__name__}: {e}"
+        _hip = None
+    return _hip
+
+
+def hip_available() -> bool:
+    return _load_hip() is not None
+
+
+def _require_hip(op_name: str):
+    m = _load_hip()
+    if m is None:
+        if os.environ.get("SEA_ALLOW_EAGER") == "1":
+            return None
+        raise RuntimeError(
+            f"HIP extension required for {op_name} on GPU but not available "
+            f"({_hip_error}). Build it with `python setup.py build_ext --inplace` "
+            f"or set SEA_ALLOW_EAGER=1 to run the slow eager fallback."
+        )
+    return m
+
+
+# ---------------------------------------------------------------------------
+
+
+def sample_weights(replacement, ratio, n, seed, device=None, base_weight=None, rank=0):
+    dev = torch.device(device) if device is not None else torch.device("cpu")
+    if dev.type == "cuda":
+        m = _require_hip("sample_weights")
+        if m is not None:
+            w = torch.empty(n, dtype=torch.float32, device=dev)
+            m.sample_weights(w, bool(replacement), float(ratio), int(seed), int(rank))
+            if base_weight is not None:
+                w *= base_weight
+            return w
+    return reference.sample_weights(replacement, ratio, n, seed, device, base_weight, rank)
+
+
+def quantile_bins(x, max_bins=256, sample_rows=262_144, seed=17):
+    # sort-based; torch sort is fine on both devices (one-time cost per fit)
+    return reference.quantile_bins(x, max_bins, sample_rows, seed)
+
+
+def bin_features(x, edges):
+    if x.is_cuda:
+        m = _require_hip("bin_features")
+        if m is not None:
+            out = torch.empty(x.shape, dtype=torch.uint8, device=x.device)
+            m.bin_features(out, x.contiguous(), edges.contiguous())
+            return out
+    return reference.bin_features(x, edges)
+
+
+def hist_build(bins, gh, row_idx, node_offsets, num_bins):
+    if bins.is_cuda:
+        m = _require_hip("hist_build")
+        if m is not None:
+            n_nodes = node_offsets.numel() - 1
+            F = bins.shape[1]
+            C = gh.shape[1]
+            out = torch.zeros(
+                n_nodes, F, num_bins, C, dtype=torch.float32, device=bins.device
+            )
+            m.hist_build(
+                out,
+                bins,
+                gh.contiguous(),
+                row_idx.to(torch.int32),
+                node_offsets.to(torch.int64).cpu(),
+                int(num_bins),
+            )
+            return out
+    return reference.hist_build(bins, gh, row_idx, node_offsets, num_bins)
+
+
+def split_search(hist, lam=1e-6, min_child_weight=0.0, min_instances=1.0, min_info_gain=0.0):
+    return reference.split_search(hist, lam, min_child_weight, min_instances, min_info_gain)
+
+
+def partition_rows(bins, row_idx, node_offsets, feat, thr):
+    if bins.is_cuda:
+        m = _require_hip("partition_rows")
+        if m is not None:
+            return _partition_hip(m, bins, row_idx, node_offsets, feat, thr)
+    return reference.partition_rows(bins, row_idx, node_offsets, feat, thr)
+
+
+def _partition_hip(m, bins, row_idx, node_offsets, feat, thr):
+    n = node_offsets.numel() - 1
+    offs_cpu = node_offsets.to(torch.int64).cpu()
+    row_idx = row_idx.to(torch.int32)
+    new_rows = torch.empty_like(row_idx)
+    left_counts = torch.zeros(n, dtype=torch.int32, device=bins.device)
+    m.partition_rows(
+        new_rows,
+        left_counts,
+        bins,
+        row_idx,
+        offs_cpu,
+        feat.to(torch.int32).to(bins.device),
+        thr.to(torch.int32).to(bins.device),
+    )
+    lc = left_counts.cpu().to(torch.int64)
+    offs = offs_cpu
+    sizes = torch.empty(2 * n, dtype=torch.int64)
+    seg = offs[1:] - offs[:-1]
+    sizes[0::2] = lc
+    sizes[1::2] = seg - lc
+    new_offs = torch.cat([torch.zeros(1, dtype=torch.int64), sizes.cumsum(0)])
+    return new_rows, new_offs, lc
+
+
+def tree_predict(x, feature, threshold, left_child, leaf_value, max_depth):
+    if x.is_cuda:
+        m = _require_hip("tree_predict")
+        if m is not None:
+            N = x.shape[0]
+            D = leaf_value.shape[1]
+            out = torch.empty(N, D, dtype=torch.float32, device=x.device)
+            m.tree_predict(
+                out,
+                x.contiguous(),
+                feature.to(torch.int32).to(x.device),
+                threshold.to(torch.float32).to(x.device),
+                left_child.to(torch.int32).to(x.device),
+                leaf_value.to(torch.float32).to(x.device).contiguous(),
+            )
+            return out
+    return reference.tree_predict(x, feature, threshold, left_child, leaf_value, max_depth)
+
+
+def forest_predict(x, trees, weights=None, max_depth=64):
+    if x.is_cuda and trees:
+        m = _require_hip("forest_predict")
+        if m is not None:
+            return _forest_predict_hip(m, x, trees, weights)
+    return reference.forest_predict(x, trees, weights, max_depth)
+
+
+def _forest_predict_hip(m, x, trees, weights):
+    # pack all trees into one node-array arena so a single kernel launch
+    # walks every (row, tree) pair
+    dev = x.device
+    feats = torch.cat([t["feature"].to(torch.int32) for t in trees]).to(dev)
+    thrs = torch.cat([t["threshold"].to(torch.float32) for t in trees]).to(dev)
+    lefts = torch.cat([t["left_child"].to(torch.int32) for t in trees]).to(dev)
+    leaves = torch.cat([t["leaf_value"].to(torch.float32) for t in trees]).to(dev)
+    sizes = torch.tensor([t["feature"].numel() for t in trees], dtype=torch.int64)
+    offsets = torch.cat([torch.zeros(1, dtype=torch.int64), sizes.cumsum(0)])[:-1]
+    offsets = offsets.to(torch.int32).to(dev)
+    D = trees[0]["leaf_value"].shape[1]
+    T = len(trees)
+    if weights is None:
+        w = torch.ones(T, dtype=torch.float32, device=dev)
+    else:
+        w = weights.to(torch.float32).to(dev)
+    out = torch.zeros(x.shape[0], D, dtype=torch.float32, device=dev)
+    m.forest_predict(out, x.contiguous(), feats, thrs, lefts, leaves.contiguous(), offsets, w, D)
+    return out

@@ -1,0 +1,104 @@
+"""Data loading: libsvm files and synthetic tabular generators.
+
+The reference's test datasets are libsvm format (reference data/*.svm,
+loaded at e.g. GBMClassifierSuite.scala:53-58); the bench uses synthetic
+data of the BASELINE.json shapes (no network access for real datasets).
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import numpy as np
+import torch
+
+from ..frame import TensorFrame
+
+
+def load_libsvm(path: str, num_features: Optional[int] = None, device=None) -> TensorFrame:
+    """Dense-load a libsvm file: 'label idx:val idx:val ...' (1-based idx)."""
+    labels = []
+    rows = []
+    cols = []
+    vals = []
+    with open(path) as f:
+        for i, line in enumerate(f):
+            parts = line.split()
+            if not parts:
+                continue
+            labels.append(float(parts[0]))
+            for tok in parts[1:]:
+                j, v = tok.split(":")
+                rows.append(i)
+                cols.append(int(j) - 1)
+                vals.append(float(v))
+    n = len(labels)
+    f_count = num_features or (max(cols) + 1 if cols else 0)
+    x = np.zeros((n, f_count), dtype=np.float32)
+    x[rows, cols] = vals
+    y = np.asarray(labels, dtype=np.float32)
+    # normalize +-1 labels to {0,1} for classification-style files
+    uniq = np.unique(y)
+    if set(uniq.tolist()) == {-1.0, 1.0}:
+        y = (y + 1.0) / 2.0
+    xt = torch.from_numpy(x)
+    yt = torch.from_numpy(y)
+    if device is not None:
+        xt, yt = xt.to(device), yt.to(device)
+    return TensorFrame(features=xt, label=yt)
+
+
+def synthetic_classification(
+    n: int,
+    f: int,
+    k: int = 2,
+    seed: int = 7,
+    device=None,
+    informative: Optional[int] = None,
+    shard: Optional[tuple] = None,
+    noise: float = 0.35,
+) -> TensorFrame:
+    """Gaussian-blob + nonlinear-boundary synthetic classification data.
+
+    Labels depend on a random linear map of the features plus interactions,
+    then argmax with noise — learnable by trees and by logistic regression,
+    not trivially separable.  ``shard=(rank, world)`` makes rank-local rows
+    of a deterministic global dataset.
+    """
+    rank, world = shard or (0, 1)
+    g = torch.Generator().manual_seed(seed * 9176 + rank)
+    n_local = n // world + (1 if rank < n % world else 0)
+    x = torch.randn(n_local, f, generator=g)
+    informative = informative or min(f, 32)
+    wm = torch.randn(informative, k, generator=g)
+    margin = x[:, :informative] @ wm
+    # nonlinear structure: pairwise interaction of first features
+    inter = (x[:, 0] * x[:, 1]).unsqueeze(1) * torch.randn(1, k, generator=g)
+    margin = margin + inter + noise * torch.randn(n_local, k, generator=g)
+    y = margin.argmax(dim=1).float()
+    if device is not None:
+        x, y = x.to(device), y.to(device)
+    return TensorFrame(features=x, label=y)
+
+
+def synthetic_regression(
+    n: int,
+    f: int,
+    seed: int = 7,
+    device=None,
+    informative: Optional[int] = None,
+    shard: Optional[tuple] = None,
+    noise: float = 0.1,
+) -> TensorFrame:
+    rank, world = shard or (0, 1)
+    g = torch.Generator().manual_seed(seed * 7919 + rank)
+    n_local = n // world + (1 if rank < n % world else 0)
+    x = torch.randn(n_local, f, generator=g)
+    informative = informative or min(f, 32)
+    w = torch.randn(informative, generator=g)
+    y = x[:, :informative] @ w
+    y = y + 0.5 * (x[:, 0] * x[:, 1]) + torch.sin(x[:, 2] * 2.0)
+    y = y + noise * torch.randn(n_local, generator=g)
+    if device is not None:
+        x, y = x.to(device), y.to(device)
+    return TensorFrame(features=x, label=y)
