@@ -1,0 +1,295 @@
+"""GBM loss family.
+
+Vectorized re-implementation of reference boosting/GBMLoss.scala:107-318
+(semantics checked line-by-line against the reference; the torch forms below
+operate on [N, D] tensors instead of per-row arrays).  These torch
+implementations are the CPU path AND the GPU goldens; the fused HIP
+``grad_hess`` / ``line_search_eval`` kernels (csrc/) reproduce them per
+loss id.
+
+Conventions (as in the reference):
+  * ``encode_label``: raw label column -> [N, D] encoded target
+    (regression: identity; LogLoss: one-hot; Exponential/Bernoulli: 2y-1)
+  * ``gradient`` is d loss / d prediction (so pseudo-residuals are
+    NEGATIVE gradients, reference GBMLoss.scala:88-93)
+  * classification losses expose ``raw2probability``; BernoulliLoss keeps
+    the reference's sign quirk (class-1 probability = 1/(1+exp(+raw)),
+    reference GBMLoss.scala:311-316 — opposite sign vs ExponentialLoss).
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+# loss ids shared with the HIP kernels (csrc/gbm_losses.hip)
+LOSS_IDS = {
+    "squared": 0,
+    "absolute": 1,
+    "logcosh": 2,
+    "scaledlogcosh": 3,
+    "huber": 4,
+    "quantile": 5,
+    "logloss": 6,
+    "exponential": 7,
+    "bernoulli": 8,
+}
+
+
+class GBMLoss:
+    name: str = "?"
+    dim: int = 1
+    has_hessian: bool = False
+    # scalar parameter forwarded to the fused kernels (alpha/delta/quantile)
+    param: float = 0.0
+
+    def encode_label(self, y: torch.Tensor) -> torch.Tensor:
+        """[N] -> [N, dim]"""
+        return y.unsqueeze(1)
+
+    def loss(self, label: torch.Tensor, pred: torch.Tensor) -> torch.Tensor:
+        """[N, dim], [N, dim] -> [N]"""
+        raise NotImplementedError
+
+    def gradient(self, label: torch.Tensor, pred: torch.Tensor) -> torch.Tensor:
+        """[N, dim] -> [N, dim]"""
+        raise NotImplementedError
+
+    def hessian(self, label: torch.Tensor, pred: torch.Tensor) -> torch.Tensor:
+        raise NotImplementedError(f"{self.name} has no hessian")
+
+    def negative_gradient(self, label, pred):
+        return -self.gradient(label, pred)
+
+    @property
+    def loss_id(self) -> int:
+        return LOSS_IDS[self.name]
+
+
+class _ClassificationLoss(GBMLoss):
+    def raw2probability(self, raw: torch.Tensor) -> torch.Tensor:
+        """[N, dim_raw] -> [N, K] probabilities"""
+        raise NotImplementedError
+
+
+class SquaredLoss(GBMLoss):
+    name = "squared"
+    has_hessian = True
+
+    def loss(self, label, pred):
+        return ((label - pred) ** 2).sum(dim=1) / 2.0
+
+    def gradient(self, label, pred):
+        return -(label - pred)
+
+    def hessian(self, label, pred):
+        return torch.ones_like(pred)
+
+
+class AbsoluteLoss(GBMLoss):
+    name = "absolute"
+
+    def loss(self, label, pred):
+        return (label - pred).abs().sum(dim=1)
+
+    def gradient(self, label, pred):
+        return -torch.sign(label - pred)
+
+
+class LogCoshLoss(GBMLoss):
+    name = "logcosh"
+    has_hessian = True
+
+    def loss(self, label, pred):
+        d = label - pred
+        # numerically stable log(cosh(d)) = |d| + log1p(exp(-2|d|)) - log 2
+        a = d.abs()
+        return (a + torch.log1p(torch.exp(-2 * a)) - 0.6931471805599453).sum(dim=1)
+
+    def gradient(self, label, pred):
+        return -torch.tanh(label - pred)
+
+    def hessian(self, label, pred):
+        return 1.0 / torch.cosh(label - pred) ** 2
+
+
+class ScaledLogCoshLoss(GBMLoss):
+    name = "scaledlogcosh"
+    has_hessian = True
+
+    def __init__(self, alpha: float):
+        self.alpha = float(alpha)
+        self.param = self.alpha
+
+    def _scale(self, label, pred):
+        return torch.where(label > pred, self.alpha, 1.0 - self.alpha)
+
+    def loss(self, label, pred):
+        return (self._scale(label, pred) * _logcosh(label - pred)).sum(dim=1)
+
+    def gradient(self, label, pred):
+        return self._scale(label, pred) * (-torch.tanh(label - pred))
+
+    def hessian(self, label, pred):
+        return self._scale(label, pred) / torch.cosh(label - pred) ** 2
+
+
+def _logcosh(d):
+    a = d.abs()
+    return a + torch.log1p(torch.exp(-2 * a)) - 0.6931471805599453
+
+
+class HuberLoss(GBMLoss):
+    name = "huber"
+
+    def __init__(self, delta: float):
+        self.delta = float(delta)
+        self.param = self.delta
+
+    def loss(self, label, pred):
+        d = label - pred
+        quad = d * d / 2.0
+        lin = self.delta * (d.abs() - self.delta / 2.0)
+        return torch.where(d.abs() <= self.delta, quad, lin).sum(dim=1)
+
+    def gradient(self, label, pred):
+        d = label - pred
+        return torch.where(d.abs() <= self.delta, -d, -self.delta * torch.sign(d))
+
+
+class QuantileLoss(GBMLoss):
+    name = "quantile"
+
+    def __init__(self, quantile: float):
+        self.quantile = float(quantile)
+        self.param = self.quantile
+
+    def loss(self, label, pred):
+        d = label - pred
+        return torch.where(d > 0, self.quantile * d, (self.quantile - 1.0) * d).sum(
+            dim=1
+        )
+
+    def gradient(self, label, pred):
+        d = label - pred
+        q = self.quantile
+        return torch.where(
+            d > 0, torch.full_like(d, -q), torch.full_like(d, 1.0 - q)
+        )
+
+
+class LogLoss(_ClassificationLoss):
+    """Multiclass cross-entropy on K raw scores (reference
+    GBMLoss.scala:196-263).  dim = K, so a GBM round fits K trees."""
+
+    name = "logloss"
+    has_hessian = True
+
+    def __init__(self, num_classes: int):
+        self.num_classes = int(num_classes)
+        self.dim = self.num_classes
+
+    def encode_label(self, y):
+        onehot = torch.zeros(
+            y.shape[0], self.num_classes, dtype=torch.float32, device=y.device
+        )
+        onehot.scatter_(1, y.long().unsqueeze(1), 1.0)
+        return onehot
+
+    def loss(self, label, pred):
+        logp = pred.log_softmax(dim=1)
+        return -(label * logp).sum(dim=1)
+
+    def gradient(self, label, pred):
+        return pred.softmax(dim=1) - label
+
+    def hessian(self, label, pred):
+        p = pred.softmax(dim=1)
+        return p * (1 - p)
+
+    def raw2probability(self, raw):
+        return raw.softmax(dim=1)
+
+
+class ExponentialLoss(_ClassificationLoss):
+    """AdaBoost exponential loss on y in {-1, 1} (reference
+    GBMLoss.scala:265-291); dim = 1."""
+
+    name = "exponential"
+    has_hessian = True
+
+    def encode_label(self, y):
+        return (2 * y - 1).unsqueeze(1)
+
+    def loss(self, label, pred):
+        return torch.exp(-label * pred).sum(dim=1)
+
+    def gradient(self, label, pred):
+        return -label * torch.exp(-label * pred)
+
+    def hessian(self, label, pred):
+        return label * label * torch.exp(-label * pred)
+
+    def raw2probability(self, raw):
+        p1 = 1.0 / (1.0 + torch.exp(-2.0 * raw[:, 0]))
+        return torch.stack([1.0 - p1, p1], dim=1)
+
+
+class BernoulliLoss(_ClassificationLoss):
+    """Logistic loss on +-1 labels (reference GBMLoss.scala:293-318);
+    NOTE the reference's raw2probability uses exp(+raw) for class 1 —
+    replicated exactly."""
+
+    name = "bernoulli"
+    has_hessian = True
+
+    def encode_label(self, y):
+        return (2 * y - 1).unsqueeze(1)
+
+    def loss(self, label, pred):
+        # log(1 + exp(-2 y p)), stable
+        z = -2 * label * pred
+        return (torch.clamp_min(z, 0) + torch.log1p(torch.exp(-z.abs()))).sum(dim=1)
+
+    def gradient(self, label, pred):
+        return -2 * label / (1 + torch.exp(2 * label * pred))
+
+    def hessian(self, label, pred):
+        e = torch.exp(2 * pred * label)
+        return 4 * e * label * label / (1 + e) ** 2
+
+    def raw2probability(self, raw):
+        # reference sign quirk (GBMLoss.scala:311-316)
+        p1 = 1.0 / (1.0 + torch.exp(raw[:, 0]))
+        return torch.stack([1.0 - p1, p1], dim=1)
+
+
+def get_regression_loss(
+    name: str, alpha: float = 0.9, delta: float = 1.0, quantile: Optional[float] = None
+) -> GBMLoss:
+    name = name.lower()
+    if name == "squared":
+        return SquaredLoss()
+    if name == "absolute":
+        return AbsoluteLoss()
+    if name == "logcosh":
+        return LogCoshLoss()
+    if name == "scaledlogcosh":
+        return ScaledLogCoshLoss(alpha)
+    if name == "huber":
+        return HuberLoss(delta)
+    if name == "quantile":
+        return QuantileLoss(quantile if quantile is not None else alpha)
+    raise ValueError(f"unknown regression loss {name!r}")
+
+
+def get_classification_loss(name: str, num_classes: int = 2) -> _ClassificationLoss:
+    name = name.lower()
+    if name == "logloss":
+        return LogLoss(num_classes)
+    if name == "exponential":
+        return ExponentialLoss()
+    if name == "bernoulli":
+        return BernoulliLoss()
+    raise ValueError(f"unknown classification loss {name!r}")

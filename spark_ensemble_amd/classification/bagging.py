@@ -1,0 +1,177 @@
+"""BaggingClassifier (reference classification/BaggingClassifier.scala).
+
+Voting (reference :55-67, 260-287): ``soft`` sums per-model
+``predictProbability``; ``hard`` sums one-hot votes of per-model ``predict``
+(default hard, :67); soft voting with a non-probabilistic base model raises
+(:275-277).  ``raw2probability`` divides by numModels (:285-287).
+Same per-learner subspace seed+i scheme and seed deviation as
+BaggingRegressor (see its docstring).
+"""
+
+from __future__ import annotations
+
+import os
+from typing import List
+
+import torch
+
+from .. import persistence
+from ..ensemble.binning import BinnedDataset
+from ..ensemble.params import (
+    HasBaseLearner,
+    HasNumBaseLearners,
+    HasParallelism,
+    HasSubBag,
+)
+from ..ensemble.utils import slice_features, subspace
+from ..estimator import (
+    ProbabilisticClassificationModel,
+    ProbabilisticClassifier,
+)
+from ..frame import TensorFrame
+from ..params import ParamValidators
+from ..parallel import get_comm
+
+
+class _BaggingClassifierParams(
+    HasNumBaseLearners, HasBaseLearner, HasParallelism, HasSubBag
+):
+    def _declare_params(self):
+        super()._declare_params()
+        self.votingStrategy = self._str_param(
+            "votingStrategy",
+            "hard (majority vote) or soft (mean probability)",
+            ParamValidators.inArray(["hard", "soft"]),
+        )
+        self.seed = self._int_param("seed", "random seed")
+        self._setDefault(numBaseLearners=10, votingStrategy="hard", seed=0)
+
+    def getVotingStrategy(self):
+        return self.getOrDefault("votingStrategy")
+
+    def setVotingStrategy(self, v):
+        return self.set("votingStrategy", v)
+
+    def setSeed(self, v):
+        return self.set("seed", v)
+
+
+class BaggingClassifier(ProbabilisticClassifier, _BaggingClassifierParams):
+    def _default_base_learner(self):
+        from ..models.tree import DecisionTreeClassifier
+
+        return DecisionTreeClassifier()
+
+    def _fit(self, dataset: TensorFrame) -> "BaggingClassificationModel":
+        comm = get_comm()
+        learner = self.getOrNone("baseLearner") or self._default_base_learner()
+        seed = self.getOrDefault("seed")
+        k = self.getNumBaseLearners()
+        x, y, w = self._extract_xyw(dataset)
+        n, num_features = x.shape
+        num_classes = int(comm.all_reduce_scalar(self._get_num_classes(dataset), "max"))
+        binned = BinnedDataset(x, dataset)
+
+        subspaces = [
+            subspace(self.getSubspaceRatio(), num_features, seed + i) for i in range(k)
+        ]
+        models = []
+        for i in range(k):
+            bag_w = self.sample_weights(
+                self.getReplacement(),
+                self.getSubsampleRatio(),
+                n,
+                seed + i,
+                x.device,
+                w,
+                comm.rank,
+            )
+            fr = binned.fit_frame(learner, y, bag_w, subspaces[i])
+            models.append(self.fit_base_learner(learner, fr, weight_col="weight"))
+
+        model = BaggingClassificationModel()
+        model._models = models
+        model._subspaces = subspaces
+        model._num_classes = num_classes
+        model._num_features = num_features
+        model.set("votingStrategy", self.getVotingStrategy())
+        for p in (
+            "featuresCol", "labelCol", "predictionCol",
+            "rawPredictionCol", "probabilityCol",
+        ):
+            model.set(p, self.getOrDefault(p))
+        return model
+
+    def _save_impl(self, path: str):
+        persistence.save_metadata(self, path)
+        self._save_learner(path)
+
+    def _load_extra(self, path: str, meta: dict):
+        self.setBaseLearner(self._load_learner(path))
+
+
+class BaggingClassificationModel(
+    ProbabilisticClassificationModel, _BaggingClassifierParams
+):
+    _models: List = []
+    _subspaces: List[torch.Tensor] = []
+
+    @property
+    def numModels(self):
+        return len(self._models)
+
+    def predictRaw(self, features: torch.Tensor) -> torch.Tensor:
+        x = features.float()
+        k = self._num_classes
+        soft = self.getVotingStrategy() == "soft"
+        acc = torch.zeros(x.shape[0], k, dtype=torch.float32, device=x.device)
+        for sub, m in zip(self._subspaces, self._models):
+            xs = slice_features(x, sub)
+            if soft:
+                if not hasattr(m, "predictProbability"):
+                    raise RuntimeError(
+                        "soft voting requires probabilistic base models "
+                        "(reference BaggingClassifier.scala:275-277)"
+                    )
+                acc += m.predictProbability(xs)
+            else:
+                pred = m.predict(xs).long()
+                acc.scatter_add_(
+                    1, pred.unsqueeze(1),
+                    torch.ones(x.shape[0], 1, device=x.device),
+                )
+        return acc
+
+    def raw2probabilityInPlace(self, raw: torch.Tensor) -> torch.Tensor:
+        raw /= len(self._models)
+        return raw
+
+    def _save_impl(self, path: str):
+        persistence.save_metadata(
+            self, path,
+            extra={
+                "numClasses": self._num_classes,
+                "numModels": len(self._models),
+                "numFeatures": self._num_features,
+            },
+        )
+        for i, m in enumerate(self._models):
+            m.save(os.path.join(path, f"model-{i}"), overwrite=True)
+            persistence.save_json_rows(
+                os.path.join(path, f"data-{i}"),
+                [{"subspace": self._subspaces[i].tolist()}],
+            )
+
+    def _load_extra(self, path: str, meta: dict):
+        self._num_classes = meta["numClasses"]
+        self._num_features = meta.get("numFeatures", -1)
+        self._models = []
+        self._subspaces = []
+        i = 0
+        while os.path.isdir(os.path.join(path, f"model-{i}")):
+            self._models.append(
+                persistence.load_instance(os.path.join(path, f"model-{i}"))
+            )
+            row = persistence.load_json_rows(os.path.join(path, f"data-{i}"))[0]
+            self._subspaces.append(torch.tensor(row["subspace"], dtype=torch.long))
+            i += 1

@@ -46,17 +46,7 @@ class HasNumBaseLearners(Params):
         return self.set("numBaseLearners", v)
 
 
-class HasBaseLearner(Params):
-    def _declare_params(self):
-        super()._declare_params()
-        self.baseLearner = self._param("baseLearner", "base learner estimator")
-
-    def getBaseLearner(self):
-        return self.getOrDefault("baseLearner")
-
-    def setBaseLearner(self, v):
-        return self.set("baseLearner", v)
-
+class FitsBaseLearners(Params):
     # The single choke-point through which every meta-estimator fits its
     # base learner (reference ensembleParams.scala:64-81): re-target the
     # label/features/prediction/weight columns of a COPY of the learner and
@@ -78,6 +68,18 @@ class HasBaseLearner(Params):
             lr.set("weightCol", weight_col)
         return lr.fit(dataset)
 
+
+class HasBaseLearner(FitsBaseLearners):
+    def _declare_params(self):
+        super()._declare_params()
+        self.baseLearner = self._param("baseLearner", "base learner estimator")
+
+    def getBaseLearner(self):
+        return self.getOrDefault("baseLearner")
+
+    def setBaseLearner(self, v):
+        return self.set("baseLearner", v)
+
     # -- persistence (learner/ subdir; ensembleParams.scala:87-103) --------
     def _save_learner(self, path: str):
         self.getBaseLearner().save(os.path.join(path, "learner"), overwrite=True)
@@ -87,7 +89,7 @@ class HasBaseLearner(Params):
         return persistence.load_instance(os.path.join(path, "learner"))
 
 
-class HasBaseLearners(Params):
+class HasBaseLearners(FitsBaseLearners):
     def _declare_params(self):
         super()._declare_params()
         self.baseLearners = self._param(
@@ -114,7 +116,7 @@ class HasBaseLearners(Params):
         return out
 
 
-class HasStacker(Params):
+class HasStacker(FitsBaseLearners):
     def _declare_params(self):
         super()._declare_params()
         self.stacker = self._param("stacker", "meta-learner fit on base outputs")
