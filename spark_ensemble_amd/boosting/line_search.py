@@ -57,7 +57,13 @@ def optimize_weight_1d(
             payload = _eval(loss, label, pred, direction, weight, a)
             if comm is not None and comm.is_distributed:
                 comm.all_reduce_(payload)
-            cache[a] = float(payload[0])
+            v = float(payload[0])
+            if not np.isfinite(v):
+                # overflowed exp-style loss: steer Brent back toward the
+                # finite region instead of feeding it inf/nan (golden-section
+                # otherwise converges to the boundary)
+                v = 1e30 * (1.0 + a)
+            cache[a] = v
         return cache[a]
 
     res = minimize_scalar(
@@ -90,7 +96,12 @@ def optimize_weight_nd(
         payload = _eval(loss, label, pred, direction, weight, a)
         if comm is not None and comm.is_distributed:
             comm.all_reduce_(payload)
-        return float(payload[0]), payload[1:].cpu().double().numpy()
+        v = float(payload[0])
+        g = payload[1:].cpu().double().numpy()
+        if not np.isfinite(v):
+            v = 1e30 * (1.0 + float(np.abs(a_np).sum()))
+            g = np.nan_to_num(g, nan=0.0, posinf=1e30, neginf=-1e30)
+        return v, g
 
     res = minimize(
         f,

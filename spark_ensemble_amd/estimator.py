@@ -160,13 +160,18 @@ class _PredictorParams(Params):
         return self.set("weightCol", v)
 
     # -- instance extraction ---------------------------------------------
-    def _extract_xyw(self, dataset: TensorFrame):
+    def _extract_xyw(self, dataset: TensorFrame, require_label: bool = True):
         """(features [N,F] f32, label [N] f32, weight [N] f32) — the analog
         of Spark's ``extractInstances`` (used at reference
         BaggingRegressor.scala:136)."""
         x = dataset[self.getFeaturesCol()].float()
         label_col = self.getLabelCol()
-        y = dataset[label_col].float() if label_col in dataset else None
+        if label_col in dataset:
+            y = dataset[label_col].float()
+        elif require_label:
+            raise KeyError(f"label column {label_col!r} not found in dataset")
+        else:
+            y = None
         wcol = self.getWeightCol()
         if wcol and wcol in dataset:
             w = dataset[wcol].float()

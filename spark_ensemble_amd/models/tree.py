@@ -161,7 +161,7 @@ class DecisionTreeRegressor(Regressor, _TreeParams):
         pass
 
 
-class DecisionTreeRegressionModel(RegressionModel, _TreeParams, _TreeModelMixin):
+class DecisionTreeRegressionModel(_TreeModelMixin, RegressionModel, _TreeParams):
     def _copy_cols_from(self, est):
         for p in ("featuresCol", "labelCol", "predictionCol"):
             self.set(p, est.getOrDefault(p))
@@ -192,7 +192,7 @@ class DecisionTreeClassifier(ProbabilisticClassifier, _TreeParams):
 
 
 class DecisionTreeClassificationModel(
-    ProbabilisticClassificationModel, _TreeParams, _TreeModelMixin
+    _TreeModelMixin, ProbabilisticClassificationModel, _TreeParams
 ):
     def _copy_cols_from(self, est):
         for p in (

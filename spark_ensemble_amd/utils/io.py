@@ -57,24 +57,28 @@ def synthetic_classification(
     informative: Optional[int] = None,
     shard: Optional[tuple] = None,
     noise: float = 0.35,
+    split: int = 0,
 ) -> TensorFrame:
-    """Gaussian-blob + nonlinear-boundary synthetic classification data.
+    """Gaussian + nonlinear-boundary synthetic classification data.
 
-    Labels depend on a random linear map of the features plus interactions,
-    then argmax with noise — learnable by trees and by logistic regression,
-    not trivially separable.  ``shard=(rank, world)`` makes rank-local rows
-    of a deterministic global dataset.
+    ``seed`` fixes the labeling FUNCTION (a random linear map plus an
+    interaction term); ``split``/``shard=(rank, world)`` select independent
+    row draws from the same task, so train (split=0) and test (split=1)
+    frames share a distribution.  Learnable by trees and by logistic
+    regression, not trivially separable.
     """
     rank, world = shard or (0, 1)
-    g = torch.Generator().manual_seed(seed * 9176 + rank)
+    task = torch.Generator().manual_seed(seed * 9176 + 4242)
+    informative = informative or min(f, 32)
+    wm = torch.randn(informative, k, generator=task)
+    wi = torch.randn(1, k, generator=task)
+
+    g = torch.Generator().manual_seed(seed * 9176 + 100003 * (split + 1) + rank)
     n_local = n // world + (1 if rank < n % world else 0)
     x = torch.randn(n_local, f, generator=g)
-    informative = informative or min(f, 32)
-    wm = torch.randn(informative, k, generator=g)
     margin = x[:, :informative] @ wm
-    # nonlinear structure: pairwise interaction of first features
-    inter = (x[:, 0] * x[:, 1]).unsqueeze(1) * torch.randn(1, k, generator=g)
-    margin = margin + inter + noise * torch.randn(n_local, k, generator=g)
+    margin = margin + (x[:, 0] * x[:, 1]).unsqueeze(1) * wi
+    margin = margin + noise * torch.randn(n_local, k, generator=g)
     y = margin.argmax(dim=1).float()
     if device is not None:
         x, y = x.to(device), y.to(device)
@@ -89,13 +93,16 @@ def synthetic_regression(
     informative: Optional[int] = None,
     shard: Optional[tuple] = None,
     noise: float = 0.1,
+    split: int = 0,
 ) -> TensorFrame:
     rank, world = shard or (0, 1)
-    g = torch.Generator().manual_seed(seed * 7919 + rank)
+    task = torch.Generator().manual_seed(seed * 7919 + 2424)
+    informative = informative or min(f, 32)
+    w = torch.randn(informative, generator=task)
+
+    g = torch.Generator().manual_seed(seed * 7919 + 100003 * (split + 1) + rank)
     n_local = n // world + (1 if rank < n % world else 0)
     x = torch.randn(n_local, f, generator=g)
-    informative = informative or min(f, 32)
-    w = torch.randn(informative, generator=g)
     y = x[:, :informative] @ w
     y = y + 0.5 * (x[:, 0] * x[:, 1]) + torch.sin(x[:, 2] * 2.0)
     y = y + noise * torch.randn(n_local, generator=g)

@@ -1,0 +1,77 @@
+import pytest
+import torch
+
+
+def pytest_configure(config):
+    config.addinivalue_line(
+        "markers", "gpu: tests that require a ROCm GPU (run on MI355X boxes)"
+    )
+    config.addinivalue_line(
+        "markers", "slow: long-running CPU tests"
+    )
+
+
+def pytest_collection_modifyitems(config, items):
+    if torch.cuda.is_available():
+        return
+    skip_gpu = pytest.mark.skip(reason="no GPU available")
+    for item in items:
+        if "gpu" in item.keywords:
+            item.add_marker(skip_gpu)
+
+
+@pytest.fixture(scope="session")
+def clf_frame():
+    from spark_ensemble_amd.utils.io import synthetic_classification
+
+    return synthetic_classification(4000, 20, k=3, seed=11)
+
+
+@pytest.fixture(scope="session")
+def clf_frame_test():
+    from spark_ensemble_amd.utils.io import synthetic_classification
+
+    return synthetic_classification(2000, 20, k=3, seed=11, split=1)
+
+
+@pytest.fixture(scope="session")
+def bin_frame():
+    from spark_ensemble_amd.utils.io import synthetic_classification
+
+    return synthetic_classification(4000, 20, k=2, seed=13)
+
+
+@pytest.fixture(scope="session")
+def bin_frame_test():
+    from spark_ensemble_amd.utils.io import synthetic_classification
+
+    return synthetic_classification(2000, 20, k=2, seed=13, split=1)
+
+
+@pytest.fixture(scope="session")
+def reg_frame():
+    from spark_ensemble_amd.utils.io import synthetic_regression
+
+    return synthetic_regression(4000, 20, seed=17)
+
+
+@pytest.fixture(scope="session")
+def reg_frame_test():
+    from spark_ensemble_amd.utils.io import synthetic_regression
+
+    return synthetic_regression(2000, 20, seed=17, split=1)
+
+
+def accuracy(model, frame):
+    out = model.transform(frame)
+    return float((out["prediction"] == frame["label"]).float().mean())
+
+
+def rmse(model, frame):
+    p = model.predict(frame["features"])
+    return float(((p - frame["label"]) ** 2).mean() ** 0.5)
+
+
+@pytest.fixture(scope="session")
+def metrics():
+    return {"accuracy": accuracy, "rmse": rmse}

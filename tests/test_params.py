@@ -1,0 +1,64 @@
+"""Param system unit tests (config surface of SURVEY.md section 5.6)."""
+
+import pytest
+
+from spark_ensemble_amd import BaggingClassifier, GBMRegressor
+from spark_ensemble_amd.models import DecisionTreeClassifier
+from spark_ensemble_amd.params import ParamValidators
+
+
+def test_defaults_match_reference():
+    # reference BaggingParams.scala:27-36, GBMParams.scala:121-129
+    b = BaggingClassifier()
+    assert b.getNumBaseLearners() == 10
+    assert b.getReplacement() is True
+    assert b.getSubsampleRatio() == 1.0
+    assert b.getSubspaceRatio() == 1.0
+    assert b.getVotingStrategy() == "hard"
+    g = GBMRegressor()
+    assert g.getOrDefault("optimizedWeights") is True
+    assert g.getOrDefault("updates") == "gradient"
+    assert g.getOrDefault("learningRate") == 1.0
+    assert g.getOrDefault("maxIter") == 100
+    assert g.getOrDefault("numRounds") == 1
+    assert g.getOrDefault("validationTol") == 0.01
+    assert g.getOrDefault("replacement") is False  # GBM overrides to False
+    assert g.getLoss() == "squared"
+
+
+def test_validators():
+    g = GBMRegressor()
+    with pytest.raises(ValueError):
+        g.set("learningRate", -1.0)
+    with pytest.raises(ValueError):
+        g.set("loss", "nonsense")
+    with pytest.raises(ValueError):
+        g.set("subsampleRatio", 0.0)
+    g.set("subsampleRatio", 0.5)
+    assert g.getSubsampleRatio() == 0.5
+
+
+def test_string_params_lowercased():
+    g = GBMRegressor().setLoss("SQUARED")
+    assert g.getLoss() == "squared"
+
+
+def test_copy_with_extra_and_nested():
+    dt = DecisionTreeClassifier().setMaxDepth(7)
+    b = BaggingClassifier().setBaseLearner(dt).setNumBaseLearners(5)
+    c = b.copy({"numBaseLearners": 3})
+    assert c.getNumBaseLearners() == 3
+    assert b.getNumBaseLearners() == 5
+    # nested estimator must be deep-copied (reference BaggingRegressor.scala:111-115)
+    assert c.getBaseLearner() is not dt
+    assert c.getBaseLearner().getOrDefault("maxDepth") == 7
+
+
+def test_explain_params():
+    text = GBMRegressor().explainParams()
+    assert "learningRate" in text and "default" in text
+
+
+def test_unknown_param_raises():
+    with pytest.raises(AttributeError):
+        GBMRegressor().set("bogus", 1)

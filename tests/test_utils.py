@@ -1,0 +1,79 @@
+"""Property tests for ensemble utils (reference HasSubBagSuite.scala:60-105,
+UtilsSuite.scala:29-67)."""
+
+import torch
+
+from spark_ensemble_amd.ensemble.utils import (
+    slice_features,
+    subspace,
+    weighted_median,
+)
+
+
+def naive_weighted_median(values, weights):
+    order = sorted(range(len(values)), key=lambda i: values[i])
+    total = sum(weights)
+    cum = 0.0
+    for i in order:
+        cum += weights[i]
+        if cum >= total / 2:
+            return values[i]
+    return values[order[-1]]
+
+
+def test_weighted_median_matches_naive():
+    g = torch.Generator().manual_seed(5)
+    for _ in range(25):
+        n = int(torch.randint(1, 30, (1,), generator=g))
+        v = torch.rand(n, generator=g)
+        w = torch.rand(n, generator=g) + 0.01
+        expect = naive_weighted_median(v.tolist(), w.tolist())
+        got = float(weighted_median(v, w))
+        assert abs(got - expect) < 1e-6
+
+
+def test_weighted_median_uniform_weights_is_median():
+    v = torch.tensor([3.0, 1.0, 2.0, 5.0, 4.0])
+    w = torch.ones(5)
+    assert float(weighted_median(v, w)) == 3.0
+
+
+def test_weighted_median_scaled_weights_invariant():
+    g = torch.Generator().manual_seed(6)
+    v = torch.rand(11, generator=g)
+    w = torch.rand(11, generator=g) + 0.1
+    assert float(weighted_median(v, w)) == float(weighted_median(v, w * 7.3))
+
+
+def test_weighted_median_rowwise():
+    v = torch.tensor([[1.0, 2.0, 3.0], [9.0, 7.0, 8.0]])
+    w = torch.ones(2, 3)
+    out = weighted_median(v, w)
+    assert out.tolist() == [2.0, 8.0]
+
+
+def test_subspace_properties():
+    for seed in range(10):
+        for ratio in (0.2, 0.5, 0.8):
+            idx = subspace(ratio, 200, seed)
+            # sorted, unique, within range
+            assert (idx[1:] > idx[:-1]).all()
+            assert idx.min() >= 0 and idx.max() < 200
+            # expected size ~ ratio * nF (loose bound)
+            assert abs(idx.numel() - ratio * 200) < 60
+        # deterministic in seed
+        assert subspace(0.5, 200, seed).tolist() == subspace(0.5, 200, seed).tolist()
+
+
+def test_subspace_ratio_one_is_identity():
+    idx = subspace(1.0, 37, 3)
+    assert idx.tolist() == list(range(37))
+
+
+def test_slice_features_gather():
+    x = torch.arange(12.0).reshape(3, 4)
+    idx = torch.tensor([0, 2])
+    out = slice_features(x, idx)
+    assert out.tolist() == [[0.0, 2.0], [4.0, 6.0], [8.0, 10.0]]
+    # identity short-circuit
+    assert slice_features(x, torch.arange(4)) is x
