@@ -44,6 +44,12 @@ class BinnedDataset:
     def get(self, max_bins: int) -> Tuple[torch.Tensor, torch.Tensor]:
         if max_bins not in self._by_bins:
             edges = ops.quantile_bins(self.x, max_bins)
+            from ..parallel import get_comm
+
+            comm = get_comm()
+            if comm.is_distributed:
+                # identical cut points on every rank (see tree_grower)
+                comm.broadcast_(edges, src=0)
             bins = ops.bin_features(self.x, edges)
             self._by_bins[max_bins] = (edges, bins)
         return self._by_bins[max_bins]

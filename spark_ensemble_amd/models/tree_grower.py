@@ -47,11 +47,20 @@ class GrowParams:
 
 def ensure_binned(frame, x: torch.Tensor, max_bins: int):
     """Bin ``x`` (cached on the frame so repeated fits over one features
-    tensor — every GBM round, every bagged learner — bin once)."""
+    tensor — every GBM round, every bagged learner — bin once).
+
+    Distributed: every rank must use IDENTICAL cut points (split thresholds
+    and histograms are only comparable across ranks then), so rank 0's
+    shard-sampled edges are broadcast before binning."""
     cached = frame.cache_get("bins", x, max_bins) if frame is not None else None
     if cached is not None:
         return cached
     edges = ops.quantile_bins(x, max_bins)
+    from ..parallel import get_comm
+
+    comm = get_comm()
+    if comm.is_distributed:
+        comm.broadcast_(edges, src=0)
     bins = ops.bin_features(x, edges)
     if frame is not None:
         frame.cache_put("bins", x, max_bins, (edges, bins))

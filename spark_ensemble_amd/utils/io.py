@@ -69,7 +69,7 @@ def synthetic_classification(
     """
     rank, world = shard or (0, 1)
     task = torch.Generator().manual_seed(seed * 9176 + 4242)
-    informative = informative or min(f, 32)
+    informative = min(informative or min(f, 32), f)
     wm = torch.randn(informative, k, generator=task)
     wi = torch.randn(1, k, generator=task)
 
@@ -97,7 +97,7 @@ def synthetic_regression(
 ) -> TensorFrame:
     rank, world = shard or (0, 1)
     task = torch.Generator().manual_seed(seed * 7919 + 2424)
-    informative = informative or min(f, 32)
+    informative = min(informative or min(f, 32), f)
     w = torch.randn(informative, generator=task)
 
     g = torch.Generator().manual_seed(seed * 7919 + 100003 * (split + 1) + rank)
