@@ -104,7 +104,7 @@ def main():
 
     def one_round(i: int):
         nonlocal predictions
-        res_label = -loss.gradient(ylab, predictions).squeeze(1)
+        res_label = -loss.grad_hess_fused(ylab, predictions)[0].squeeze(1)
         fr = binned.fit_frame(learner, res_label)
         model = gbm.fit_base_learner(learner, fr)
         direction = model.predict(x).unsqueeze(1)

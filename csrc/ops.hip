@@ -5,18 +5,20 @@
 //   * hist_build      — LDS-staged per-(node, feature, bin) grad/hess/count
 //                       histograms (replaces MLlib DecisionTree's
 //                       treeAggregate histogram rounds)
-//   * partition_rows  — single-pass two-ended node partition
+//   * partition_rows  — block-aggregated two-pass node partition
 //   * bin_features    — quantile binning (raw f32 -> uint8 bin ids)
 //   * tree_predict /
-//     forest_predict  — batched node-array tree walks (per-row model.predict
-//                       loops of every ensemble model)
+//     forest_predict  — batched node-array tree walks
 //   * sample_weights  — counter-based Poisson/Bernoulli row sampling
-//                       (RDD.sample semantics as weight vectors)
+//   * grad_hess /
+//     line_search_eval— fused per-row GBM loss work (one pass over the
+//                       shard + one short reduction payload)
 //
 // Design notes (MI355X_MICROARCH.md): 64-wide waves, 256-thread blocks;
-// histograms live in LDS (dynamic, <= 64 KiB per block keeps 2 blocks/CU);
-// global accumulation via device atomics once per block per bin; grids are
-// (row-chunk x feature-group) so launches have >> 256 workgroups.
+// histograms live in LDS (dynamic, <= 48 KiB keeps 3 blocks/CU resident);
+// chunk sizes are chosen host-side so the grid is ~2-3x the resident block
+// capacity (minimizes the global-atomic flush that dominated the naive
+// version); single-chunk nodes flush with plain stores.
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -41,8 +43,7 @@ __device__ inline uint64_t splitmix64(uint64_t z) {
 }
 
 __device__ inline float u01(uint64_t z) {
-  // upper 24 bits -> (0, 1]
-  return ((z >> 40) + 1) * (1.0f / 16777216.0f);
+  return ((z >> 40) + 1) * (1.0f / 16777216.0f);  // upper 24 bits -> (0, 1]
 }
 
 __global__ void sample_weights_kernel(float* __restrict__ out, int64_t n,
@@ -87,6 +88,9 @@ void sample_weights(torch::Tensor out, bool replacement, double ratio,
 
 // ---------------------------------------------------------------------------
 // bin_features: [N, F] f32 + [F, B-1] edges -> [N, F] u8
+//   edges for one feature stay in registers? no — binary search in L1/L2.
+//   Thread covers one (row, feature); consecutive threads cover consecutive
+//   features of one row so the x reads coalesce.
 // ---------------------------------------------------------------------------
 
 __global__ void bin_features_kernel(uint8_t* __restrict__ out,
@@ -100,8 +104,7 @@ __global__ void bin_features_kernel(uint8_t* __restrict__ out,
     int fi = (int)(i % f);
     float v = x[i];
     const float* e = edges + (int64_t)fi * nedges;
-    // first index with e[idx] >= v  (torch.searchsorted right=False)
-    int lo = 0, hi = nedges;
+    int lo = 0, hi = nedges;  // first index with e[idx] >= v
     while (lo < hi) {
       int mid = (lo + hi) >> 1;
       if (e[mid] >= v) hi = mid; else lo = mid + 1;
@@ -127,7 +130,9 @@ void bin_features(torch::Tensor out, torch::Tensor x, torch::Tensor edges) {
 // ---------------------------------------------------------------------------
 // hist_build
 //   grid = (row_chunks, feature_groups); LDS histogram [FG][B][C] f32.
-//   chunks: int32 [n_chunks, 3] = (node, start, len) segments of row_idx.
+//   chunks: int32 [n_chunks, 4] = (node, start, len, single_chunk_flag).
+//   Chunk sizes are host-chosen so the grid ~ fills the chip 2-3x over;
+//   a node covered by ONE chunk flushes with plain stores (no atomics).
 // ---------------------------------------------------------------------------
 
 __global__ void hist_build_kernel(
@@ -135,40 +140,63 @@ __global__ void hist_build_kernel(
     const uint8_t* __restrict__ bins,   // [N, F]
     const float* __restrict__ gh,       // [N, C]
     const int* __restrict__ row_idx,    // [M]
-    const int* __restrict__ chunks,     // [n_chunks, 3]
+    const int* __restrict__ chunks,     // [n_chunks, 4]
     int F, int B, int C, int FG) {
   extern __shared__ float lds[];  // FG * B * C
   const int chunk = blockIdx.x;
   const int fg = blockIdx.y;
   const int f0 = fg * FG;
   const int nf = min(FG, F - f0);
-  const int node = chunks[chunk * 3 + 0];
-  const int start = chunks[chunk * 3 + 1];
-  const int len = chunks[chunk * 3 + 2];
+  const int node = chunks[chunk * 4 + 0];
+  const int start = chunks[chunk * 4 + 1];
+  const int len = chunks[chunk * 4 + 2];
+  const int single = chunks[chunk * 4 + 3];
 
   const int lds_size = FG * B * C;
   for (int i = threadIdx.x; i < lds_size; i += blockDim.x) lds[i] = 0.0f;
   __syncthreads();
 
+  // bins rows are read 16 bytes at a time (uint4) when the group start is
+  // 16-aligned and inside the row
+  const bool vec16 = (FG == 16) && (f0 + 16 <= F) && ((F & 15) == 0);
+
   for (int i = threadIdx.x; i < len; i += blockDim.x) {
     const int r = row_idx[start + i];
     const float* g = gh + (int64_t)r * C;
-    const uint8_t* br = bins + (int64_t)r * F + f0;
-    for (int f = 0; f < nf; ++f) {
-      const int b = br[f];
-      float* cell = lds + ((f * B) + b) * C;
-      for (int c = 0; c < C; ++c) {
-        atomicAdd(cell + c, g[c]);
+    float gloc[8];
+    for (int c = 0; c < C; ++c) gloc[c] = g[c];
+    if (vec16) {
+      const uint4 bv = *reinterpret_cast<const uint4*>(
+          bins + (int64_t)r * F + f0);
+      const unsigned w[4] = {bv.x, bv.y, bv.z, bv.w};
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const int b = (w[q] >> (8 * j)) & 0xff;
+          float* cell = lds + (((q * 4 + j) * B) + b) * C;
+          for (int c = 0; c < C; ++c) atomicAdd(cell + c, gloc[c]);
+        }
+      }
+    } else {
+      const uint8_t* br = bins + (int64_t)r * F + f0;
+      for (int f = 0; f < nf; ++f) {
+        const int b = br[f];
+        float* cell = lds + ((f * B) + b) * C;
+        for (int c = 0; c < C; ++c) atomicAdd(cell + c, gloc[c]);
       }
     }
   }
   __syncthreads();
 
-  // flush LDS -> global (atomic: several chunks may hit one node)
   float* dst = out + (((int64_t)node * F + f0) * B) * C;
-  for (int i = threadIdx.x; i < nf * B * C; i += blockDim.x) {
-    float v = lds[i];
-    if (v != 0.0f) atomicAdd(dst + i, v);
+  if (single) {
+    for (int i = threadIdx.x; i < nf * B * C; i += blockDim.x) dst[i] = lds[i];
+  } else {
+    for (int i = threadIdx.x; i < nf * B * C; i += blockDim.x) {
+      float v = lds[i];
+      if (v != 0.0f) atomicAdd(dst + i, v);
+    }
   }
 }
 
@@ -181,26 +209,38 @@ void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
   const int F = (int)bins.size(1);
   const int B = (int)num_bins;
   const int C = (int)gh.size(1);
+  TORCH_CHECK(C <= 8, "gh channels capped at 8 (grad dims + hess + count)");
 
-  // feature-group size: keep LDS <= 48 KiB so >= 3 blocks/CU stay resident
+  // feature-group size: prefer 16 (vectorized 16-B bin reads); keep
+  // LDS <= 48 KiB so 3 blocks/CU stay resident
   int FG = std::max<int>(1, std::min<int>(F, 49152 / (B * C * 4)));
+  if (FG >= 16) FG = 16;
   const int n_groups = (int)ceil_div(F, FG);
 
-  // chunk table on host
-  const int64_t CHUNK = 16384;
-  auto offs = node_offsets.accessor<int64_t, 1>();
-  std::vector<int> chunk_v;
+  // ---- adaptive chunking: target ~resident-grid x OVERSUB blocks --------
   const int n_nodes = (int)node_offsets.numel() - 1;
+  auto offs = node_offsets.accessor<int64_t, 1>();
+  int64_t total_rows = 0;
+  for (int nd = 0; nd < n_nodes; ++nd) total_rows += offs[nd + 1] - offs[nd];
+  const int64_t resident = 256 * 3;  // CUs x blocks/CU at 48 KiB LDS
+  const int64_t target_chunks =
+      std::max<int64_t>(1, (resident * 3) / std::max(1, n_groups));
+  int64_t chunk_rows = std::max<int64_t>(
+      4096, ceil_div(total_rows, target_chunks));
+
+  std::vector<int> chunk_v;
   for (int nd = 0; nd < n_nodes; ++nd) {
     int64_t s = offs[nd], e = offs[nd + 1];
-    for (int64_t c = s; c < e; c += CHUNK) {
+    const int single = (e - s) <= chunk_rows ? 1 : 0;
+    for (int64_t c = s; c < e; c += chunk_rows) {
       chunk_v.push_back(nd);
       chunk_v.push_back((int)c);
-      chunk_v.push_back((int)std::min<int64_t>(CHUNK, e - c));
+      chunk_v.push_back((int)std::min<int64_t>(chunk_rows, e - c));
+      chunk_v.push_back(single);
     }
   }
   if (chunk_v.empty()) return;
-  const int n_chunks = (int)(chunk_v.size() / 3);
+  const int n_chunks = (int)(chunk_v.size() / 4);
   auto chunks = torch::from_blob(chunk_v.data(), {(int64_t)chunk_v.size()},
                                  torch::kInt32)
                     .to(bins.device(), /*non_blocking=*/false);
@@ -215,7 +255,9 @@ void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
 }
 
 // ---------------------------------------------------------------------------
-// partition_rows: single pass, two-ended (left fills up, right fills down)
+// partition_rows: two passes per chunk.
+//   pass 1: block counts its left rows, ONE global atomic per block per side
+//   pass 2: replay, scatter via per-wave ballot + LDS block cursors
 // ---------------------------------------------------------------------------
 
 __global__ void partition_kernel(
@@ -227,22 +269,64 @@ __global__ void partition_kernel(
     const int* __restrict__ feat,      // [n_nodes]
     const int* __restrict__ thr,       // [n_nodes]
     int F) {
+  __shared__ int base_l, base_r, loc_l, loc_r;
   const int chunk = blockIdx.x;
   const int node = chunks[chunk * 3 + 0];
   const int start = chunks[chunk * 3 + 1];
   const int len = chunks[chunk * 3 + 2];
   const int f = feat[node];
   const int t = thr[node];
+
+  // pass 1: count left
+  int my_l = 0;
   for (int i = threadIdx.x; i < len; i += blockDim.x) {
     const int r = row_idx[start + i];
-    bool left = (f < 0) || (bins[(int64_t)r * F + f] <= t);
-    int pos;
-    if (left) {
-      pos = atomicAdd(cursors + node * 2 + 0, 1);
-    } else {
-      pos = atomicAdd(cursors + node * 2 + 1, -1) - 1;
+    my_l += (f < 0) || (bins[(int64_t)r * F + f] <= t) ? 1 : 0;
+  }
+  // wave then block reduction
+  for (int off = 32; off > 0; off >>= 1) my_l += __shfl_down(my_l, off, 64);
+  __shared__ int wl[8];
+  const int wave = threadIdx.x / 64;
+  if ((threadIdx.x & 63) == 0) wl[wave] = my_l;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    int tot = 0;
+    for (int w0 = 0; w0 < (int)(blockDim.x / 64); ++w0) tot += wl[w0];
+    base_l = atomicAdd(cursors + node * 2 + 0, tot);
+    base_r = atomicAdd(cursors + node * 2 + 1, -(len - tot)) - (len - tot);
+    loc_l = 0;
+    loc_r = 0;
+  }
+  __syncthreads();
+
+  // pass 2: scatter (per-wave ballot + LDS cursors)
+  for (int i = threadIdx.x; i < ((len + 63) & ~63); i += blockDim.x) {
+    const bool active = i < len;
+    int r = 0;
+    bool left = false;
+    if (active) {
+      r = row_idx[start + i];
+      left = (f < 0) || (bins[(int64_t)r * F + f] <= t);
     }
-    new_rows[pos] = r;
+    const unsigned long long lmask = __ballot(active && left);
+    const unsigned long long rmask = __ballot(active && !left);
+    const int lane = threadIdx.x & 63;
+    int lbase = 0, rbase = 0;
+    if (lane == 0) {
+      lbase = atomicAdd(&loc_l, (int)__popcll(lmask));
+      rbase = atomicAdd(&loc_r, (int)__popcll(rmask));
+    }
+    lbase = __shfl(lbase, 0, 64);
+    rbase = __shfl(rbase, 0, 64);
+    if (active) {
+      if (left) {
+        const int pos = base_l + lbase + (int)__popcll(lmask & ((1ull << lane) - 1ull));
+        new_rows[pos] = r;
+      } else {
+        const int pos = base_r + rbase + (int)__popcll(rmask & ((1ull << lane) - 1ull));
+        new_rows[pos] = r;
+      }
+    }
   }
 }
 
@@ -258,14 +342,16 @@ void partition_rows(torch::Tensor new_rows, torch::Tensor left_counts,
 
   std::vector<int> cur_v(n_nodes * 2);
   std::vector<int> chunk_v;
-  const int64_t CHUNK = 16384;
+  int64_t total_rows = offs[n_nodes];
+  int64_t chunk_rows =
+      std::max<int64_t>(8192, ceil_div(total_rows, (int64_t)2048));
   for (int nd = 0; nd < n_nodes; ++nd) {
     cur_v[nd * 2 + 0] = (int)offs[nd];
     cur_v[nd * 2 + 1] = (int)offs[nd + 1];
-    for (int64_t c = offs[nd]; c < offs[nd + 1]; c += CHUNK) {
+    for (int64_t c = offs[nd]; c < offs[nd + 1]; c += chunk_rows) {
       chunk_v.push_back(nd);
       chunk_v.push_back((int)c);
-      chunk_v.push_back((int)std::min<int64_t>(CHUNK, offs[nd + 1] - c));
+      chunk_v.push_back((int)std::min<int64_t>(chunk_rows, offs[nd + 1] - c));
     }
   }
   auto stream = at::hip::getCurrentHIPStream();
@@ -282,7 +368,6 @@ void partition_rows(torch::Tensor new_rows, torch::Tensor left_counts,
                        chunks.data_ptr<int>(), feat.data_ptr<int>(),
                        thr.data_ptr<int>(), F);
   }
-  // left_counts[nd] = final lcur - seg_start
   auto lcur = cursors.view({n_nodes, 2}).select(1, 0);
   auto seg_start =
       torch::from_blob(cur_v.data(), {n_nodes, 2}, torch::kInt32)
@@ -380,11 +465,241 @@ void forest_predict(torch::Tensor out, torch::Tensor x, torch::Tensor feat,
                      (int)D, T);
 }
 
+// ---------------------------------------------------------------------------
+// fused GBM loss kernels
+//   loss ids match spark_ensemble_amd/boosting/losses.py LOSS_IDS
+// ---------------------------------------------------------------------------
+
+#define L_SQUARED 0
+#define L_ABSOLUTE 1
+#define L_LOGCOSH 2
+#define L_SCALEDLOGCOSH 3
+#define L_HUBER 4
+#define L_QUANTILE 5
+#define L_LOGLOSS 6
+#define L_EXPONENTIAL 7
+#define L_BERNOULLI 8
+
+__device__ inline float logcosh_f(float d) {
+  float a = fabsf(d);
+  return a + log1pf(__expf(-2.0f * a)) - 0.6931471805599453f;
+}
+
+// scalar (dim=1) losses: returns loss, writes d(loss)/d(pred) and hessian
+__device__ inline float scalar_loss_grad(int loss_id, float param, float y,
+                                         float p, float* grad, float* hess) {
+  switch (loss_id) {
+    case L_SQUARED: {
+      float d = y - p;
+      *grad = -d;
+      *hess = 1.0f;
+      return 0.5f * d * d;
+    }
+    case L_ABSOLUTE: {
+      float d = y - p;
+      *grad = -copysignf(1.0f, d);
+      *hess = 0.0f;
+      return fabsf(d);
+    }
+    case L_LOGCOSH: {
+      float d = y - p;
+      float t = tanhf(d);
+      *grad = -t;
+      *hess = 1.0f - t * t;
+      return logcosh_f(d);
+    }
+    case L_SCALEDLOGCOSH: {
+      float d = y - p;
+      float s = (y > p) ? param : 1.0f - param;
+      float t = tanhf(d);
+      *grad = -s * t;
+      *hess = s * (1.0f - t * t);
+      return s * logcosh_f(d);
+    }
+    case L_HUBER: {
+      float d = y - p;
+      float ad = fabsf(d);
+      if (ad <= param) {
+        *grad = -d;
+        *hess = 1.0f;
+        return 0.5f * d * d;
+      }
+      *grad = -param * copysignf(1.0f, d);
+      *hess = 0.0f;
+      return param * (ad - 0.5f * param);
+    }
+    case L_QUANTILE: {
+      float d = y - p;
+      *grad = (d > 0.0f) ? -param : (1.0f - param);
+      *hess = 0.0f;
+      return (d > 0.0f) ? param * d : (param - 1.0f) * d;
+    }
+    case L_EXPONENTIAL: {
+      // y in {-1, 1}
+      float e = __expf(-y * p);
+      *grad = -y * e;
+      *hess = e;  // y^2 = 1
+      return e;
+    }
+    case L_BERNOULLI: {
+      float z = 2.0f * y * p;
+      float e = __expf(z);
+      *grad = -2.0f * y / (1.0f + e);
+      float ip = 1.0f + e;
+      *hess = 4.0f * e / (ip * ip);
+      float az = fabsf(z);
+      return fmaxf(-z, 0.0f) + log1pf(__expf(-az));
+    }
+  }
+  *grad = 0.0f;
+  *hess = 0.0f;
+  return 0.0f;
+}
+
+// grad_hess: per-row fused pseudo-residual inputs.
+//   dim == 1 scalar losses, or logloss with dim = K (softmax).
+__global__ void grad_hess_kernel(float* __restrict__ grad,  // [N, D]
+                                 float* __restrict__ hess,  // [N, D] or null
+                                 const float* __restrict__ label,  // [N, D]
+                                 const float* __restrict__ pred,   // [N, D]
+                                 int64_t n, int D, int loss_id, float param,
+                                 int want_hess) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; i < n; i += stride) {
+    if (loss_id == L_LOGLOSS) {
+      const float* pr = pred + i * D;
+      const float* lr = label + i * D;
+      float m = pr[0];
+      for (int d = 1; d < D; ++d) m = fmaxf(m, pr[d]);
+      float s = 0.0f;
+      for (int d = 0; d < D; ++d) s += __expf(pr[d] - m);
+      float inv = 1.0f / s;
+      for (int d = 0; d < D; ++d) {
+        float q = __expf(pr[d] - m) * inv;
+        grad[i * D + d] = q - lr[d];
+        if (want_hess) hess[i * D + d] = q * (1.0f - q);
+      }
+    } else {
+      float g, h;
+      scalar_loss_grad(loss_id, param, label[i], pred[i], &g, &h);
+      grad[i] = g;
+      if (want_hess) hess[i] = h;
+    }
+  }
+}
+
+void grad_hess(torch::Tensor grad, torch::Tensor hess, torch::Tensor label,
+               torch::Tensor pred, int64_t loss_id, double param,
+               bool want_hess) {
+  CHECK_GPU(grad); CHECK_GPU(label); CHECK_GPU(pred);
+  int64_t n = pred.size(0);
+  int D = (int)pred.size(1);
+  auto stream = at::hip::getCurrentHIPStream();
+  int blocks = (int)std::min<int64_t>(ceil_div(n, 256), 16384);
+  hipLaunchKernelGGL(grad_hess_kernel, dim3(blocks), dim3(256), 0, stream,
+                     grad.data_ptr<float>(),
+                     want_hess ? hess.data_ptr<float>() : nullptr,
+                     label.data_ptr<float>(), pred.data_ptr<float>(), n, D,
+                     (int)loss_id, (float)param, want_hess ? 1 : 0);
+}
+
+// line_search_eval: one pass computing
+//   payload[0]   = sum_i w_i * loss(y_i, p_i + a . d_i)
+//   payload[1+d] = sum_i w_i * d_id * dloss/dp_id
+__global__ void line_search_eval_kernel(
+    float* __restrict__ payload,        // [1 + D] (pre-zeroed)
+    const float* __restrict__ label,    // [N, D]
+    const float* __restrict__ pred,     // [N, D]
+    const float* __restrict__ dir,      // [N, D]
+    const float* __restrict__ weight,   // [N]
+    const float* __restrict__ coeff,    // [D]
+    int64_t n, int D, int loss_id, float param) {
+  extern __shared__ float acc[];  // blockDim.x warps... use [65 * (D+1)]? keep simple: [D+1] block acc
+  // block-level accumulators in LDS
+  for (int i = threadIdx.x; i < D + 1; i += blockDim.x) acc[i] = 0.0f;
+  __syncthreads();
+
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  float loss_sum = 0.0f;
+  float gsum[8];
+  for (int d = 0; d < D; ++d) gsum[d] = 0.0f;
+
+  for (; i < n; i += stride) {
+    const float w = weight[i];
+    if (loss_id == L_LOGLOSS) {
+      const float* pr = pred + i * D;
+      const float* dr = dir + i * D;
+      const float* lr = label + i * D;
+      float pv[8];
+      float m = -1e30f;
+      for (int d = 0; d < D; ++d) {
+        pv[d] = pr[d] + coeff[d] * dr[d];
+        m = fmaxf(m, pv[d]);
+      }
+      float s = 0.0f;
+      for (int d = 0; d < D; ++d) s += __expf(pv[d] - m);
+      float lse = m + __logf(s);
+      float l = 0.0f;
+      float inv = 1.0f / s;
+      for (int d = 0; d < D; ++d) {
+        l += -lr[d] * (pv[d] - lse);
+        float q = __expf(pv[d] - m) * inv;
+        gsum[d] += w * dr[d] * (q - lr[d]);
+      }
+      loss_sum += w * l;
+    } else {
+      float g, h;
+      float l = scalar_loss_grad(loss_id, param, label[i],
+                                 pred[i] + coeff[0] * dir[i], &g, &h);
+      if (!isfinite(l)) l = 3.0e38f;
+      loss_sum += w * l;
+      gsum[0] += w * dir[i] * g;
+    }
+  }
+
+  // wave reduce then LDS
+  for (int off = 32; off > 0; off >>= 1) {
+    loss_sum += __shfl_down(loss_sum, off, 64);
+    for (int d = 0; d < D; ++d) gsum[d] += __shfl_down(gsum[d], off, 64);
+  }
+  if ((threadIdx.x & 63) == 0) {
+    atomicAdd(&acc[0], loss_sum);
+    for (int d = 0; d < D; ++d) atomicAdd(&acc[1 + d], gsum[d]);
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    atomicAdd(&payload[0], acc[0]);
+    for (int d = 0; d < D; ++d) atomicAdd(&payload[1 + d], acc[1 + d]);
+  }
+}
+
+void line_search_eval(torch::Tensor payload, torch::Tensor label,
+                      torch::Tensor pred, torch::Tensor dir,
+                      torch::Tensor weight, torch::Tensor coeff,
+                      int64_t loss_id, double param) {
+  CHECK_GPU(payload); CHECK_GPU(pred);
+  int64_t n = pred.size(0);
+  int D = (int)pred.size(1);
+  TORCH_CHECK(D <= 8, "line_search_eval supports dim <= 8");
+  auto stream = at::hip::getCurrentHIPStream();
+  int blocks = (int)std::min<int64_t>(ceil_div(n, 256 * 8), 2048);
+  hipLaunchKernelGGL(line_search_eval_kernel, dim3(blocks), dim3(256),
+                     (D + 1) * 4, stream, payload.data_ptr<float>(),
+                     label.data_ptr<float>(), pred.data_ptr<float>(),
+                     dir.data_ptr<float>(), weight.data_ptr<float>(),
+                     coeff.data_ptr<float>(), n, D, (int)loss_id,
+                     (float)param);
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sample_weights", &sample_weights, "counter-based Poisson/Bernoulli row weights");
   m.def("bin_features", &bin_features, "quantile binning f32 -> u8");
   m.def("hist_build", &hist_build, "LDS-staged node histograms");
-  m.def("partition_rows", &partition_rows, "two-ended node partition");
+  m.def("partition_rows", &partition_rows, "block-aggregated node partition");
   m.def("tree_predict", &tree_predict, "single-tree batched predict");
   m.def("forest_predict", &forest_predict, "packed-forest weighted predict");
+  m.def("grad_hess", &grad_hess, "fused per-row loss gradient/hessian");
+  m.def("line_search_eval", &line_search_eval, "fused loss+grad line-search payload");
 }

@@ -26,7 +26,25 @@ from .losses import GBMLoss
 
 def _eval(loss, label, pred, direction, weight, coeff):
     """Returns (weighted loss sum, per-dim gradient sums) as a flat tensor
-    [1 + dim] on the data's device."""
+    [1 + dim] on the data's device.  On GPU this is ONE fused kernel pass
+    (csrc line_search_eval) instead of ~6 eager passes."""
+    if pred.is_cuda:
+        from ..ops import dispatch
+
+        m = dispatch._require_hip("line_search_eval")
+        if m is not None:
+            D = pred.shape[1]
+            if isinstance(coeff, torch.Tensor):
+                ct = coeff.reshape(-1).float().to(pred.device)
+            else:
+                ct = torch.full((D,), float(coeff), device=pred.device)
+            payload = torch.zeros(1 + D, dtype=torch.float32, device=pred.device)
+            m.line_search_eval(
+                payload, label.contiguous(), pred.contiguous(),
+                direction.contiguous(), weight.contiguous(), ct,
+                loss.loss_id, float(loss.param),
+            )
+            return payload
     p = pred + direction * coeff
     l = (loss.loss(label, p) * weight).sum()
     g = loss.gradient(label, p) * direction * weight.unsqueeze(1)

@@ -66,6 +66,25 @@ class GBMLoss:
     def loss_id(self) -> int:
         return LOSS_IDS[self.name]
 
+    def grad_hess_fused(self, label, pred, want_hess: bool = False):
+        """(gradient [N,D], hessian [N,D] or None) in ONE fused kernel pass
+        on GPU (csrc grad_hess), torch ops otherwise."""
+        if pred.is_cuda:
+            from ..ops import dispatch
+
+            m = dispatch._require_hip("grad_hess")
+            if m is not None:
+                grad = torch.empty_like(pred)
+                hess = torch.empty_like(pred) if want_hess else grad
+                m.grad_hess(
+                    grad, hess, label.contiguous(), pred.contiguous(),
+                    self.loss_id, float(self.param), bool(want_hess),
+                )
+                return grad, (hess if want_hess else None)
+        g = self.gradient(label, pred)
+        h = self.hessian(label, pred) if want_hess else None
+        return g, h
+
 
 class _ClassificationLoss(GBMLoss):
     def raw2probability(self, raw: torch.Tensor) -> torch.Tensor:

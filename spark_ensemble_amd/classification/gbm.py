@@ -165,15 +165,16 @@ class GBMClassifier(ProbabilisticClassifier, _GBMClassifierParams):
             )
             bag_w = bag_m * w
 
-            if use_newton and loss.has_hessian:
-                h = loss.hessian(ylab, predictions).clamp_min(1e-2)  # [N, dim]
+            newton = use_newton and loss.has_hessian
+            g, h = loss.grad_hess_fused(ylab, predictions, want_hess=newton)
+            if newton:
+                h = h.clamp_min(1e-2)  # [N, dim]
                 sum_h = (h * bag_m.unsqueeze(1)).sum(dim=0)
                 comm.all_reduce_(sum_h)
-                neg_grad = -loss.gradient(ylab, predictions)
-                res_label = neg_grad / h  # [N, dim]
+                res_label = -g / h  # [N, dim]
                 res_weight = 0.5 * h / sum_h.unsqueeze(0) * bag_w.unsqueeze(1)
             else:
-                res_label = -loss.gradient(ylab, predictions)
+                res_label = -g
                 res_weight = bag_w.unsqueeze(1).expand(-1, dim)
 
             # K per-class base-regressor fits (reference parallel Futures

@@ -189,14 +189,15 @@ class GBMRegressor(Regressor, _GBMRegressorParams):
 
             ylab = y.unsqueeze(1)
             pred2 = predictions.unsqueeze(1)
-            if use_newton and loss.has_hessian:
-                h = loss.hessian(ylab, pred2).squeeze(1).clamp_min(1e-2)
+            newton = use_newton and loss.has_hessian
+            g, h = loss.grad_hess_fused(ylab, pred2, want_hess=newton)
+            if newton:
+                h = h.squeeze(1).clamp_min(1e-2)
                 sum_h = comm.all_reduce_scalar(float((h * bag_m).sum()))
-                neg_grad = -loss.gradient(ylab, pred2).squeeze(1)
-                res_label = neg_grad / h
+                res_label = -g.squeeze(1) / h
                 res_weight = 0.5 * h / sum_h * bag_w
             else:
-                res_label = -loss.gradient(ylab, pred2).squeeze(1)
+                res_label = -g.squeeze(1)
                 res_weight = bag_w
 
             fr = binned.fit_frame(learner, res_label, res_weight, idx, xs)

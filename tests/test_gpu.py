@@ -209,3 +209,53 @@ def test_gpu_losses_match_cpu():
     grad_gpu = loss.gradient(lab.to(DEV), pred.to(DEV)).cpu()
     assert torch.allclose(l_cpu, l_gpu, atol=1e-5)
     assert torch.allclose(grad_cpu, grad_gpu, atol=1e-5)
+
+
+def test_grad_hess_kernel_matches_torch(hip):
+    from spark_ensemble_amd.boosting.losses import (
+        BernoulliLoss, ExponentialLoss, HuberLoss, LogCoshLoss, LogLoss,
+        QuantileLoss, ScaledLogCoshLoss, SquaredLoss, AbsoluteLoss,
+    )
+
+    g = torch.Generator().manual_seed(11)
+    n = 20000
+    for loss in [SquaredLoss(), AbsoluteLoss(), LogCoshLoss(),
+                 ScaledLogCoshLoss(0.3), HuberLoss(0.8), QuantileLoss(0.7),
+                 ExponentialLoss(), BernoulliLoss(), LogLoss(4)]:
+        if loss.name == "logloss":
+            y = torch.randint(0, 4, (n,), generator=g).float()
+        elif loss.name in ("exponential", "bernoulli"):
+            y = torch.randint(0, 2, (n,), generator=g).float()
+        else:
+            y = torch.randn(n, generator=g)
+        lab = loss.encode_label(y)
+        pred = torch.randn(n, loss.dim, generator=g)
+        want_g = loss.gradient(lab, pred)
+        got_g, got_h = loss.grad_hess_fused(
+            lab.to(DEV), pred.to(DEV), want_hess=loss.has_hessian
+        )
+        assert torch.allclose(got_g.cpu(), want_g, atol=1e-4), loss.name
+        if loss.has_hessian:
+            want_h = loss.hessian(lab, pred)
+            assert torch.allclose(got_h.cpu(), want_h, atol=1e-4), loss.name
+
+
+def test_line_search_eval_matches_torch(hip):
+    from spark_ensemble_amd.boosting.line_search import _eval
+    from spark_ensemble_amd.boosting.losses import BernoulliLoss, LogLoss
+
+    g = torch.Generator().manual_seed(12)
+    n = 40000
+    for loss, D in [(BernoulliLoss(), 1), (LogLoss(3), 3)]:
+        y = torch.randint(0, max(2, D), (n,), generator=g).float()
+        lab = loss.encode_label(y)
+        pred = torch.randn(n, D, generator=g)
+        dr = torch.randn(n, D, generator=g)
+        w = torch.rand(n, generator=g)
+        coeff = torch.rand(D, generator=g) if D > 1 else 0.7
+        want = _eval(loss, lab, pred, dr, w, coeff)
+        got = _eval(
+            loss, lab.to(DEV), pred.to(DEV), dr.to(DEV), w.to(DEV),
+            coeff.to(DEV) if isinstance(coeff, torch.Tensor) else coeff,
+        ).cpu()
+        assert torch.allclose(got, want, rtol=2e-3, atol=2e-2), (loss.name, got, want)
