@@ -135,13 +135,14 @@ void bin_features(torch::Tensor out, torch::Tensor x, torch::Tensor edges) {
 //   a node covered by ONE chunk flushes with plain stores (no atomics).
 // ---------------------------------------------------------------------------
 
+template <int C>
 __global__ void hist_build_kernel(
     float* __restrict__ out,            // [n_nodes, F, B, C]
     const uint8_t* __restrict__ bins,   // [N, F]
     const float* __restrict__ gh,       // [N, C]
     const int* __restrict__ row_idx,    // [M]
     const int* __restrict__ chunks,     // [n_chunks, 4]
-    int F, int B, int C, int FG) {
+    int F, int B, int FG) {
   extern __shared__ float lds[];  // FG * B * C
   const int chunk = blockIdx.x;
   const int fg = blockIdx.y;
@@ -163,7 +164,8 @@ __global__ void hist_build_kernel(
   for (int i = threadIdx.x; i < len; i += blockDim.x) {
     const int r = row_idx[start + i];
     const float* g = gh + (int64_t)r * C;
-    float gloc[8];
+    float gloc[C];
+#pragma unroll
     for (int c = 0; c < C; ++c) gloc[c] = g[c];
     if (vec16) {
       const uint4 bv = *reinterpret_cast<const uint4*>(
@@ -175,6 +177,7 @@ __global__ void hist_build_kernel(
         for (int j = 0; j < 4; ++j) {
           const int b = (w[q] >> (8 * j)) & 0xff;
           float* cell = lds + (((q * 4 + j) * B) + b) * C;
+#pragma unroll
           for (int c = 0; c < C; ++c) atomicAdd(cell + c, gloc[c]);
         }
       }
@@ -183,6 +186,7 @@ __global__ void hist_build_kernel(
       for (int f = 0; f < nf; ++f) {
         const int b = br[f];
         float* cell = lds + ((f * B) + b) * C;
+#pragma unroll
         for (int c = 0; c < C; ++c) atomicAdd(cell + c, gloc[c]);
       }
     }
@@ -247,11 +251,23 @@ void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
 
   auto stream = at::hip::getCurrentHIPStream();
   const size_t lds_bytes = (size_t)FG * B * C * 4;
-  hipLaunchKernelGGL(hist_build_kernel, dim3(n_chunks, n_groups), dim3(256),
-                     lds_bytes, stream, out.data_ptr<float>(),
-                     bins.data_ptr<uint8_t>(), gh.data_ptr<float>(),
-                     row_idx.data_ptr<int>(), chunks.data_ptr<int>(), F, B, C,
-                     FG);
+#define HB_LAUNCH(CC)                                                        \
+  hipLaunchKernelGGL(hist_build_kernel<CC>, dim3(n_chunks, n_groups),        \
+                     dim3(256), lds_bytes, stream, out.data_ptr<float>(),    \
+                     bins.data_ptr<uint8_t>(), gh.data_ptr<float>(),         \
+                     row_idx.data_ptr<int>(), chunks.data_ptr<int>(), F, B,  \
+                     FG)
+  switch (C) {
+    case 2: HB_LAUNCH(2); break;
+    case 3: HB_LAUNCH(3); break;
+    case 4: HB_LAUNCH(4); break;
+    case 5: HB_LAUNCH(5); break;
+    case 6: HB_LAUNCH(6); break;
+    case 7: HB_LAUNCH(7); break;
+    case 8: HB_LAUNCH(8); break;
+    default: TORCH_CHECK(false, "hist_build: unsupported channel count ", C);
+  }
+#undef HB_LAUNCH
 }
 
 // ---------------------------------------------------------------------------
@@ -568,6 +584,8 @@ __global__ void grad_hess_kernel(float* __restrict__ grad,  // [N, D]
   int64_t stride = gridDim.x * (int64_t)blockDim.x;
   for (; i < n; i += stride) {
     if (loss_id == L_LOGLOSS) {
+      // two passes over d; no local arrays (rule: runtime-indexed locals
+      // become register-select chains or scratch)
       const float* pr = pred + i * D;
       const float* lr = label + i * D;
       float m = pr[0];
@@ -607,6 +625,7 @@ void grad_hess(torch::Tensor grad, torch::Tensor hess, torch::Tensor label,
 // line_search_eval: one pass computing
 //   payload[0]   = sum_i w_i * loss(y_i, p_i + a . d_i)
 //   payload[1+d] = sum_i w_i * d_id * dloss/dp_id
+template <int D>
 __global__ void line_search_eval_kernel(
     float* __restrict__ payload,        // [1 + D] (pre-zeroed)
     const float* __restrict__ label,    // [N, D]
@@ -614,35 +633,42 @@ __global__ void line_search_eval_kernel(
     const float* __restrict__ dir,      // [N, D]
     const float* __restrict__ weight,   // [N]
     const float* __restrict__ coeff,    // [D]
-    int64_t n, int D, int loss_id, float param) {
-  extern __shared__ float acc[];  // blockDim.x warps... use [65 * (D+1)]? keep simple: [D+1] block acc
-  // block-level accumulators in LDS
+    int64_t n, int loss_id, float param) {
+  __shared__ float acc[1 + D];
   for (int i = threadIdx.x; i < D + 1; i += blockDim.x) acc[i] = 0.0f;
   __syncthreads();
+
+  float cf[D];
+#pragma unroll
+  for (int d = 0; d < D; ++d) cf[d] = coeff[d];
 
   int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
   int64_t stride = gridDim.x * (int64_t)blockDim.x;
   float loss_sum = 0.0f;
-  float gsum[8];
+  float gsum[D];
+#pragma unroll
   for (int d = 0; d < D; ++d) gsum[d] = 0.0f;
 
   for (; i < n; i += stride) {
     const float w = weight[i];
-    if (loss_id == L_LOGLOSS) {
+    if (D > 1 || loss_id == L_LOGLOSS) {
       const float* pr = pred + i * D;
       const float* dr = dir + i * D;
       const float* lr = label + i * D;
-      float pv[8];
+      float pv[D];
       float m = -1e30f;
+#pragma unroll
       for (int d = 0; d < D; ++d) {
-        pv[d] = pr[d] + coeff[d] * dr[d];
+        pv[d] = pr[d] + cf[d] * dr[d];
         m = fmaxf(m, pv[d]);
       }
       float s = 0.0f;
+#pragma unroll
       for (int d = 0; d < D; ++d) s += __expf(pv[d] - m);
       float lse = m + __logf(s);
       float l = 0.0f;
       float inv = 1.0f / s;
+#pragma unroll
       for (int d = 0; d < D; ++d) {
         l += -lr[d] * (pv[d] - lse);
         float q = __expf(pv[d] - m) * inv;
@@ -652,7 +678,7 @@ __global__ void line_search_eval_kernel(
     } else {
       float g, h;
       float l = scalar_loss_grad(loss_id, param, label[i],
-                                 pred[i] + coeff[0] * dir[i], &g, &h);
+                                 pred[i] + cf[0] * dir[i], &g, &h);
       if (!isfinite(l)) l = 3.0e38f;
       loss_sum += w * l;
       gsum[0] += w * dir[i] * g;
@@ -662,15 +688,18 @@ __global__ void line_search_eval_kernel(
   // wave reduce then LDS
   for (int off = 32; off > 0; off >>= 1) {
     loss_sum += __shfl_down(loss_sum, off, 64);
+#pragma unroll
     for (int d = 0; d < D; ++d) gsum[d] += __shfl_down(gsum[d], off, 64);
   }
   if ((threadIdx.x & 63) == 0) {
     atomicAdd(&acc[0], loss_sum);
+#pragma unroll
     for (int d = 0; d < D; ++d) atomicAdd(&acc[1 + d], gsum[d]);
   }
   __syncthreads();
   if (threadIdx.x == 0) {
     atomicAdd(&payload[0], acc[0]);
+#pragma unroll
     for (int d = 0; d < D; ++d) atomicAdd(&payload[1 + d], acc[1 + d]);
   }
 }
@@ -683,14 +712,27 @@ void line_search_eval(torch::Tensor payload, torch::Tensor label,
   int64_t n = pred.size(0);
   int D = (int)pred.size(1);
   TORCH_CHECK(D <= 8, "line_search_eval supports dim <= 8");
+  TORCH_CHECK(D == 1 || loss_id == L_LOGLOSS,
+              "vector line search only for logloss");
   auto stream = at::hip::getCurrentHIPStream();
   int blocks = (int)std::min<int64_t>(ceil_div(n, 256 * 8), 2048);
-  hipLaunchKernelGGL(line_search_eval_kernel, dim3(blocks), dim3(256),
-                     (D + 1) * 4, stream, payload.data_ptr<float>(),
-                     label.data_ptr<float>(), pred.data_ptr<float>(),
-                     dir.data_ptr<float>(), weight.data_ptr<float>(),
-                     coeff.data_ptr<float>(), n, D, (int)loss_id,
-                     (float)param);
+#define LS_LAUNCH(DD)                                                         \
+  hipLaunchKernelGGL(line_search_eval_kernel<DD>, dim3(blocks), dim3(256), 0, \
+                     stream, payload.data_ptr<float>(),                       \
+                     label.data_ptr<float>(), pred.data_ptr<float>(),         \
+                     dir.data_ptr<float>(), weight.data_ptr<float>(),         \
+                     coeff.data_ptr<float>(), n, (int)loss_id, (float)param)
+  switch (D) {
+    case 1: LS_LAUNCH(1); break;
+    case 2: LS_LAUNCH(2); break;
+    case 3: LS_LAUNCH(3); break;
+    case 4: LS_LAUNCH(4); break;
+    case 5: LS_LAUNCH(5); break;
+    case 6: LS_LAUNCH(6); break;
+    case 7: LS_LAUNCH(7); break;
+    case 8: LS_LAUNCH(8); break;
+  }
+#undef LS_LAUNCH
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {

@@ -75,6 +75,7 @@ def grow_tree(
     params: GrowParams,
     comm: Optional[Comm] = None,
     row_mask: Optional[torch.Tensor] = None,
+    hess_is_count: Optional[bool] = None,
 ) -> Dict[str, torch.Tensor]:
     """Grow one tree; returns flat node arrays:
 
@@ -84,11 +85,23 @@ def grow_tree(
     device = bins.device
     N, F = bins.shape
     D = grad.shape[1]
-    C = D + 2
     B = params.max_bins
 
-    cnt = torch.ones(N, dtype=torch.float32, device=device)
-    gh = torch.cat([grad, hess.unsqueeze(1), cnt.unsqueeze(1)], dim=1).contiguous()
+    # Count channel is only materialized when hess cannot serve as the row
+    # count (non-unit weights); with unit hessians (the default gini /
+    # gradient-mode path) C = D + 1 and the hess channel doubles as count —
+    # one fewer LDS atomic per cell in the histogram kernel.
+    if hess_is_count is None:
+        hess_is_count = bool((hess == 1).all())
+    if hess_is_count:
+        gh = torch.cat([grad, hess.unsqueeze(1)], dim=1).contiguous()
+    else:
+        cnt = torch.ones(N, dtype=torch.float32, device=device)
+        gh = torch.cat(
+            [grad, hess.unsqueeze(1), cnt.unsqueeze(1)], dim=1
+        ).contiguous()
+    C = gh.shape[1]
+    idx_c = C - 1
 
     if row_mask is not None:
         row_idx = row_mask.nonzero(as_tuple=True)[0].to(torch.int32)
@@ -185,6 +198,7 @@ def grow_tree(
             params.min_child_weight,
             params.min_instances_per_node,
             params.min_info_gain,
+            d_dims=D,
         )
         gain_cpu = gain.cpu()
         feat_cpu = feat.cpu()
@@ -240,8 +254,8 @@ def grow_tree(
             lcid, rcid = child_ids[i]
             l_stats = left_stats[i]
             r_stats = totals[i] - l_stats
-            l_cnt = float(l_stats[D + 1])
-            r_cnt = float(r_stats[D + 1])
+            l_cnt = float(l_stats[idx_c])
+            r_cnt = float(r_stats[idx_c])
             ls, le, re = offs_list[2 * i], offs_list[2 * i + 1], offs_list[2 * i + 2]
             keep_segs.append((ls, re))
             for (cid, s, e, st, built) in (

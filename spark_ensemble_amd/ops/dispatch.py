@@ -104,8 +104,8 @@ def hist_build(bins, gh, row_idx, node_offsets, num_bins):
     return reference.hist_build(bins, gh, row_idx, node_offsets, num_bins)
 
 
-def split_search(hist, lam=1e-6, min_child_weight=0.0, min_instances=1.0, min_info_gain=0.0):
-    return reference.split_search(hist, lam, min_child_weight, min_instances, min_info_gain)
+def split_search(hist, lam=1e-6, min_child_weight=0.0, min_instances=1.0, min_info_gain=0.0, d_dims=-1):
+    return reference.split_search(hist, lam, min_child_weight, min_instances, min_info_gain, d_dims)
 
 
 def partition_rows(bins, row_idx, node_offsets, feat, thr):

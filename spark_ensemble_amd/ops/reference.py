@@ -153,6 +153,7 @@ def split_search(
     min_child_weight: float = 0.0,
     min_instances: float = 1.0,
     min_info_gain: float = 0.0,
+    d_dims: int = -1,
 ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor, torch.Tensor]:
     """Best variance-reduction / newton-gain split per node.
 
@@ -165,7 +166,8 @@ def split_search(
     gain = -inf where no valid split exists.
     """
     n, F, B, C = hist.shape
-    D = C - 2
+    D = d_dims if d_dims > 0 else C - 2
+    idx_c = C - 1  # count channel; == hess channel when C == D + 1
     total = hist.sum(dim=2)  # [n, F, B, C] -> [n, F, C]
     parent = total[:, 0, :]  # same for every feature: [n, C]
     cum = hist.cumsum(dim=2)  # left stats if split at bin b (x <= edge_b)
@@ -180,8 +182,8 @@ def split_search(
     gain = score(left) + score(right) - score(parent)[:, None, None]
     hl = left[..., D]
     hr = right[..., D]
-    cl = left[..., D + 1]
-    cr = right[..., D + 1]
+    cl = left[..., idx_c]
+    cr = right[..., idx_c]
     valid = (
         (hl >= min_child_weight)
         & (hr >= min_child_weight)
