@@ -73,6 +73,10 @@ class BoostingClassifier(ProbabilisticClassifier, _BoostingClassifierParams):
         return DecisionTreeClassifier()
 
     def _fit(self, dataset: TensorFrame) -> "BoostingClassificationModel":
+        from ..utils.instrumentation import Instrumentation
+
+        instr = Instrumentation(self, dataset)
+        instr.log_params(self, "algorithm", "numBaseLearners")
         comm = get_comm()
         learner = self.getOrNone("baseLearner") or self._default_base_learner()
         k = self.getNumBaseLearners()
@@ -134,8 +138,12 @@ class BoostingClassifier(ProbabilisticClassifier, _BoostingClassifierParams):
                     torch.tensor(inv_beta, device=x.device), err01
                 )
             sum_w = comm.all_reduce_scalar(float(boosting_w.sum()))
+            instr.log_round(i, error=est_err,
+                            weight=est_weights[-1] if est_weights else 0.0,
+                            sum_w=sum_w)
             i += 1
 
+        instr.finish()
         model = BoostingClassificationModel()
         model._models = models
         model._weights = est_weights

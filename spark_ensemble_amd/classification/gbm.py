@@ -88,6 +88,11 @@ class GBMClassifier(ProbabilisticClassifier, _GBMClassifierParams):
         return DecisionTreeRegressor()
 
     def _fit(self, dataset: TensorFrame) -> "GBMClassificationModel":
+        from ..utils.instrumentation import Instrumentation
+
+        instr = Instrumentation(self, dataset)
+        instr.log_params(self, "loss", "numBaseLearners", "updates",
+                         "learningRate", "optimizedWeights")
         comm = get_comm()
         learner = self.getOrNone("baseLearner") or self._default_base_learner()
         seed = self.getOrDefault("seed")
@@ -225,9 +230,14 @@ class GBMClassifier(ProbabilisticClassifier, _GBMClassifierParams):
                 elif err < best_err:
                     best_err = err
                     v = 0
+                instr.log_round(i, weight=float(iweights[0]), val_loss=err,
+                                patience=v)
+            else:
+                instr.log_round(i, weight=float(iweights[0]))
             i += 1
 
         keep = i - v
+        instr.finish()
         model = GBMClassificationModel()
         model._init = init
         model._models = models[:keep]

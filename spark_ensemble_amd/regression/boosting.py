@@ -87,6 +87,10 @@ class BoostingRegressor(Regressor, _BoostingRegressorParams):
         return DecisionTreeRegressor()
 
     def _fit(self, dataset: TensorFrame) -> "BoostingRegressionModel":
+        from ..utils.instrumentation import Instrumentation
+
+        instr = Instrumentation(self, dataset)
+        instr.log_params(self, "lossType", "numBaseLearners", "votingStrategy")
         comm = get_comm()
         learner = self.getOrNone("baseLearner") or self._default_base_learner()
         k = self.getNumBaseLearners()
@@ -129,8 +133,10 @@ class BoostingRegressor(Regressor, _BoostingRegressorParams):
 
             models.append(model)
             est_weights.append(est_weight)
+            instr.log_round(i, error=est_err, weight=est_weight, sum_w=sum_w)
             i += 1
 
+        instr.finish()
         model = BoostingRegressionModel()
         model._models = models
         model._weights = est_weights

@@ -98,6 +98,11 @@ class GBMRegressor(Regressor, _GBMRegressorParams):
         return DecisionTreeRegressor()
 
     def _fit(self, dataset: TensorFrame) -> "GBMRegressionModel":
+        from ..utils.instrumentation import Instrumentation
+
+        instr = Instrumentation(self, dataset)
+        instr.log_params(self, "loss", "numBaseLearners", "updates",
+                         "learningRate", "optimizedWeights")
         comm = get_comm()
         learner = (
             self.getOrNone("baseLearner") or self._default_base_learner()
@@ -235,7 +240,11 @@ class GBMRegressor(Regressor, _GBMRegressorParams):
                 elif err < best_err:
                     best_err = err
                     v = 0
+                instr.log_round(i, weight=weight, val_loss=err, patience=v)
+            else:
+                instr.log_round(i, weight=weight)
             i += 1
+        instr.finish()
 
         keep = i - v
         model = GBMRegressionModel()
