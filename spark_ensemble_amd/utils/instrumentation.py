@@ -60,6 +60,10 @@ class Instrumentation:
     def __init__(self, estimator, dataset=None):
         self.stage = type(estimator).__name__
         self.uid = getattr(estimator, "uid", "?")
+        try:
+            estimator._instr = self  # expose the live record on the estimator
+        except Exception:  # noqa: BLE001
+            pass
         self.history: List[Dict] = []
         self.timers: Dict[str, float] = {}
         self._t0 = time.perf_counter()
