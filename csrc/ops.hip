@@ -735,7 +735,15 @@ void line_search_eval(torch::Tensor payload, torch::Tensor label,
 #undef LS_LAUNCH
 }
 
+// csrc/linear.hip
+void logreg_loss_grad(torch::Tensor payload, torch::Tensor x, torch::Tensor y,
+                      torch::Tensor w, torch::Tensor wmat, bool has_bias);
+bool logreg_fused_supported(int64_t F, int64_t K);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("logreg_loss_grad", &logreg_loss_grad,
+        "single-pass fused logistic loss+gradient");
+  m.def("logreg_fused_supported", &logreg_fused_supported);
   m.def("sample_weights", &sample_weights, "counter-based Poisson/Bernoulli row weights");
   m.def("bin_features", &bin_features, "quantile binning f32 -> u8");
   m.def("hist_build", &hist_build, "LDS-staged node histograms");

@@ -155,33 +155,23 @@ class LogisticRegression(ProbabilisticClassifier, _LinearParams):
 
         fp = f + (1 if intercept else 0)
 
+        from ..ops import dispatch
+
+        y_int = yl.to(torch.int32)
+
         def eval_loss_grad(theta_np):
             theta = torch.from_numpy(theta_np.astype(np.float32)).to(x.device)
             wt = theta.view(fp, k)
-            wm = wt[:f]
-            b = wt[f] if intercept else None
-            raw = x @ wm
-            if b is not None:
-                raw = raw + b
-            logp = raw.log_softmax(dim=1)
-            nll = -(logp.gather(1, yl.unsqueeze(1)).squeeze(1) * w).sum()
-            p = logp.exp()
-            onehot = torch.zeros_like(p)
-            onehot.scatter_(1, yl.unsqueeze(1), 1.0)
-            gmat = (p - onehot) * w.unsqueeze(1)  # [N, K]
-            gw = x.T @ gmat  # [F, K]
-            if intercept:
-                gb = gmat.sum(dim=0, keepdim=True)
-                g = torch.cat([gw, gb], dim=0)
-            else:
-                g = gw
-            payload = torch.cat([nll.reshape(1), g.reshape(-1)])
+            # single fused pass over X: margins + softmax loss + grad
+            # outer-product (HIP kernel csrc/linear.hip on GPU)
+            payload = dispatch.logreg_loss_grad(x, y_int, w, wt, intercept)
             if comm.is_distributed:
                 comm.all_reduce_(payload)
             loss = float(payload[0]) / total_w
-            grad = (payload[1:] / total_w).reshape(fp, k)
+            grad_full = (payload[1:] / total_w).reshape(f + 1, k)
+            grad = grad_full[:fp] if intercept else grad_full[:f]
             if lam > 0:
-                reg_w = wm
+                reg_w = wt[:f]
                 loss += 0.5 * lam * float((reg_w * reg_w).sum())
                 grad[:f] += lam * reg_w
             return loss, grad.cpu().double().numpy().ravel()

@@ -188,3 +188,27 @@ def _forest_predict_hip(m, x, trees, weights):
     out = torch.zeros(x.shape[0], D, dtype=torch.float32, device=dev)
     m.forest_predict(out, x.contiguous(), feats, thrs, lefts, leaves.contiguous(), offsets, w, D)
     return out
+
+
+def logreg_loss_grad(x, y_int, w, wmat, has_bias):
+    """Single-pass fused logistic loss+gradient.
+
+    Returns payload [1 + (F+1)*K]: [loss_sum, grad(F,K) flat, grad_bias(K)]
+    (unnormalized sums; caller all-reduces and divides by total weight).
+    On GPU uses the hand-written gfx950 kernel (csrc/linear.hip) when the
+    (F, K) shape fits the register budget, else the torch fallback.
+    """
+    f = x.shape[1]
+    k = wmat.shape[1]
+    if x.is_cuda:
+        m = _require_hip("logreg_loss_grad")
+        if m is not None and m.logreg_fused_supported(f, k):
+            payload = torch.zeros(
+                1 + (f + 1) * k, dtype=torch.float32, device=x.device
+            )
+            m.logreg_loss_grad(
+                payload, x.contiguous(), y_int.contiguous(),
+                w.contiguous(), wmat.contiguous(), bool(has_bias),
+            )
+            return payload
+    return reference.logreg_loss_grad(x, y_int, w, wmat, has_bias)

@@ -298,3 +298,24 @@ def forest_predict(
             p = p * weights[i]
         out = p if out is None else out + p
     return out
+
+
+def logreg_loss_grad(x, y_int, w, wmat, has_bias):
+    """Torch reference for the fused logistic loss+grad payload
+    (dispatch.logreg_loss_grad contract)."""
+    f = x.shape[1]
+    k = wmat.shape[1]
+    wm = wmat[:f]
+    raw = x @ wm
+    if has_bias:
+        raw = raw + wmat[f]
+    logp = torch.log_softmax(raw, dim=1)
+    yl = y_int.long()
+    nll = -(logp.gather(1, yl.unsqueeze(1)).squeeze(1) * w).sum()
+    p = logp.exp()
+    onehot = torch.zeros_like(p)
+    onehot.scatter_(1, yl.unsqueeze(1), 1.0)
+    gmat = (p - onehot) * w.unsqueeze(1)
+    gw = x.T @ gmat
+    gb = gmat.sum(dim=0) if has_bias else torch.zeros(k, device=x.device)
+    return torch.cat([nll.reshape(1), gw.reshape(-1), gb.reshape(-1)])

@@ -259,3 +259,33 @@ def test_line_search_eval_matches_torch(hip):
             coeff.to(DEV) if isinstance(coeff, torch.Tensor) else coeff,
         ).cpu()
         assert torch.allclose(got, want, rtol=2e-3, atol=2e-2), (loss.name, got, want)
+
+
+@pytest.mark.parametrize("f,k,bias", [(256, 2, True), (1024, 2, True),
+                                      (100, 3, False), (512, 4, True)])
+def test_logreg_loss_grad_matches_reference(hip, ref, f, k, bias):
+    g = torch.Generator().manual_seed(11)
+    n = 20000
+    x = torch.randn(n, f, generator=g).to(DEV)
+    y = torch.randint(0, k, (n,), generator=g, dtype=torch.int32).to(DEV)
+    w = torch.rand(n, generator=g).to(DEV)
+    fp = f + (1 if bias else 0)
+    wmat = (torch.randn(fp, k, generator=g) * 0.05).to(DEV)
+    got = hip.logreg_loss_grad(x, y, w, wmat, bias)
+    want = ref.logreg_loss_grad(x, y, w, wmat, bias)
+    # loss and gradient sums over 20k rows: fp32 atomics, tolerate 1e-3 rel
+    assert torch.allclose(got[0], want[0], rtol=1e-3, atol=1e-2)
+    assert torch.allclose(got[1:], want[1:], rtol=1e-3, atol=5e-2), (
+        (got[1:] - want[1:]).abs().max()
+    )
+
+
+def test_logistic_regression_gpu_fit():
+    import spark_ensemble_amd as sea
+    from spark_ensemble_amd.utils.io import synthetic_classification
+
+    df = synthetic_classification(30000, 64, k=2, seed=7, device=DEV)
+    m = sea.LogisticRegression().setMaxIter(40).fit(df)
+    out = m.transform(df)
+    acc = float((out["prediction"] == df["label"]).float().mean())
+    assert acc > 0.8, acc

@@ -16,7 +16,10 @@ import sys
 import sysconfig
 
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
-SRC = [os.path.join(REPO, "csrc", "ops.hip")]
+SRC = [
+    os.path.join(REPO, "csrc", "ops.hip"),
+    os.path.join(REPO, "csrc", "linear.hip"),
+]
 OUT = os.path.join(REPO, "spark_ensemble_amd", "_hip_ops.so")
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 
