@@ -735,6 +735,201 @@ void line_search_eval(torch::Tensor payload, torch::Tensor label,
 #undef LS_LAUNCH
 }
 
+// ---------------------------------------------------------------------------
+// split_argmax: fused best-split search over node histograms.
+//   Replaces the eager chain cumsum -> score -> where -> argmax -> gather of
+//   reference.split_search (ops/reference.py:150-204; semantics: XGBoost-style
+//   newton gain |G|^2/(H+lam), reference analog: the split evaluation inside
+//   MLlib DecisionTree invoked via fitBaseLearner, ensembleParams.scala:64-81).
+//   One wave scans one feature's B bins (lane-chunked + wave prefix scan in
+//   registers), packs (sortable gain, feat*B+bin) into a u64 and atomicMax's
+//   a per-node cell — no [n,F,B,C] cumsum/gain tensors ever touch HBM.
+// ---------------------------------------------------------------------------
+
+__device__ inline unsigned f32_sortable(float f) {
+  unsigned u = __float_as_uint(f);
+  return (u & 0x80000000u) ? ~u : (u | 0x80000000u);
+}
+__device__ inline float sortable_f32(unsigned s) {
+  unsigned u = (s & 0x80000000u) ? (s & 0x7FFFFFFFu) : ~s;
+  return __uint_as_float(u);
+}
+
+template <int C>
+__global__ void split_argmax_kernel(
+    unsigned long long* __restrict__ best,  // [n_nodes] pre-zeroed
+    const float* __restrict__ hist,         // [n_nodes, F, B, C]
+    int F, int B, int D, float lam, float min_child_weight,
+    float min_instances) {
+  const int node = blockIdx.x;
+  const int f = blockIdx.y * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  const int lane = threadIdx.x & 63;
+  if (f >= F) return;
+  const float* h = hist + (((int64_t)node * F + f) * (int64_t)B) * C;
+  const int BPL = (B + 63) >> 6;  // <= 4 for B <= 256
+  const int idx_c = C - 1;        // count channel
+
+  float loc[4][C];
+  float tot[C];
+#pragma unroll
+  for (int c = 0; c < C; ++c) tot[c] = 0.0f;
+  for (int i = 0; i < BPL; ++i) {
+    const int b = lane * BPL + i;
+#pragma unroll
+    for (int c = 0; c < C; ++c) loc[i][c] = 0.0f;
+    if (b < B) {
+#pragma unroll
+      for (int c = 0; c < C; ++c) {
+        loc[i][c] = h[(int64_t)b * C + c];
+        tot[c] += loc[i][c];
+      }
+    }
+  }
+  // wave-wide exclusive prefix + parent totals (shfl scan per channel)
+  float pre[C], parent[C];
+#pragma unroll
+  for (int c = 0; c < C; ++c) {
+    float v = tot[c];
+    for (int off = 1; off < 64; off <<= 1) {
+      const float u = __shfl_up(v, off, 64);
+      if (lane >= off) v += u;
+    }
+    pre[c] = v - tot[c];
+    parent[c] = __shfl(v, 63, 64);
+  }
+  float pg2 = 0.0f;
+#pragma unroll
+  for (int d = 0; d < C; ++d)
+    if (d < D) pg2 += parent[d] * parent[d];
+  const float parent_score = pg2 / (parent[D] + lam);
+
+  float run[C];
+#pragma unroll
+  for (int c = 0; c < C; ++c) run[c] = pre[c];
+  unsigned best_s = 0u;
+  int best_b = -1;
+  for (int i = 0; i < BPL; ++i) {
+    const int b = lane * BPL + i;
+    if (b >= B) break;
+#pragma unroll
+    for (int c = 0; c < C; ++c) run[c] += loc[i][c];
+    if (b < B - 1) {  // last bin cannot split
+      const float hl = run[D], hr = parent[D] - run[D];
+      const float cl = run[idx_c], cr = parent[idx_c] - run[idx_c];
+      if (hl >= min_child_weight && hr >= min_child_weight &&
+          cl >= min_instances && cr >= min_instances) {
+        float gl = 0.0f, gr = 0.0f;
+#pragma unroll
+        for (int d = 0; d < C; ++d)
+          if (d < D) {
+            const float l = run[d], r = parent[d] - run[d];
+            gl += l * l;
+            gr += r * r;
+          }
+        const float gain =
+            gl / (hl + lam) + gr / (hr + lam) - parent_score;
+        const unsigned s = f32_sortable(gain);
+        if (s > best_s) { best_s = s; best_b = b; }
+      }
+    }
+  }
+  // wave argmax
+  for (int off = 32; off > 0; off >>= 1) {
+    const unsigned os = __shfl_down(best_s, off, 64);
+    const int ob = __shfl_down(best_b, off, 64);
+    if (os > best_s) { best_s = os; best_b = ob; }
+  }
+  if (lane == 0 && best_b >= 0) {
+    const unsigned long long packed =
+        ((unsigned long long)best_s << 32) | (unsigned)(f * B + best_b);
+    atomicMax(&best[node], packed);
+  }
+}
+
+template <int C>
+__global__ void split_decode_kernel(
+    float* __restrict__ gain, int* __restrict__ feat, int* __restrict__ bin,
+    float* __restrict__ left_stats,  // [n, C]
+    const unsigned long long* __restrict__ best,
+    const float* __restrict__ hist, int F, int B, float min_info_gain) {
+  const int node = blockIdx.x;
+  const unsigned long long p = best[node];
+  const float g = sortable_f32((unsigned)(p >> 32));
+  __shared__ float acc[C];
+  if (threadIdx.x < C) acc[threadIdx.x] = 0.0f;
+  __syncthreads();
+  if (p == 0ull || !(g >= min_info_gain)) {
+    if (threadIdx.x == 0) {
+      gain[node] = -INFINITY;
+      feat[node] = -1;
+      bin[node] = -1;
+    }
+    if (threadIdx.x < C) left_stats[node * C + threadIdx.x] = 0.0f;
+    return;
+  }
+  const int fb = (int)(p & 0xFFFFFFFFull);
+  const int f = fb / B, b = fb % B;
+  const float* h = hist + (((int64_t)node * F + f) * (int64_t)B) * C;
+  float part[C];
+#pragma unroll
+  for (int c = 0; c < C; ++c) part[c] = 0.0f;
+  for (int i = threadIdx.x; i <= b; i += blockDim.x)
+#pragma unroll
+    for (int c = 0; c < C; ++c) part[c] += h[(int64_t)i * C + c];
+#pragma unroll
+  for (int c = 0; c < C; ++c)
+    if (part[c] != 0.0f) atomicAdd(&acc[c], part[c]);
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    gain[node] = g;
+    feat[node] = f;
+    bin[node] = b;
+  }
+  if (threadIdx.x < C) left_stats[node * C + threadIdx.x] = acc[threadIdx.x];
+}
+
+void split_argmax(torch::Tensor gain, torch::Tensor feat, torch::Tensor bin,
+                  torch::Tensor left_stats, torch::Tensor hist, int64_t d_dims,
+                  double lam, double min_child_weight, double min_instances,
+                  double min_info_gain) {
+  CHECK_GPU(hist); CHECK_CONTIG(hist);
+  const int n = (int)hist.size(0);
+  const int F = (int)hist.size(1);
+  const int B = (int)hist.size(2);
+  const int C = (int)hist.size(3);
+  const int D = d_dims > 0 ? (int)d_dims : C - 2;
+  TORCH_CHECK(C >= 2 && C <= 8, "split_argmax: 2 <= C <= 8");
+  TORCH_CHECK(B >= 2 && B <= 256, "split_argmax: 2 <= B <= 256");
+  TORCH_CHECK(D >= 1 && D < C, "split_argmax: bad D");
+  auto best = torch::zeros({n}, hist.options().dtype(torch::kInt64));
+  auto stream = at::hip::getCurrentHIPStream();
+  const int fpb = 4;  // features (waves) per block
+#define SA_LAUNCH(CC)                                                         \
+  do {                                                                        \
+    hipLaunchKernelGGL(split_argmax_kernel<CC>,                               \
+                       dim3(n, (F + fpb - 1) / fpb), dim3(64 * fpb), 0,       \
+                       stream,                                                \
+                       (unsigned long long*)best.data_ptr<int64_t>(),         \
+                       hist.data_ptr<float>(), F, B, D, (float)lam,           \
+                       (float)min_child_weight, (float)min_instances);        \
+    hipLaunchKernelGGL(split_decode_kernel<CC>, dim3(n), dim3(64), 0, stream, \
+                       gain.data_ptr<float>(), feat.data_ptr<int>(),          \
+                       bin.data_ptr<int>(), left_stats.data_ptr<float>(),     \
+                       (unsigned long long*)best.data_ptr<int64_t>(),         \
+                       hist.data_ptr<float>(), F, B, (float)min_info_gain);   \
+  } while (0)
+  switch (C) {
+    case 2: SA_LAUNCH(2); break;
+    case 3: SA_LAUNCH(3); break;
+    case 4: SA_LAUNCH(4); break;
+    case 5: SA_LAUNCH(5); break;
+    case 6: SA_LAUNCH(6); break;
+    case 7: SA_LAUNCH(7); break;
+    case 8: SA_LAUNCH(8); break;
+  }
+#undef SA_LAUNCH
+}
+
 // csrc/linear.hip
 void logreg_loss_grad(torch::Tensor payload, torch::Tensor x, torch::Tensor y,
                       torch::Tensor w, torch::Tensor wmat, bool has_bias);
@@ -744,6 +939,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("logreg_loss_grad", &logreg_loss_grad,
         "single-pass fused logistic loss+gradient");
   m.def("logreg_fused_supported", &logreg_fused_supported);
+  m.def("split_argmax", &split_argmax,
+        "fused best-split gain scan + argmax over node histograms");
   m.def("sample_weights", &sample_weights, "counter-based Poisson/Bernoulli row weights");
   m.def("bin_features", &bin_features, "quantile binning f32 -> u8");
   m.def("hist_build", &hist_build, "LDS-staged node histograms");

@@ -105,6 +105,20 @@ def hist_build(bins, gh, row_idx, node_offsets, num_bins):
 
 
 def split_search(hist, lam=1e-6, min_child_weight=0.0, min_instances=1.0, min_info_gain=0.0, d_dims=-1):
+    if hist.is_cuda:
+        m = _require_hip("split_argmax")
+        if m is not None:
+            n, f, b, c = hist.shape
+            gain = torch.empty(n, dtype=torch.float32, device=hist.device)
+            feat = torch.empty(n, dtype=torch.int32, device=hist.device)
+            bin_ = torch.empty(n, dtype=torch.int32, device=hist.device)
+            left_stats = torch.empty(n, c, dtype=torch.float32, device=hist.device)
+            m.split_argmax(
+                gain, feat, bin_, left_stats, hist.contiguous(),
+                int(d_dims), float(lam), float(min_child_weight),
+                float(min_instances), float(min_info_gain),
+            )
+            return gain, feat.long(), bin_.long(), left_stats
     return reference.split_search(hist, lam, min_child_weight, min_instances, min_info_gain, d_dims)
 
 

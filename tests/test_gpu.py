@@ -289,3 +289,23 @@ def test_logistic_regression_gpu_fit():
     out = m.transform(df)
     acc = float((out["prediction"] == df["label"]).float().mean())
     assert acc > 0.8, acc
+
+
+@pytest.mark.parametrize("n,f,b,c,d", [(3, 40, 32, 4, 2), (8, 64, 256, 2, 1),
+                                       (1, 256, 256, 3, 1)])
+def test_split_argmax_matches_reference(hip, ref, n, f, b, c, d):
+    g = torch.Generator().manual_seed(21)
+    hist = torch.rand(n, f, b, c, generator=g)
+    hist[..., :d] -= 0.5  # signed gradient channels
+    hist_gpu = hist.to(DEV)
+    for mig, mcw in [(0.0, 0.0), (0.05, 0.3)]:
+        got = hip.split_search(hist_gpu, 1e-6, mcw, 1.0, mig, d_dims=d)
+        want = ref.split_search(hist, 1e-6, mcw, 1.0, mig, d_dims=d)
+        gg, gf, gb, gls = [t.cpu() for t in got]
+        wg, wf, wb, wls = want
+        fin = torch.isfinite(wg)
+        assert torch.equal(torch.isfinite(gg), fin)
+        assert torch.allclose(gg[fin], wg[fin], rtol=1e-3, atol=1e-4)
+        assert torch.equal(gf[fin], wf[fin]), (gf, wf, gg, wg)
+        assert torch.equal(gb[fin], wb[fin])
+        assert torch.allclose(gls[fin], wls[fin].float(), rtol=1e-4, atol=1e-4)
