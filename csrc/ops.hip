@@ -129,32 +129,49 @@ void bin_features(torch::Tensor out, torch::Tensor x, torch::Tensor edges) {
 
 // ---------------------------------------------------------------------------
 // hist_build
-//   grid = (row_chunks, feature_groups); LDS histogram [FG][B][C] f32.
-//   chunks: int32 [n_chunks, 4] = (node, start, len, single_chunk_flag).
-//   Chunk sizes are host-chosen so the grid ~ fills the chip 2-3x over;
-//   a node covered by ONE chunk flushes with plain stores (no atomics).
+//   grid = (row_chunks, feature_groups); LDS histogram [FG][B][CELLS] u64.
+//
+//   KEY gfx950 design point (measured, tools/probe_hist2.hip +
+//   profiles/r01_hist_probe.md): LDS f32 atomicAdd (ds_add_f32) runs ~13x
+//   slower than LDS u64 integer atomicAdd (ds_add_u64) — 101 vs 1395 G
+//   bump/s at B=256 — so the histogram accumulates in FIXED POINT: each
+//   64-bit cell packs one SIGNED channel (gradient, high 32) and one
+//   NON-NEGATIVE channel (hessian/count, low 32) and is bumped with ONE
+//   ds_add_u64.  Channel convention (tree_grower.py): channels [0, D) are
+//   signed gradients, channels [D, C) are non-negative hess/count, so
+//   CELLS = max(D, C - D).  Host-chosen per-channel scales guarantee
+//   |sum| < 2^30 per chunk per bin (no field overflow, low->high carries
+//   impossible since low fields are non-negative and bounded).
+//
+//   chunks: int32 [n_chunks, 4] = (node, start, len, single_chunk_flag);
+//   quantization scales ride in the same upload (one H2D per launch).
+//   Flush converts back to f32 into out [n_nodes, F, B, C]; single-chunk
+//   nodes use plain stores, multi-chunk nodes f32 global atomics.
 // ---------------------------------------------------------------------------
 
-template <int C>
+template <int DC, int NC>
 __global__ void hist_build_kernel(
     float* __restrict__ out,            // [n_nodes, F, B, C]
     const uint8_t* __restrict__ bins,   // [N, F]
     const float* __restrict__ gh,       // [N, C]
     const int* __restrict__ row_idx,    // [M]
-    const int* __restrict__ chunks,     // [n_chunks, 4]
+    const int* __restrict__ chunks,     // [2*C + n_chunks*4] (scales first)
     int F, int B, int FG) {
-  extern __shared__ float lds[];  // FG * B * C
-  const int chunk = blockIdx.x;
+  constexpr int C = DC + NC;
+  constexpr int CELLS = DC > NC ? DC : NC;
+  extern __shared__ unsigned long long lds64[];  // FG * B * CELLS
+  const float* scales = reinterpret_cast<const float*>(chunks);
+  const int* chk = chunks + 2 * C + blockIdx.x * 4;
   const int fg = blockIdx.y;
   const int f0 = fg * FG;
   const int nf = min(FG, F - f0);
-  const int node = chunks[chunk * 4 + 0];
-  const int start = chunks[chunk * 4 + 1];
-  const int len = chunks[chunk * 4 + 2];
-  const int single = chunks[chunk * 4 + 3];
+  const int node = chk[0];
+  const int start = chk[1];
+  const int len = chk[2];
+  const int single = chk[3];
 
-  const int lds_size = FG * B * C;
-  for (int i = threadIdx.x; i < lds_size; i += blockDim.x) lds[i] = 0.0f;
+  const int lds_cells = FG * B * CELLS;
+  for (int i = threadIdx.x; i < lds_cells; i += blockDim.x) lds64[i] = 0ull;
   __syncthreads();
 
   // bins rows are read 16 bytes at a time (uint4) when the group start is
@@ -164,9 +181,20 @@ __global__ void hist_build_kernel(
   for (int i = threadIdx.x; i < len; i += blockDim.x) {
     const int r = row_idx[start + i];
     const float* g = gh + (int64_t)r * C;
-    float gloc[C];
+    // quantize once per row, pack one u64 addend per cell
+    unsigned long long addend[CELLS];
 #pragma unroll
-    for (int c = 0; c < C; ++c) gloc[c] = g[c];
+    for (int c = 0; c < CELLS; ++c) addend[c] = 0ull;
+#pragma unroll
+    for (int d = 0; d < DC; ++d) {
+      const int iv = __float2int_rn(g[d] * scales[d]);
+      addend[d] |= ((unsigned long long)(unsigned)iv) << 32;
+    }
+#pragma unroll
+    for (int c = 0; c < NC; ++c) {
+      const int iv = __float2int_rn(g[DC + c] * scales[DC + c]);
+      addend[c] |= (unsigned)iv;
+    }
     if (vec16) {
       const uint4 bv = *reinterpret_cast<const uint4*>(
           bins + (int64_t)r * F + f0);
@@ -176,48 +204,70 @@ __global__ void hist_build_kernel(
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
           const int b = (w[q] >> (8 * j)) & 0xff;
-          float* cell = lds + (((q * 4 + j) * B) + b) * C;
+          unsigned long long* cell = lds64 + (((q * 4 + j) * B) + b) * CELLS;
 #pragma unroll
-          for (int c = 0; c < C; ++c) atomicAdd(cell + c, gloc[c]);
+          for (int c = 0; c < CELLS; ++c) atomicAdd(cell + c, addend[c]);
         }
       }
     } else {
       const uint8_t* br = bins + (int64_t)r * F + f0;
       for (int f = 0; f < nf; ++f) {
         const int b = br[f];
-        float* cell = lds + ((f * B) + b) * C;
+        unsigned long long* cell = lds64 + ((f * B) + b) * CELLS;
 #pragma unroll
-        for (int c = 0; c < C; ++c) atomicAdd(cell + c, gloc[c]);
+        for (int c = 0; c < CELLS; ++c) atomicAdd(cell + c, addend[c]);
       }
     }
   }
   __syncthreads();
 
+  // flush: decode fixed point -> f32 channels
   float* dst = out + (((int64_t)node * F + f0) * B) * C;
-  if (single) {
-    for (int i = threadIdx.x; i < nf * B * C; i += blockDim.x) dst[i] = lds[i];
-  } else {
-    for (int i = threadIdx.x; i < nf * B * C; i += blockDim.x) {
-      float v = lds[i];
-      if (v != 0.0f) atomicAdd(dst + i, v);
+  for (int i = threadIdx.x; i < nf * B; i += blockDim.x) {
+    const int f = i / B, b = i - f * B;
+    const unsigned long long* cell = lds64 + ((f * B) + b) * CELLS;
+    float vals[C];
+#pragma unroll
+    for (int d = 0; d < DC; ++d)
+      vals[d] = (float)(int)(unsigned)(cell[d] >> 32) / scales[d];
+#pragma unroll
+    for (int c = 0; c < NC; ++c)
+      vals[DC + c] =
+          (float)(int)(unsigned)(cell[c] & 0xFFFFFFFFull) / scales[DC + c];
+    float* o = dst + ((int64_t)f * B + b) * C;
+    if (single) {
+#pragma unroll
+      for (int c = 0; c < C; ++c) o[c] = vals[c];
+    } else {
+#pragma unroll
+      for (int c = 0; c < C; ++c)
+        if (vals[c] != 0.0f) atomicAdd(o + c, vals[c]);
     }
   }
 }
 
 void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
                 torch::Tensor row_idx, torch::Tensor node_offsets,
-                int64_t num_bins) {
+                int64_t num_bins, int64_t d_dims, torch::Tensor max_abs) {
   CHECK_GPU(out); CHECK_GPU(bins); CHECK_GPU(gh); CHECK_GPU(row_idx);
   CHECK_CONTIG(out); CHECK_CONTIG(bins); CHECK_CONTIG(gh); CHECK_CONTIG(row_idx);
   TORCH_CHECK(!node_offsets.is_cuda(), "node_offsets stays on host");
+  TORCH_CHECK(!max_abs.is_cuda(), "max_abs stays on host");
   const int F = (int)bins.size(1);
   const int B = (int)num_bins;
   const int C = (int)gh.size(1);
+  const int D = d_dims > 0 ? (int)d_dims : (C >= 3 ? C - 2 : C - 1);
+  const int NN = C - D;
   TORCH_CHECK(C <= 8, "gh channels capped at 8 (grad dims + hess + count)");
+  TORCH_CHECK(D >= 1 && NN >= 1 && NN <= 2,
+              "hist_build: channels must be D grads + 1-2 nonneg, got C=", C,
+              " D=", D);
+  TORCH_CHECK(max_abs.numel() == C, "max_abs must have C entries");
+  const int CELLS = std::max(D, NN);
 
-  // feature-group size: prefer 16 (vectorized 16-B bin reads); keep
-  // LDS <= 48 KiB so 3 blocks/CU stay resident
-  int FG = std::max<int>(1, std::min<int>(F, 49152 / (B * C * 4)));
+  // feature-group size: prefer 16 (vectorized 16-B bin reads); LDS budget
+  // 64 KiB/block (2 blocks/CU of 160 KiB; atomic-rate bound, not occupancy)
+  int FG = std::max<int>(1, std::min<int>(F, 65536 / (B * CELLS * 8)));
   if (FG >= 16) FG = 16;
   const int n_groups = (int)ceil_div(F, FG);
 
@@ -226,13 +276,27 @@ void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
   auto offs = node_offsets.accessor<int64_t, 1>();
   int64_t total_rows = 0;
   for (int nd = 0; nd < n_nodes; ++nd) total_rows += offs[nd + 1] - offs[nd];
-  const int64_t resident = 256 * 3;  // CUs x blocks/CU at 48 KiB LDS
+  const int64_t resident = 256 * 2;  // CUs x blocks/CU at 64 KiB LDS
   const int64_t target_chunks =
       std::max<int64_t>(1, (resident * 3) / std::max(1, n_groups));
   int64_t chunk_rows = std::max<int64_t>(
       4096, ceil_div(total_rows, target_chunks));
+  // fixed-point headroom: per-chunk per-bin |sum| must stay < 2^30
+  const int64_t CHUNK_CAP = 1 << 20;
+  chunk_rows = std::min(chunk_rows, CHUNK_CAP);
 
-  std::vector<int> chunk_v;
+  // quantization scales: scale_c = 2^30 / (chunk_rows * max_abs_c)
+  auto ma = max_abs.to(torch::kFloat32).accessor<float, 1>();
+  std::vector<int> chunk_v(2 * C);
+  float* scales_f = reinterpret_cast<float*>(chunk_v.data());
+  for (int c = 0; c < C; ++c) {
+    float m = ma[c];
+    if (!(m > 0.0f) || !std::isfinite(m)) m = 1.0f;
+    double s = (double)(1u << 30) / ((double)chunk_rows * (double)m);
+    // clamp so a single value cannot overflow int32 either
+    s = std::min(s, (double)(1u << 30) / (double)m);
+    scales_f[c] = (float)s;
+  }
   for (int nd = 0; nd < n_nodes; ++nd) {
     int64_t s = offs[nd], e = offs[nd + 1];
     const int single = (e - s) <= chunk_rows ? 1 : 0;
@@ -243,29 +307,36 @@ void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
       chunk_v.push_back(single);
     }
   }
-  if (chunk_v.empty()) return;
-  const int n_chunks = (int)(chunk_v.size() / 4);
+  if ((int)chunk_v.size() == 2 * C) return;
+  const int n_chunks = (int)((chunk_v.size() - 2 * C) / 4);
   auto chunks = torch::from_blob(chunk_v.data(), {(int64_t)chunk_v.size()},
                                  torch::kInt32)
                     .to(bins.device(), /*non_blocking=*/false);
 
   auto stream = at::hip::getCurrentHIPStream();
-  const size_t lds_bytes = (size_t)FG * B * C * 4;
-#define HB_LAUNCH(CC)                                                        \
-  hipLaunchKernelGGL(hist_build_kernel<CC>, dim3(n_chunks, n_groups),        \
-                     dim3(256), lds_bytes, stream, out.data_ptr<float>(),    \
-                     bins.data_ptr<uint8_t>(), gh.data_ptr<float>(),         \
-                     row_idx.data_ptr<int>(), chunks.data_ptr<int>(), F, B,  \
-                     FG)
-  switch (C) {
-    case 2: HB_LAUNCH(2); break;
-    case 3: HB_LAUNCH(3); break;
-    case 4: HB_LAUNCH(4); break;
-    case 5: HB_LAUNCH(5); break;
-    case 6: HB_LAUNCH(6); break;
-    case 7: HB_LAUNCH(7); break;
-    case 8: HB_LAUNCH(8); break;
-    default: TORCH_CHECK(false, "hist_build: unsupported channel count ", C);
+  const size_t lds_bytes = (size_t)FG * B * CELLS * 8;
+#define HB_LAUNCH(DD, NNN)                                                   \
+  hipLaunchKernelGGL((hist_build_kernel<DD, NNN>),                           \
+                     dim3(n_chunks, n_groups), dim3(256), lds_bytes, stream, \
+                     out.data_ptr<float>(), bins.data_ptr<uint8_t>(),        \
+                     gh.data_ptr<float>(), row_idx.data_ptr<int>(),          \
+                     chunks.data_ptr<int>(), F, B, FG)
+  const int key = D * 10 + NN;
+  switch (key) {
+    case 11: HB_LAUNCH(1, 1); break;
+    case 12: HB_LAUNCH(1, 2); break;
+    case 21: HB_LAUNCH(2, 1); break;
+    case 22: HB_LAUNCH(2, 2); break;
+    case 31: HB_LAUNCH(3, 1); break;
+    case 32: HB_LAUNCH(3, 2); break;
+    case 41: HB_LAUNCH(4, 1); break;
+    case 42: HB_LAUNCH(4, 2); break;
+    case 51: HB_LAUNCH(5, 1); break;
+    case 52: HB_LAUNCH(5, 2); break;
+    case 61: HB_LAUNCH(6, 1); break;
+    case 62: HB_LAUNCH(6, 2); break;
+    case 71: HB_LAUNCH(7, 1); break;
+    default: TORCH_CHECK(false, "hist_build: unsupported D/NN ", D, "/", NN);
   }
 #undef HB_LAUNCH
 }

@@ -102,6 +102,9 @@ def grow_tree(
         ).contiguous()
     C = gh.shape[1]
     idx_c = C - 1
+    # per-channel abs maxima for the kernel's fixed-point quantization
+    # (computed ONCE per fit; one device sync amortized over all levels)
+    gh_max = gh.abs().amax(dim=0).cpu() if bins.is_cuda else None
 
     if row_mask is not None:
         row_idx = row_mask.nonzero(as_tuple=True)[0].to(torch.int32)
@@ -155,7 +158,7 @@ def grow_tree(
         # ----- histograms for this level ---------------------------------
         if hists is None:
             # root level: build everything
-            new_h = ops.hist_build(bins, gh, row_idx, offsets, B)
+            new_h = ops.hist_build(bins, gh, row_idx, offsets, B, D, gh_max)
             if comm is not None:
                 comm.all_reduce_(new_h)
             hists = new_h
@@ -179,7 +182,8 @@ def grow_tree(
                     else row_idx[segs[0][0] : segs[0][1]]
                 )
                 bh = ops.hist_build(
-                    bins, gh, build_rows, torch.tensor(b_off, dtype=torch.int64), B
+                    bins, gh, build_rows, torch.tensor(b_off, dtype=torch.int64),
+                    B, D, gh_max,
                 )
                 if comm is not None:
                     comm.all_reduce_(bh)

@@ -82,13 +82,25 @@ def bin_features(x, edges):
     return reference.bin_features(x, edges)
 
 
-def hist_build(bins, gh, row_idx, node_offsets, num_bins):
+def hist_build(bins, gh, row_idx, node_offsets, num_bins, d_dims=-1, max_abs=None):
+    """Per-(node, feature, bin) channel sums.
+
+    Channel contract (tree_grower.py): gh[:, :d_dims] are SIGNED gradient
+    channels, gh[:, d_dims:] are NON-NEGATIVE hess/count channels — the
+    gfx950 kernel accumulates in packed fixed-point u64 LDS cells
+    (ds_add_u64 is ~13x faster than ds_add_f32 on gfx950, see
+    profiles/r01_hist_probe.md).  ``max_abs`` (host [C] tensor of per-channel
+    abs maxima) sizes the quantization; pass it from the caller to avoid a
+    device sync per level — when None it is computed here (one sync).
+    """
     if bins.is_cuda:
         m = _require_hip("hist_build")
         if m is not None:
             n_nodes = node_offsets.numel() - 1
             F = bins.shape[1]
             C = gh.shape[1]
+            if max_abs is None:
+                max_abs = gh.abs().amax(dim=0).cpu()
             out = torch.zeros(
                 n_nodes, F, num_bins, C, dtype=torch.float32, device=bins.device
             )
@@ -99,6 +111,8 @@ def hist_build(bins, gh, row_idx, node_offsets, num_bins):
                 row_idx.to(torch.int32),
                 node_offsets.to(torch.int64).cpu(),
                 int(num_bins),
+                int(d_dims),
+                max_abs.to(torch.float32),
             )
             return out
     return reference.hist_build(bins, gh, row_idx, node_offsets, num_bins)

@@ -57,10 +57,13 @@ def test_hist_build_matches_reference(hip, ref):
     g = torch.Generator().manual_seed(2)
     n, f, b, d = 30000, 40, 32, 2
     bins = torch.randint(0, b, (n, f), generator=g, dtype=torch.uint8).to(DEV)
-    gh = torch.randn(n, d + 2, generator=g).to(DEV)
+    # channel contract: first d signed grads, rest non-negative hess/count
+    gh = torch.cat(
+        [torch.randn(n, d, generator=g), torch.rand(n, 2, generator=g)], dim=1
+    ).to(DEV)
     rows = torch.randperm(n, generator=g)[: n - 100].to(torch.int32).to(DEV)
     offs = torch.tensor([0, 9000, 9000, n - 100])
-    got = hip.hist_build(bins, gh, rows, offs, b)
+    got = hip.hist_build(bins, gh, rows, offs, b, d)
     want = ref.hist_build(bins.cpu(), gh.cpu(), rows.cpu(), offs, b)
     assert torch.allclose(got.cpu(), want, atol=2e-2, rtol=1e-4)
 
@@ -69,10 +72,12 @@ def test_hist_build_256bins_multiclass(hip, ref):
     g = torch.Generator().manual_seed(3)
     n, f, b, d = 20000, 17, 256, 5
     bins = torch.randint(0, b, (n, f), generator=g, dtype=torch.uint8).to(DEV)
-    gh = torch.rand(n, d + 2, generator=g).to(DEV)
+    gh = torch.cat(
+        [torch.randn(n, d, generator=g), torch.rand(n, 2, generator=g)], dim=1
+    ).to(DEV)
     rows = torch.arange(n, dtype=torch.int32).to(DEV)
     offs = torch.tensor([0, n])
-    got = hip.hist_build(bins, gh, rows, offs, b)
+    got = hip.hist_build(bins, gh, rows, offs, b, d)
     want = ref.hist_build(bins.cpu(), gh.cpu(), rows.cpu(), offs, b)
     assert torch.allclose(got.cpu(), want, atol=2e-2, rtol=1e-4)
 
