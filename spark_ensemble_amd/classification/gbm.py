@@ -155,6 +155,30 @@ class GBMClassifier(ProbabilisticClassifier, _GBMClassifierParams):
         weights: List[List[float]] = []
         i = 0
         v = 0
+
+        # -- resume from a round-state checkpoint ----------
+        from ..utils import checkpoint as ckpt
+
+        ckpt_dir = self.getCheckpointDir()
+        resumed = ckpt.load_round_state(ckpt_dir)
+        if resumed:
+            r0, models, weights, extra = resumed
+            r0 = min(r0, k_stages)
+            models, weights = models[:r0], weights[:r0]
+            for j, (ms, wts) in enumerate(zip(models, weights)):
+                xs_j = slice_features(x, subspaces[j])
+                wt_t = torch.tensor(wts, dtype=torch.float32, device=device)
+                dirs = torch.stack([m.predict(xs_j) for m in ms], dim=1)
+                predictions = predictions + dirs * wt_t.unsqueeze(0)
+                if val is not None:
+                    xvs_j = slice_features(xv, subspaces[j])
+                    vdirs = torch.stack([m.predict(xvs_j) for m in ms], dim=1)
+                    val_pred = val_pred + vdirs * wt_t.unsqueeze(0)
+            best_err = extra.get("best_err", best_err)
+            v = int(extra.get("v", 0))
+            i = r0
+            instr.log_named_value("resumed_from_round", r0)
+
         while i < k_stages and v < self.getOrDefault("numRounds"):
             idx = subspaces[i]
             xs = binned.sliced_features(idx)
@@ -234,6 +258,12 @@ class GBMClassifier(ProbabilisticClassifier, _GBMClassifierParams):
                                 patience=v)
             else:
                 instr.log_round(i, weight=float(iweights[0]))
+            interval = self.getCheckpointInterval()
+            if ckpt_dir and interval > 0 and (i + 1) % interval == 0:
+                ckpt.save_round_state(
+                    ckpt_dir, i + 1, models, weights,
+                    extra={"best_err": best_err, "v": v},
+                )
             i += 1
 
         keep = i - v

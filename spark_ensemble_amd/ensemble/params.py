@@ -166,12 +166,29 @@ class HasCheckpointInterval(Params):
             lambda v: v == -1 or v >= 1,
         )
         self._setDefault(checkpointInterval=10)
+        self._declare_checkpoint_dir()
 
     def getCheckpointInterval(self):
         return self.getOrDefault("checkpointInterval")
 
     def setCheckpointInterval(self, v):
         return self.set("checkpointInterval", v)
+
+    # durable round-state dumps (SURVEY.md §5.4 rebuild of the reference's
+    # PeriodicRDDCheckpointer — here fit() CAN resume a half-trained
+    # ensemble from the dump; the reference's cannot)
+    def getCheckpointDir(self):
+        return self.getOrNone("checkpointDir")
+
+    def setCheckpointDir(self, v):
+        return self.set("checkpointDir", v)
+
+    def _declare_checkpoint_dir(self):
+        self.checkpointDir = self._str_param(
+            "checkpointDir",
+            "directory for durable round-state snapshots (resume on refit)",
+            lower=False,
+        )
 
 
 class HasAggregationDepth(Params):

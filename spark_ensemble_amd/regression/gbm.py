@@ -172,6 +172,28 @@ class GBMRegressor(Regressor, _GBMRegressorParams):
         weights: List[float] = []
         i = 0
         v = 0
+
+        # -- resume from a round-state checkpoint ----------
+        from ..utils import checkpoint as ckpt
+
+        ckpt_dir = self.getCheckpointDir()
+        resumed = ckpt.load_round_state(ckpt_dir)
+        if resumed:
+            r0, models, weights, extra = resumed
+            r0 = min(r0, k_stages)
+            models, weights = models[:r0], weights[:r0]
+            for j, (m, wt) in enumerate(zip(models, weights)):
+                xs_j = slice_features(x, subspaces[j])
+                predictions = predictions + wt * m.predict(xs_j)
+                if val is not None:
+                    val_pred = val_pred + wt * m.predict(
+                        slice_features(xv, subspaces[j])
+                    )
+            best_err = extra.get("best_err", best_err)
+            v = int(extra.get("v", 0))
+            i = r0
+            instr.log_named_value("resumed_from_round", r0)
+
         while i < k_stages and v < self.getOrDefault("numRounds"):
             if loss_name == "huber":
                 delta = dist_quantile((y - predictions).abs(), alpha, None, comm)
@@ -243,6 +265,12 @@ class GBMRegressor(Regressor, _GBMRegressorParams):
                 instr.log_round(i, weight=weight, val_loss=err, patience=v)
             else:
                 instr.log_round(i, weight=weight)
+            interval = self.getCheckpointInterval()
+            if ckpt_dir and interval > 0 and (i + 1) % interval == 0:
+                ckpt.save_round_state(
+                    ckpt_dir, i + 1, models, weights,
+                    extra={"best_err": best_err, "v": v},
+                )
             i += 1
         instr.finish()
 
