@@ -118,6 +118,13 @@ class TensorFrame:
             idx = mask
         return TensorFrame({k: v[idx] for k, v in self._cols.items()})._with_cache_of(self)
 
+    def random_split(self, frac: float, seed: int = 0):
+        """(train, test) split: Bernoulli(frac) per row (Spark
+        ``randomSplit`` analog), deterministic in ``seed``."""
+        g = torch.Generator(device="cpu").manual_seed(seed)
+        mask = (torch.rand(self.count(), generator=g) < frac).to(self.device)
+        return self.filter(mask), self.filter(~mask)
+
     def to(self, device) -> "TensorFrame":
         return TensorFrame({k: v.to(device) for k, v in self._cols.items()})._with_cache_of(self)
 
