@@ -286,7 +286,8 @@ void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
   chunk_rows = std::min(chunk_rows, CHUNK_CAP);
 
   // quantization scales: scale_c = 2^30 / (chunk_rows * max_abs_c)
-  auto ma = max_abs.to(torch::kFloat32).accessor<float, 1>();
+  auto max_abs_f = max_abs.to(torch::kFloat32).contiguous();
+  auto ma = max_abs_f.accessor<float, 1>();
   std::vector<int> chunk_v(2 * C);
   float* scales_f = reinterpret_cast<float*>(chunk_v.data());
   for (int c = 0; c < C; ++c) {
