@@ -299,9 +299,20 @@ def test_logistic_regression_gpu_fit():
 @pytest.mark.parametrize("n,f,b,c,d", [(3, 40, 32, 4, 2), (8, 64, 256, 2, 1),
                                        (1, 256, 256, 3, 1)])
 def test_split_argmax_matches_reference(hip, ref, n, f, b, c, d):
+    # hist must be a REAL histogram (per-feature totals equal — the
+    # hist_build invariant the kernel relies on: the reference scores the
+    # parent from feature 0's totals, the kernel from each feature's own)
     g = torch.Generator().manual_seed(21)
-    hist = torch.rand(n, f, b, c, generator=g)
-    hist[..., :d] -= 0.5  # signed gradient channels
+    rows_n = 4000
+    bins = torch.randint(0, b, (rows_n, f), generator=g, dtype=torch.uint8)
+    gh = torch.cat(
+        [torch.randn(rows_n, d, generator=g),
+         torch.rand(rows_n, c - d, generator=g)], dim=1
+    )
+    seg = rows_n // n
+    offs = torch.tensor([i * seg for i in range(n)] + [rows_n])
+    rows = torch.arange(rows_n, dtype=torch.int32)
+    hist = ref.hist_build(bins, gh, rows, offs, b).float()
     hist_gpu = hist.to(DEV)
     for mig, mcw in [(0.0, 0.0), (0.05, 0.3)]:
         got = hip.split_search(hist_gpu, 1e-6, mcw, 1.0, mig, d_dims=d)
