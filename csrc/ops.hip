@@ -746,17 +746,19 @@ void grad_hess(torch::Tensor grad, torch::Tensor hess, torch::Tensor label,
 // line_search_eval: one pass computing
 //   payload[0]   = sum_i w_i * loss(y_i, p_i + a . d_i)
 //   payload[1+d] = sum_i w_i * d_id * dloss/dp_id
+//   payload[1+D] = sum_i w_i * d_i^2 * d2loss/dp_i^2   (want_hess, D == 1;
+//                  feeds the safeguarded-Newton stage-weight search)
 template <int D>
 __global__ void line_search_eval_kernel(
-    float* __restrict__ payload,        // [1 + D] (pre-zeroed)
+    float* __restrict__ payload,        // [1 + D (+1)] (pre-zeroed)
     const float* __restrict__ label,    // [N, D]
     const float* __restrict__ pred,     // [N, D]
     const float* __restrict__ dir,      // [N, D]
     const float* __restrict__ weight,   // [N]
     const float* __restrict__ coeff,    // [D]
-    int64_t n, int loss_id, float param) {
-  __shared__ float acc[1 + D];
-  for (int i = threadIdx.x; i < D + 1; i += blockDim.x) acc[i] = 0.0f;
+    int64_t n, int loss_id, float param, int want_hess) {
+  __shared__ float acc[2 + D];
+  for (int i = threadIdx.x; i < D + 2; i += blockDim.x) acc[i] = 0.0f;
   __syncthreads();
 
   float cf[D];
@@ -766,6 +768,7 @@ __global__ void line_search_eval_kernel(
   int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
   int64_t stride = gridDim.x * (int64_t)blockDim.x;
   float loss_sum = 0.0f;
+  float hsum = 0.0f;
   float gsum[D];
 #pragma unroll
   for (int d = 0; d < D; ++d) gsum[d] = 0.0f;
@@ -803,23 +806,27 @@ __global__ void line_search_eval_kernel(
       if (!isfinite(l)) l = 3.0e38f;
       loss_sum += w * l;
       gsum[0] += w * dir[i] * g;
+      if (want_hess) hsum += w * dir[i] * dir[i] * h;
     }
   }
 
   // wave reduce then LDS
   for (int off = 32; off > 0; off >>= 1) {
     loss_sum += __shfl_down(loss_sum, off, 64);
+    hsum += __shfl_down(hsum, off, 64);
 #pragma unroll
     for (int d = 0; d < D; ++d) gsum[d] += __shfl_down(gsum[d], off, 64);
   }
   if ((threadIdx.x & 63) == 0) {
     atomicAdd(&acc[0], loss_sum);
+    if (want_hess) atomicAdd(&acc[1 + D], hsum);
 #pragma unroll
     for (int d = 0; d < D; ++d) atomicAdd(&acc[1 + d], gsum[d]);
   }
   __syncthreads();
   if (threadIdx.x == 0) {
     atomicAdd(&payload[0], acc[0]);
+    if (want_hess) atomicAdd(&payload[1 + D], acc[1 + D]);
 #pragma unroll
     for (int d = 0; d < D; ++d) atomicAdd(&payload[1 + d], acc[1 + d]);
   }
@@ -828,13 +835,15 @@ __global__ void line_search_eval_kernel(
 void line_search_eval(torch::Tensor payload, torch::Tensor label,
                       torch::Tensor pred, torch::Tensor dir,
                       torch::Tensor weight, torch::Tensor coeff,
-                      int64_t loss_id, double param) {
+                      int64_t loss_id, double param, bool want_hess) {
   CHECK_GPU(payload); CHECK_GPU(pred);
   int64_t n = pred.size(0);
   int D = (int)pred.size(1);
   TORCH_CHECK(D <= 8, "line_search_eval supports dim <= 8");
   TORCH_CHECK(D == 1 || loss_id == L_LOGLOSS,
               "vector line search only for logloss");
+  TORCH_CHECK(!want_hess || D == 1, "hessian accumulation is scalar-only");
+  TORCH_CHECK(payload.numel() >= 1 + D + (want_hess ? 1 : 0), "payload size");
   auto stream = at::hip::getCurrentHIPStream();
   int blocks = (int)std::min<int64_t>(ceil_div(n, 256 * 8), 2048);
 #define LS_LAUNCH(DD)                                                         \
@@ -842,7 +851,8 @@ void line_search_eval(torch::Tensor payload, torch::Tensor label,
                      stream, payload.data_ptr<float>(),                       \
                      label.data_ptr<float>(), pred.data_ptr<float>(),         \
                      dir.data_ptr<float>(), weight.data_ptr<float>(),         \
-                     coeff.data_ptr<float>(), n, (int)loss_id, (float)param)
+                     coeff.data_ptr<float>(), n, (int)loss_id, (float)param,  \
+                     want_hess ? 1 : 0)
   switch (D) {
     case 1: LS_LAUNCH(1); break;
     case 2: LS_LAUNCH(2); break;

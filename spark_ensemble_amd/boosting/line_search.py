@@ -24,10 +24,12 @@ from ..parallel import Comm
 from .losses import GBMLoss
 
 
-def _eval(loss, label, pred, direction, weight, coeff):
-    """Returns (weighted loss sum, per-dim gradient sums) as a flat tensor
-    [1 + dim] on the data's device.  On GPU this is ONE fused kernel pass
-    (csrc line_search_eval) instead of ~6 eager passes."""
+def _eval(loss, label, pred, direction, weight, coeff, want_hess=False):
+    """Returns (weighted loss sum, per-dim gradient sums[, hessian sum]) as
+    a flat tensor [1 + dim (+1)] on the data's device.  On GPU this is ONE
+    fused kernel pass (csrc line_search_eval) instead of ~6 eager passes.
+    ``want_hess`` (scalar dim only) additionally accumulates
+    sum_i w_i d_i^2 loss''(p_i + a d_i) for the Newton weight search."""
     if pred.is_cuda:
         from ..ops import dispatch
 
@@ -38,17 +40,24 @@ def _eval(loss, label, pred, direction, weight, coeff):
                 ct = coeff.reshape(-1).float().to(pred.device)
             else:
                 ct = torch.full((D,), float(coeff), device=pred.device)
-            payload = torch.zeros(1 + D, dtype=torch.float32, device=pred.device)
+            payload = torch.zeros(
+                1 + D + (1 if want_hess else 0),
+                dtype=torch.float32, device=pred.device,
+            )
             m.line_search_eval(
                 payload, label.contiguous(), pred.contiguous(),
                 direction.contiguous(), weight.contiguous(), ct,
-                loss.loss_id, float(loss.param),
+                loss.loss_id, float(loss.param), bool(want_hess),
             )
             return payload
     p = pred + direction * coeff
     l = (loss.loss(label, p) * weight).sum()
     g = loss.gradient(label, p) * direction * weight.unsqueeze(1)
-    return torch.cat([l.reshape(1), g.sum(dim=0)])
+    out = [l.reshape(1), g.sum(dim=0)]
+    if want_hess:
+        h = loss.hessian(label, p) * direction * direction * weight.unsqueeze(1)
+        out.append(h.sum(dim=0))
+    return torch.cat(out)
 
 
 def optimize_weight_1d(
@@ -63,8 +72,21 @@ def optimize_weight_1d(
     lo: float = 0.0,
     hi: float = 100.0,
 ) -> float:
-    """Brent minimization of the 1-D stage weight on [0, 100] (the
-    reference's SearchInterval — GBMRegressor.scala:413-421)."""
+    """1-D stage-weight minimization on [0, 100] (the reference's Brent
+    SearchInterval — GBMRegressor.scala:413-421).
+
+    For losses with a CONTINUOUS second derivative (squared, logcosh,
+    scaledlogcosh, exponential, bernoulli) a safeguarded Newton iteration
+    on phi'(a) converges in 3-6 fused evaluations instead of Brent's
+    ~25-40 — same minimizer, far fewer host<->device round trips (each
+    evaluation is a kernel launch + an RCCL all-reduce + a sync).  Brent
+    remains for the non-smooth losses (absolute, huber, quantile) and as
+    the fallback when Newton fails to bracket."""
+    if getattr(loss, "smooth", False) and loss.has_hessian:
+        a = _newton_1d(loss, label, pred, direction, weight, comm,
+                       max_iter, tol, lo, hi)
+        if a is not None:
+            return a
     from scipy.optimize import minimize_scalar
 
     cache = {}
@@ -130,3 +152,44 @@ def optimize_weight_nd(
         options={"maxiter": max_iter, "ftol": tol, "gtol": tol},
     )
     return res.x.astype(np.float64)
+
+
+def _newton_1d(loss, label, pred, direction, weight, comm, max_iter, tol,
+               lo, hi):
+    """Safeguarded Newton on phi'(a) over [lo, hi]; returns None to fall
+    back to Brent (non-finite values, no interior minimum)."""
+
+    def eval_at(a):
+        payload = _eval(loss, label, pred, direction, weight, float(a),
+                        want_hess=True)
+        if comm is not None and comm.is_distributed:
+            comm.all_reduce_(payload)
+        return float(payload[0]), float(payload[1]), float(payload[2])
+
+    a = 1.0  # natural stage weight
+    blo, bhi = lo, hi
+    best_a, best_f = None, float("inf")
+    for _ in range(max(8, min(max_iter, 20))):
+        f, g, h = eval_at(a)
+        if not (np.isfinite(f) and np.isfinite(g) and np.isfinite(h)):
+            return None
+        if f < best_f:
+            best_f, best_a = f, a
+        # shrink the safeguard bracket using the sign of phi'
+        if g > 0:
+            bhi = a
+        else:
+            blo = a
+        if abs(g) <= tol * max(1.0, abs(f)) or (bhi - blo) <= tol:
+            return a
+        if h > 1e-12:
+            step = -g / h
+            nxt = a + step
+            if not (blo < nxt < bhi):
+                nxt = 0.5 * (blo + bhi)  # bisect when Newton leaves bracket
+            if abs(nxt - a) <= tol * max(1.0, abs(a)):
+                return nxt
+            a = nxt
+        else:
+            a = 0.5 * (blo + bhi)
+    return best_a

@@ -41,6 +41,9 @@ class GBMLoss:
     name: str = "?"
     dim: int = 1
     has_hessian: bool = False
+    # continuous second derivative => eligible for the Newton stage-weight
+    # search (line_search._newton_1d); huber/absolute/quantile stay False
+    smooth: bool = False
     # scalar parameter forwarded to the fused kernels (alpha/delta/quantile)
     param: float = 0.0
 
@@ -93,6 +96,7 @@ class _ClassificationLoss(GBMLoss):
 
 
 class SquaredLoss(GBMLoss):
+    smooth = True
     name = "squared"
     has_hessian = True
 
@@ -117,6 +121,7 @@ class AbsoluteLoss(GBMLoss):
 
 
 class LogCoshLoss(GBMLoss):
+    smooth = True
     name = "logcosh"
     has_hessian = True
 
@@ -134,6 +139,7 @@ class LogCoshLoss(GBMLoss):
 
 
 class ScaledLogCoshLoss(GBMLoss):
+    smooth = True
     name = "scaledlogcosh"
     has_hessian = True
 
@@ -232,6 +238,7 @@ class LogLoss(_ClassificationLoss):
 
 
 class ExponentialLoss(_ClassificationLoss):
+    smooth = True
     """AdaBoost exponential loss on y in {-1, 1} (reference
     GBMLoss.scala:265-291); dim = 1."""
 
@@ -256,6 +263,7 @@ class ExponentialLoss(_ClassificationLoss):
 
 
 class BernoulliLoss(_ClassificationLoss):
+    smooth = True
     """Logistic loss on +-1 labels (reference GBMLoss.scala:293-318);
     NOTE the reference's raw2probability uses exp(+raw) for class 1 —
     replicated exactly."""

@@ -110,3 +110,44 @@ def test_negative_gradient():
     assert torch.allclose(
         loss.negative_gradient(label, pred), -(loss.gradient(label, pred))
     )
+
+
+def test_newton_line_search_matches_brent():
+    """The safeguarded-Newton stage-weight search (smooth losses) must land
+    on the same minimizer as Brent within tolerance."""
+    import torch
+    from spark_ensemble_amd.boosting.line_search import (
+        _newton_1d, optimize_weight_1d,
+    )
+    from spark_ensemble_amd.boosting.losses import (
+        BernoulliLoss, LogCoshLoss, SquaredLoss,
+    )
+    from scipy.optimize import minimize_scalar
+
+    g = torch.Generator().manual_seed(31)
+    n = 5000
+    for loss in (SquaredLoss(), LogCoshLoss(), BernoulliLoss()):
+        if loss.name == "bernoulli":
+            y = loss.encode_label(torch.randint(0, 2, (n,), generator=g).float())
+        else:
+            y = torch.randn(n, generator=g)
+        y = y.reshape(n, 1)
+        pred = torch.randn(n, 1, generator=g) * 0.3
+        direction = (y - pred) * 0.5 + torch.randn(n, 1, generator=g) * 0.1
+        w = torch.rand(n, generator=g) + 0.5
+
+        a_newton = _newton_1d(loss, y, pred, direction, w, None, 100, 1e-8,
+                              0.0, 100.0)
+        assert a_newton is not None
+
+        def phi(a):
+            p = pred + direction * float(a)
+            return float((loss.loss(y, p) * w).sum())
+
+        res = minimize_scalar(phi, bounds=(0.0, 100.0), method="bounded",
+                              options={"xatol": 1e-10})
+        assert abs(a_newton - float(res.x)) < 1e-3, (
+            loss.name, a_newton, float(res.x))
+        # and the public entry point routes smooth losses through Newton
+        a_pub = optimize_weight_1d(loss, y, pred, direction, w, None)
+        assert abs(a_pub - float(res.x)) < 1e-3
