@@ -325,3 +325,34 @@ def test_split_argmax_matches_reference(hip, ref, n, f, b, c, d):
         assert torch.equal(gf[fin], wf[fin]), (gf, wf, gg, wg)
         assert torch.equal(gb[fin], wb[fin])
         assert torch.allclose(gls[fin], wls[fin].float(), rtol=1e-4, atol=1e-4)
+
+
+def test_hist_build_deterministic(hip):
+    """instead of the reference's no-op race story, the
+    rebuild pins determinism — fixed-point integer LDS accumulation makes
+    hist_build bitwise-reproducible across runs (f32 atomics would not be)."""
+    g = torch.Generator().manual_seed(33)
+    n, f, b, d = 200000, 32, 256, 1
+    bins = torch.randint(0, b, (n, f), generator=g, dtype=torch.uint8).to(DEV)
+    gh = torch.cat(
+        [torch.randn(n, d, generator=g), torch.rand(n, 1, generator=g)], dim=1
+    ).to(DEV)
+    rows = torch.randperm(n, generator=g).to(torch.int32).to(DEV)
+    offs = torch.tensor([0, n // 3, n])
+    h1 = hip.hist_build(bins, gh, rows, offs, b, d)
+    h2 = hip.hist_build(bins, gh, rows, offs, b, d)
+    assert torch.equal(h1, h2)
+
+
+def test_tree_fit_deterministic():
+    """Two identical fits on GPU produce identical predictions (tested at
+    the estimator level: split decisions + leaf values reproduce)."""
+    import spark_ensemble_amd as sea
+    from spark_ensemble_amd.utils.io import synthetic_classification
+
+    df = synthetic_classification(100000, 24, k=2, seed=44, device=DEV)
+    m1 = sea.GBMClassifier().setNumBaseLearners(3).setSeed(9).fit(df)
+    m2 = sea.GBMClassifier().setNumBaseLearners(3).setSeed(9).fit(df)
+    r1 = m1.transform(df)["rawPrediction"]
+    r2 = m2.transform(df)["rawPrediction"]
+    assert torch.equal(r1, r2)
