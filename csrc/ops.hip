@@ -216,9 +216,10 @@ __global__ void hist_build_kernel(
   for (int i = threadIdx.x; i < lds_cells; i += blockDim.x) lds64[i] = 0ull;
   __syncthreads();
 
-  // bins rows are read 16 bytes at a time (uint4) when the group start is
-  // 16-aligned and inside the row
+  // bins rows are read 16 (or 2x16) bytes at a time (uint4) when the
+  // group start is 16-aligned and inside the row
   const bool vec16 = (FG == 16) && (f0 + 16 <= F) && ((F & 15) == 0);
+  const bool vec32 = (FG == 32) && (f0 + 32 <= F) && ((F & 31) == 0);
 
   for (int i = threadIdx.x; i < len; i += blockDim.x) {
     const int r = row_idx[start + i];
@@ -237,18 +238,22 @@ __global__ void hist_build_kernel(
       const int iv = __float2int_rn(g[DC + c] * scales[DC + c]);
       addend[c] |= (unsigned)iv;
     }
-    if (vec16) {
-      const uint4 bv = *reinterpret_cast<const uint4*>(
-          bins + (int64_t)r * F + f0);
-      const unsigned w[4] = {bv.x, bv.y, bv.z, bv.w};
+    if (vec16 || vec32) {
+      const int nh = vec32 ? 2 : 1;
+      for (int hh = 0; hh < nh; ++hh) {
+        const uint4 bv = *reinterpret_cast<const uint4*>(
+            bins + (int64_t)r * F + f0 + 16 * hh);
+        const unsigned w[4] = {bv.x, bv.y, bv.z, bv.w};
 #pragma unroll
-      for (int q = 0; q < 4; ++q) {
+        for (int q = 0; q < 4; ++q) {
 #pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          const int b = (w[q] >> (8 * j)) & 0xff;
-          unsigned long long* cell = lds64 + (((q * 4 + j) * B) + b) * CELLS;
+          for (int j = 0; j < 4; ++j) {
+            const int b = (w[q] >> (8 * j)) & 0xff;
+            unsigned long long* cell =
+                lds64 + (((16 * hh + q * 4 + j) * B) + b) * CELLS;
 #pragma unroll
-          for (int c = 0; c < CELLS; ++c) atomicAdd(cell + c, addend[c]);
+            for (int c = 0; c < CELLS; ++c) atomicAdd(cell + c, addend[c]);
+          }
         }
       }
     } else {
@@ -307,10 +312,15 @@ void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
   TORCH_CHECK(max_abs.numel() == C, "max_abs must have C entries");
   const int CELLS = std::max(D, NN);
 
-  // feature-group size: prefer 16 (vectorized 16-B bin reads); LDS budget
-  // 64 KiB/block (2 blocks/CU of 160 KiB; atomic-rate bound, not occupancy)
+  // feature-group size: 16 (32-KiB LDS at CELLS=1) or 32 (64-KiB, halves
+  // the random-row read amplification at deep levels); SEA_HIST_FG overrides
+  static const int fg_env = []() {
+    const char* e = getenv("SEA_HIST_FG");
+    return e ? atoi(e) : 0;
+  }();
   int FG = std::max<int>(1, std::min<int>(F, 65536 / (B * CELLS * 8)));
-  if (FG >= 16) FG = 16;
+  if (FG >= 32 && (F % 32) == 0 && (fg_env == 32)) FG = 32;
+  else if (FG >= 16) FG = 16;
   const int n_groups = (int)ceil_div(F, FG);
 
   // ---- adaptive chunking: target ~resident-grid x OVERSUB blocks --------

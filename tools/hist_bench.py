@@ -36,12 +36,14 @@ def main():
     rows = torch.arange(N, dtype=torch.int32, device=dev)
     offs1 = torch.tensor([0, N])
 
+    gh2 = gh3[:, :2].contiguous()
     bench("root B=256 C=3 contiguous", bins, gh3, rows, offs1, 256)
-    bench("root B=256 C=2 contiguous", bins, gh3[:, :2].contiguous(), rows, offs1, 256)
+    bench("root B=256 C=2 contiguous", bins, gh2, rows, offs1, 256)
     bench("root B=64  C=3 contiguous", (bins & 63), gh3, rows, offs1, 64)
 
     # shuffled rows (deep-level access pattern)
     perm = torch.randperm(N, generator=g, device=dev).to(torch.int32)
+    bench("root B=256 C=2 shuffled", bins, gh2, perm, offs1, 256)
     bench("root B=256 C=3 shuffled", bins, gh3, perm, offs1, 256)
 
     # 64-node level, half the rows
@@ -49,6 +51,7 @@ def main():
     seg = N // 2 // n_nodes
     offs = torch.tensor([i * seg for i in range(n_nodes + 1)])
     rows_half = perm[: seg * n_nodes]
+    bench("64 nodes B=256 C=2 shuffled half-rows", bins, gh2, rows_half, offs, 256)
     bench("64 nodes B=256 C=3 shuffled half-rows", bins, gh3, rows_half, offs, 256)
 
 
