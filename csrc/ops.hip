@@ -318,9 +318,15 @@ void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
     const char* e = getenv("SEA_HIST_FG");
     return e ? atoi(e) : 0;
   }();
-  int FG = std::max<int>(1, std::min<int>(F, 65536 / (B * CELLS * 8)));
+  static const int lds_budget = []() {
+    const char* e = getenv("SEA_HIST_LDS");
+    return e ? atoi(e) : 65536;
+  }();
+  int FG = std::max<int>(1, std::min<int>(F, lds_budget / (B * CELLS * 8)));
   if (FG >= 32 && (F % 32) == 0 && (fg_env == 32)) FG = 32;
   else if (FG >= 16) FG = 16;
+  else if (FG >= 8) FG = 8;
+  else if (FG >= 4) FG = 4;
   const int n_groups = (int)ceil_div(F, FG);
 
   // ---- adaptive chunking: target ~resident-grid x OVERSUB blocks --------

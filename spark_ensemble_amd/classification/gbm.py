@@ -3,9 +3,10 @@
 Re-creates reference classification/GBMClassifier.scala:219-496:
   * losses {logloss, exponential, bernoulli} (default logloss); LogLoss has
     dim = K, so each round fits K base regressors on per-class
-    pseudo-residuals (reference fits them in parallel Futures :377-411 — on
-    one MI355X they are fused into a single multi-output histogram pass when
-    the base learner is the built-in tree, else fitted sequentially),
+    pseudo-residuals (reference fits them in parallel Futures :377-411;
+    here they are fitted sequentially — each tree fit is itself a fully
+    GPU-parallel histogram pass, so there is no idle hardware to recover
+    the way the reference's driver threads do),
   * init {prior, uniform}; binary + dim-1 prior -> constant log-odds model
     (reference :275-283),
   * newton pseudo-residuals with per-dim hessian floor 1e-2 and weights
@@ -217,9 +218,12 @@ class GBMClassifier(ProbabilisticClassifier, _GBMClassifierParams):
                 )
                 imodels.append(self.fit_base_learner(learner, fr, weight_col="weight"))
 
-            directions = torch.stack(
-                [m.predict(xs) for m in imodels], dim=1
-            )  # [N, dim]
+            def _dir(m):
+                tp = getattr(m, "_train_pred", None)
+                return tp if tp is not None and tp.shape[0] == xs.shape[0] \
+                    else m.predict(xs)
+
+            directions = torch.stack([_dir(m) for m in imodels], dim=1)
 
             if optimized:
                 if dim == 1:

@@ -151,9 +151,14 @@ class DecisionTreeRegressor(Regressor, _TreeParams):
         total_w = comm.all_reduce_scalar(float(w.sum()))
         gp = self._grow_params(total_w)
         mask = w > 0 if bool((w == 0).any()) else None
-        tree = grow_tree(bins, edges, grad, w, gp, comm, row_mask=mask)
+        tp_out: list = []
+        tree = grow_tree(bins, edges, grad, w, gp, comm, row_mask=mask,
+                         train_pred_out=tp_out)
         model = DecisionTreeRegressionModel()
         model._set_tree(tree, x.shape[1])
+        # training-row predictions captured during growth (leaf scatter
+        # instead of a post-hoc tree walk; GBM's margin update uses this)
+        model._train_pred = tp_out[0].squeeze(1) if tp_out else None
         model._copy_cols_from(self)
         return model
 

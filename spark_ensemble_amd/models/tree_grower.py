@@ -76,11 +76,18 @@ def grow_tree(
     comm: Optional[Comm] = None,
     row_mask: Optional[torch.Tensor] = None,
     hess_is_count: Optional[bool] = None,
+    train_pred_out: Optional[list] = None,
 ) -> Dict[str, torch.Tensor]:
     """Grow one tree; returns flat node arrays:
 
     feature [n] int32 (-1 leaf), threshold [n] f32, left_child [n] int32
     (right = left+1), leaf_value [n, D] f32.
+
+    ``train_pred_out``: pass a list to ALSO receive the training-row
+    predictions [N, D] (appended) — training already partitions every row
+    to its leaf, so this is a leaf-value scatter instead of the full tree
+    walk GBM would otherwise pay per round (only produced when row_mask is
+    None, i.e. the tree saw every row).
     """
     device = bins.device
     N, F = bins.shape
@@ -147,12 +154,28 @@ def grow_tree(
 
     edges_cpu = edges.cpu()
 
+    # leaf-wise (rows, value) capture for train_pred_out
+    pred_rows: List[torch.Tensor] = []
+    pred_vals: List[torch.Tensor] = []
+    capture = train_pred_out is not None and row_mask is None
+
+    def _capture_leaves(act_idx, offs, ridx, ids):
+        if not capture:
+            return
+        ol = offs.tolist() if isinstance(offs, torch.Tensor) else offs
+        for i in act_idx:
+            s0, e0 = int(ol[i]), int(ol[i + 1])
+            if e0 > s0:
+                pred_rows.append(ridx[s0:e0])
+                pred_vals.append(leaves[ids[i]])
+
     for depth in range(params.max_depth + 1):
         n_active = len(node_ids)
         if n_active == 0:
             break
         if depth == params.max_depth:
             _finalize_leaves(node_ids, totals, leaves, params, D)
+            _capture_leaves(range(n_active), offsets, row_idx, node_ids)
             break
 
         # ----- histograms for this level ---------------------------------
@@ -219,6 +242,7 @@ def grow_tree(
                 params,
                 D,
             )
+            _capture_leaves(ns_idx.tolist(), offsets, row_idx, node_ids)
         if not bool(do_split.any()):
             break
 
@@ -296,6 +320,14 @@ def grow_tree(
         "left_child": torch.tensor(lefts, dtype=torch.int32),
         "leaf_value": leaf_value,
     }
+    if capture and pred_rows:
+        rows_cat = torch.cat(pred_rows).long()
+        lv_mat = torch.stack([v.reshape(D) for v in pred_vals]).to(device)
+        counts = torch.tensor([r.numel() for r in pred_rows], device=device)
+        vals = torch.repeat_interleave(lv_mat, counts, dim=0)
+        tp = torch.zeros(N, D, dtype=torch.float32, device=device)
+        tp[rows_cat] = vals
+        train_pred_out.append(tp)
     return {k: v.to(device) for k, v in tree.items()}
 
 

@@ -230,7 +230,11 @@ class GBMRegressor(Regressor, _GBMRegressorParams):
             fr = binned.fit_frame(learner, res_label, res_weight, idx, xs)
             model = self.fit_base_learner(learner, fr, weight_col="weight")
 
-            direction = model.predict(xs)  # [N]
+            # tree base learners capture train-row predictions during
+            # growth (leaf scatter) — avoids a full tree walk per round
+            direction = getattr(model, "_train_pred", None)
+            if direction is None or direction.shape[0] != xs.shape[0]:
+                direction = model.predict(xs)  # [N]
 
             if optimized:
                 sol = optimize_weight_1d(
