@@ -194,6 +194,7 @@ void bin_features(torch::Tensor out, torch::Tensor x, torch::Tensor edges) {
 template <int DC, int NC>
 __global__ void hist_build_kernel(
     float* __restrict__ out,            // [n_nodes, F, B, C]
+    long long* __restrict__ stage,      // [n_nodes, F, B, C] i64 (multi-chunk)
     const uint8_t* __restrict__ bins,   // [N, F]
     const float* __restrict__ gh,       // [N, C]
     const int* __restrict__ row_idx,    // [M]
@@ -268,28 +269,59 @@ __global__ void hist_build_kernel(
   }
   __syncthreads();
 
-  // flush: decode fixed point -> f32 channels
-  float* dst = out + (((int64_t)node * F + f0) * B) * C;
-  for (int i = threadIdx.x; i < nf * B; i += blockDim.x) {
-    const int f = i / B, b = i - f * B;
-    const unsigned long long* cell = lds64 + ((f * B) + b) * CELLS;
-    float vals[C];
+  // flush. Single-chunk nodes: decode fixed point -> f32 straight into
+  // out (plain stores).  Multi-chunk nodes: accumulate INTEGER partials
+  // into the i64 staging buffer — integer atomics are order-independent,
+  // so histograms stay bitwise deterministic across runs (decode kernel
+  // below converts once at the end).
+  if (single) {
+    float* dst = out + (((int64_t)node * F + f0) * B) * C;
+    for (int i = threadIdx.x; i < nf * B; i += blockDim.x) {
+      const int f = i / B, b = i - f * B;
+      const unsigned long long* cell = lds64 + ((f * B) + b) * CELLS;
+      float* o = dst + ((int64_t)f * B + b) * C;
 #pragma unroll
-    for (int d = 0; d < DC; ++d)
-      vals[d] = (float)(int)(unsigned)(cell[d] >> 32) / scales[d];
+      for (int d = 0; d < DC; ++d)
+        o[d] = (float)(int)(unsigned)(cell[d] >> 32) / scales[d];
 #pragma unroll
-    for (int c = 0; c < NC; ++c)
-      vals[DC + c] =
-          (float)(int)(unsigned)(cell[c] & 0xFFFFFFFFull) / scales[DC + c];
-    float* o = dst + ((int64_t)f * B + b) * C;
-    if (single) {
-#pragma unroll
-      for (int c = 0; c < C; ++c) o[c] = vals[c];
-    } else {
-#pragma unroll
-      for (int c = 0; c < C; ++c)
-        if (vals[c] != 0.0f) atomicAdd(o + c, vals[c]);
+      for (int c = 0; c < NC; ++c)
+        o[DC + c] =
+            (float)(int)(unsigned)(cell[c] & 0xFFFFFFFFull) / scales[DC + c];
     }
+  } else {
+    long long* sdst = stage + (((int64_t)node * F + f0) * B) * C;
+    for (int i = threadIdx.x; i < nf * B; i += blockDim.x) {
+      const int f = i / B, b = i - f * B;
+      const unsigned long long* cell = lds64 + ((f * B) + b) * CELLS;
+      long long* o = sdst + ((int64_t)f * B + b) * C;
+#pragma unroll
+      for (int d = 0; d < DC; ++d) {
+        const int v = (int)(unsigned)(cell[d] >> 32);
+        if (v) atomicAdd((unsigned long long*)(o + d),
+                         (unsigned long long)(long long)v);
+      }
+#pragma unroll
+      for (int c = 0; c < NC; ++c) {
+        const int v = (int)(unsigned)(cell[c] & 0xFFFFFFFFull);
+        if (v) atomicAdd((unsigned long long*)(o + DC + c),
+                         (unsigned long long)(long long)v);
+      }
+    }
+  }
+}
+
+// decode the i64 staging sums of multi-chunk nodes into f32 out
+__global__ void hist_decode_kernel(float* __restrict__ out,
+                                   const long long* __restrict__ stage,
+                                   const int* __restrict__ nodes,  // multi-chunk node ids (after scales)
+                                   const float* __restrict__ scales, int FBC,
+                                   int C) {
+  const int node = nodes[blockIdx.y];
+  const int64_t base = (int64_t)node * FBC;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < FBC;
+       i += gridDim.x * (int64_t)blockDim.x) {
+    const int c = (int)(i % C);
+    out[base + i] = (float)stage[base + i] / scales[c];
   }
 }
 
@@ -363,9 +395,11 @@ void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
     s = std::min(s, (double)(1u << 30) / (double)m);
     scales_f[c] = (float)s;
   }
+  std::vector<int> multi_nodes;
   for (int nd = 0; nd < n_nodes; ++nd) {
     int64_t s = offs[nd], e = offs[nd + 1];
     const int single = (e - s) <= chunk_rows ? 1 : 0;
+    if (!single) multi_nodes.push_back(nd);
     for (int64_t c = s; c < e; c += chunk_rows) {
       chunk_v.push_back(nd);
       chunk_v.push_back((int)c);
@@ -379,12 +413,23 @@ void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
                                  torch::kInt32)
                     .to(bins.device(), /*non_blocking=*/false);
 
+  // i64 integer staging for multi-chunk nodes (order-independent flush =>
+  // bitwise-deterministic histograms)
+  torch::Tensor stage;
+  long long* stage_ptr = nullptr;
+  if (!multi_nodes.empty()) {
+    stage = torch::zeros({(int64_t)n_nodes, (int64_t)F, (int64_t)B, (int64_t)C},
+                         out.options().dtype(torch::kInt64));
+    stage_ptr = stage.data_ptr<long long>();
+  }
+
   auto stream = at::hip::getCurrentHIPStream();
   const size_t lds_bytes = (size_t)FG * B * CELLS * 8;
 #define HB_LAUNCH(DD, NNN)                                                   \
   hipLaunchKernelGGL((hist_build_kernel<DD, NNN>),                           \
                      dim3(n_chunks, n_groups), dim3(256), lds_bytes, stream, \
-                     out.data_ptr<float>(), bins.data_ptr<uint8_t>(),        \
+                     out.data_ptr<float>(), stage_ptr,                       \
+                     bins.data_ptr<uint8_t>(),                               \
                      gh.data_ptr<float>(), row_idx.data_ptr<int>(),          \
                      chunks.data_ptr<int>(), F, B, FG)
   const int key = D * 10 + NN;
@@ -405,12 +450,29 @@ void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
     default: TORCH_CHECK(false, "hist_build: unsupported D/NN ", D, "/", NN);
   }
 #undef HB_LAUNCH
+  if (!multi_nodes.empty()) {
+    auto nodes_t = torch::from_blob(multi_nodes.data(),
+                                    {(int64_t)multi_nodes.size()},
+                                    torch::kInt32)
+                       .to(bins.device(), /*non_blocking=*/false);
+    const int FBC = F * B * C;
+    const int dblocks = (int)std::min<int64_t>(ceil_div(FBC, 256), 1024);
+    hipLaunchKernelGGL(hist_decode_kernel,
+                       dim3(dblocks, (int)multi_nodes.size()), dim3(256), 0,
+                       stream, out.data_ptr<float>(), stage_ptr,
+                       nodes_t.data_ptr<int>(),
+                       reinterpret_cast<const float*>(chunks.data_ptr<int>()),
+                       FBC, C);
+  }
 }
 
 // ---------------------------------------------------------------------------
-// partition_rows: two passes per chunk.
-//   pass 1: block counts its left rows, ONE global atomic per block per side
-//   pass 2: replay, scatter via per-wave ballot + LDS block cursors
+// partition_rows: two passes per chunk, ONE bins read.
+//   pass 1: decide left/right per row, stash the decision as a ballot
+//           bitmask in LDS (1 bit/row, chunk <= 64 Ki rows = 8 KiB), count
+//           via popcount, ONE global atomic per block per side
+//   pass 2: replay from the LDS bitmask (no second random bins gather),
+//           scatter via ballot prefix + LDS block cursors
 // ---------------------------------------------------------------------------
 
 __global__ void partition_kernel(
@@ -423,24 +485,33 @@ __global__ void partition_kernel(
     const int* __restrict__ thr,       // [n_nodes]
     int F) {
   __shared__ int base_l, base_r, loc_l, loc_r;
+  __shared__ int wl[8];
+  extern __shared__ unsigned long long bits[];  // ceil(chunk_rows/64)
   const int chunk = blockIdx.x;
   const int node = chunks[chunk * 3 + 0];
   const int start = chunks[chunk * 3 + 1];
   const int len = chunks[chunk * 3 + 2];
   const int f = feat[node];
   const int t = thr[node];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
 
-  // pass 1: count left
+  // pass 1: decide + stash bitmask + count
   int my_l = 0;
-  for (int i = threadIdx.x; i < len; i += blockDim.x) {
-    const int r = row_idx[start + i];
-    my_l += (f < 0) || (bins[(int64_t)r * F + f] <= t) ? 1 : 0;
+  for (int i = threadIdx.x; i < ((len + 63) & ~63); i += blockDim.x) {
+    const bool active = i < len;
+    bool left = false;
+    if (active) {
+      const int r = row_idx[start + i];
+      left = (f < 0) || (bins[(int64_t)r * F + f] <= t);
+    }
+    const unsigned long long m = __ballot(active && left);
+    if (lane == 0) {
+      bits[i >> 6] = m;
+      my_l += (int)__popcll(m);
+    }
   }
-  // wave then block reduction
-  for (int off = 32; off > 0; off >>= 1) my_l += __shfl_down(my_l, off, 64);
-  __shared__ int wl[8];
-  const int wave = threadIdx.x / 64;
-  if ((threadIdx.x & 63) == 0) wl[wave] = my_l;
+  if (lane == 0) wl[wave] = my_l;
   __syncthreads();
   if (threadIdx.x == 0) {
     int tot = 0;
@@ -452,18 +523,14 @@ __global__ void partition_kernel(
   }
   __syncthreads();
 
-  // pass 2: scatter (per-wave ballot + LDS cursors)
+  // pass 2: scatter from the stashed decisions (row_idx re-read is
+  // sequential; bins is NOT touched again)
   for (int i = threadIdx.x; i < ((len + 63) & ~63); i += blockDim.x) {
     const bool active = i < len;
-    int r = 0;
-    bool left = false;
-    if (active) {
-      r = row_idx[start + i];
-      left = (f < 0) || (bins[(int64_t)r * F + f] <= t);
-    }
-    const unsigned long long lmask = __ballot(active && left);
-    const unsigned long long rmask = __ballot(active && !left);
-    const int lane = threadIdx.x & 63;
+    const unsigned long long word = bits[i >> 6];
+    const bool left = (word >> lane) & 1ull;
+    const unsigned long long lmask = word;
+    const unsigned long long rmask = __ballot(active) & ~word;
     int lbase = 0, rbase = 0;
     if (lane == 0) {
       lbase = atomicAdd(&loc_l, (int)__popcll(lmask));
@@ -472,6 +539,7 @@ __global__ void partition_kernel(
     lbase = __shfl(lbase, 0, 64);
     rbase = __shfl(rbase, 0, 64);
     if (active) {
+      const int r = row_idx[start + i];
       if (left) {
         const int pos = base_l + lbase + (int)__popcll(lmask & ((1ull << lane) - 1ull));
         new_rows[pos] = r;
@@ -498,6 +566,7 @@ void partition_rows(torch::Tensor new_rows, torch::Tensor left_counts,
   int64_t total_rows = offs[n_nodes];
   int64_t chunk_rows =
       std::max<int64_t>(8192, ceil_div(total_rows, (int64_t)2048));
+  chunk_rows = std::min<int64_t>(chunk_rows, 65536);  // 8 KiB LDS bitmask
   for (int nd = 0; nd < n_nodes; ++nd) {
     cur_v[nd * 2 + 0] = (int)offs[nd];
     cur_v[nd * 2 + 1] = (int)offs[nd + 1];
@@ -515,7 +584,9 @@ void partition_rows(torch::Tensor new_rows, torch::Tensor left_counts,
                                    torch::kInt32)
                       .to(bins.device());
     const int n_chunks = (int)(chunk_v.size() / 3);
-    hipLaunchKernelGGL(partition_kernel, dim3(n_chunks), dim3(256), 0, stream,
+    const size_t bit_lds = (size_t)((chunk_rows + 63) / 64) * 8;
+    hipLaunchKernelGGL(partition_kernel, dim3(n_chunks), dim3(256), bit_lds,
+                       stream,
                        new_rows.data_ptr<int>(), cursors.data_ptr<int>(),
                        bins.data_ptr<uint8_t>(), row_idx.data_ptr<int>(),
                        chunks.data_ptr<int>(), feat.data_ptr<int>(),

@@ -140,11 +140,18 @@ def partition_rows(bins, row_idx, node_offsets, feat, thr):
     if bins.is_cuda:
         m = _require_hip("partition_rows")
         if m is not None:
-            return _partition_hip(m, bins, row_idx, node_offsets, feat, thr)
+            new_rows, lc_dev, offs_cpu = partition_rows_async(
+                bins, row_idx, node_offsets, feat, thr
+            )
+            return partition_rows_finish(new_rows, lc_dev, offs_cpu)
     return reference.partition_rows(bins, row_idx, node_offsets, feat, thr)
 
 
-def _partition_hip(m, bins, row_idx, node_offsets, feat, thr):
+def partition_rows_async(bins, row_idx, node_offsets, feat, thr):
+    """Launch the partition kernel WITHOUT syncing (feat/thr may be live
+    device tensors straight from split_argmax); call
+    :func:`partition_rows_finish` after the caller's own device sync."""
+    m = _require_hip("partition_rows")
     n = node_offsets.numel() - 1
     offs_cpu = node_offsets.to(torch.int64).cpu()
     row_idx = row_idx.to(torch.int32)
@@ -159,10 +166,14 @@ def _partition_hip(m, bins, row_idx, node_offsets, feat, thr):
         feat.to(torch.int32).to(bins.device),
         thr.to(torch.int32).to(bins.device),
     )
+    return new_rows, left_counts, offs_cpu
+
+
+def partition_rows_finish(new_rows, left_counts, offs_cpu):
+    n = offs_cpu.numel() - 1
     lc = left_counts.cpu().to(torch.int64)
-    offs = offs_cpu
     sizes = torch.empty(2 * n, dtype=torch.int64)
-    seg = offs[1:] - offs[:-1]
+    seg = offs_cpu[1:] - offs_cpu[:-1]
     sizes[0::2] = lc
     sizes[1::2] = seg - lc
     new_offs = torch.cat([torch.zeros(1, dtype=torch.int64), sizes.cumsum(0)])
