@@ -420,7 +420,7 @@ void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
   if (!multi_nodes.empty()) {
     stage = torch::zeros({(int64_t)n_nodes, (int64_t)F, (int64_t)B, (int64_t)C},
                          out.options().dtype(torch::kInt64));
-    stage_ptr = stage.data_ptr<long long>();
+    stage_ptr = reinterpret_cast<long long*>(stage.data_ptr<int64_t>());
   }
 
   auto stream = at::hip::getCurrentHIPStream();

@@ -227,6 +227,13 @@ def grow_tree(
             params.min_info_gain,
             d_dims=D,
         )
+        # overlap: on GPU launch the partition straight off the device
+        # split outputs, BEFORE the host fetch below drains the stream —
+        # one device sync per level instead of two
+        pr_async = None
+        if bins.is_cuda:
+            pr_async = ops.partition_rows_async(bins, row_idx, offsets, feat, b)
+
         gain_cpu = gain.cpu()
         feat_cpu = feat.cpu()
         b_cpu = b.cpu()
@@ -263,9 +270,13 @@ def grow_tree(
                 child_ids.append(None)
 
         # partition rows of splitting nodes
-        new_rows, new_offs, _ = ops.partition_rows(
-            bins, row_idx, offsets, split_feat.to(torch.int32), b_cpu.to(torch.int32)
-        )
+        if pr_async is not None:
+            new_rows, new_offs, _ = ops.partition_rows_finish(*pr_async)
+        else:
+            new_rows, new_offs, _ = ops.partition_rows(
+                bins, row_idx, offsets,
+                split_feat.to(torch.int32), b_cpu.to(torch.int32),
+            )
         offs_list = new_offs.tolist()
 
         # ----- next level bookkeeping ------------------------------------
