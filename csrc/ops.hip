@@ -911,12 +911,27 @@ __global__ void line_search_eval_kernel(
     for (int d = 0; d < D; ++d) atomicAdd(&acc[1 + d], gsum[d]);
   }
   __syncthreads();
+  // per-block partial (no global atomics) -> deterministic fixed-order
+  // reduction in ls_reduce_kernel
   if (threadIdx.x == 0) {
-    atomicAdd(&payload[0], acc[0]);
-    if (want_hess) atomicAdd(&payload[1 + D], acc[1 + D]);
+    float* row = payload + (int64_t)blockIdx.x * (2 + D);
+    row[0] = acc[0];
+    if (want_hess) row[1 + D] = acc[1 + D];
 #pragma unroll
-    for (int d = 0; d < D; ++d) atomicAdd(&payload[1 + d], acc[1 + d]);
+    for (int d = 0; d < D; ++d) row[1 + d] = acc[1 + d];
   }
+}
+
+// sum the [n_blocks, 2+D] partials in block order (single block, fixed
+// order => bitwise deterministic across runs)
+__global__ void ls_reduce_kernel(float* __restrict__ out,      // [2 + D]
+                                 const float* __restrict__ partials,
+                                 int n_blocks, int width) {
+  const int k = threadIdx.x;  // one thread per payload slot
+  if (k >= width) return;
+  float s = 0.0f;
+  for (int b = 0; b < n_blocks; ++b) s += partials[(int64_t)b * width + k];
+  out[k] = s;
 }
 
 void line_search_eval(torch::Tensor payload, torch::Tensor label,
@@ -933,9 +948,12 @@ void line_search_eval(torch::Tensor payload, torch::Tensor label,
   TORCH_CHECK(payload.numel() >= 1 + D + (want_hess ? 1 : 0), "payload size");
   auto stream = at::hip::getCurrentHIPStream();
   int blocks = (int)std::min<int64_t>(ceil_div(n, 256 * 8), 2048);
+  const int width = 2 + D;
+  auto partials = torch::empty({(int64_t)blocks * width},
+                               payload.options());
 #define LS_LAUNCH(DD)                                                         \
   hipLaunchKernelGGL(line_search_eval_kernel<DD>, dim3(blocks), dim3(256), 0, \
-                     stream, payload.data_ptr<float>(),                       \
+                     stream, partials.data_ptr<float>(),                      \
                      label.data_ptr<float>(), pred.data_ptr<float>(),         \
                      dir.data_ptr<float>(), weight.data_ptr<float>(),         \
                      coeff.data_ptr<float>(), n, (int)loss_id, (float)param,  \
@@ -951,6 +969,9 @@ void line_search_eval(torch::Tensor payload, torch::Tensor label,
     case 8: LS_LAUNCH(8); break;
   }
 #undef LS_LAUNCH
+  hipLaunchKernelGGL(ls_reduce_kernel, dim3(1), dim3(64), 0, stream,
+                     payload.data_ptr<float>(), partials.data_ptr<float>(),
+                     blocks, std::min(width, (int)payload.numel()));
 }
 
 // ---------------------------------------------------------------------------

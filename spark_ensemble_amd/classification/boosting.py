@@ -90,14 +90,27 @@ class BoostingClassifier(ProbabilisticClassifier, _BoostingClassifierParams):
         boosting_w = w.clone()
         sum_w = comm.all_reduce_scalar(float(boosting_w.sum()))
 
+        # resume: replay saved models through the reweighting loop (same
+        # stats recomputed, no refit) — SURVEY.md §5.4
+        from ..utils import checkpoint as ckpt
+
+        ckpt_dir = self.getCheckpointDir()
+        resumed = ckpt.load_round_state(ckpt_dir)
+        replay = resumed[1][:k] if resumed else []
+        if replay:
+            instr.log_named_value("resumed_from_round", len(replay))
+
         models: List = []
         est_weights: List[float] = []
         i = 0
         done = False
         while i < k and not done and sum_w > 0:
             norm_w = boosting_w / sum_w
-            fr = binned.fit_frame(learner, y, norm_w)
-            model = self.fit_base_learner(learner, fr, weight_col="weight")
+            if i < len(replay):
+                model = replay[i]
+            else:
+                fr = binned.fit_frame(learner, y, norm_w)
+                model = self.fit_base_learner(learner, fr, weight_col="weight")
 
             if algo == "real":
                 if not hasattr(model, "predictProbability"):
@@ -141,6 +154,10 @@ class BoostingClassifier(ProbabilisticClassifier, _BoostingClassifierParams):
             instr.log_round(i, error=est_err,
                             weight=est_weights[-1] if est_weights else 0.0,
                             sum_w=sum_w)
+            interval = self.getCheckpointInterval()
+            if (ckpt_dir and i >= len(replay) and interval > 0
+                    and (i + 1) % interval == 0):
+                ckpt.save_round_state(ckpt_dir, i + 1, models, est_weights)
             i += 1
 
         instr.finish()

@@ -80,3 +80,20 @@ def test_save_load_round_state_roundtrip(tmp_path):
     assert torch.allclose(models[0].predict(x), m1.predict(x))
     ck.clear(str(tmp_path / "s"))
     assert ck.load_round_state(str(tmp_path / "s")) is None
+
+
+def test_boosting_resume_matches_straight_fit(tmp_path):
+    df = synthetic_classification(600, 10, k=2, seed=13)
+
+    def mk(k, ck):
+        e = sea.BoostingClassifier().setNumBaseLearners(k).setSeed(2)
+        if ck:
+            e.setCheckpointInterval(2).setCheckpointDir(str(tmp_path / "bk"))
+        return e
+
+    m_full = mk(5, False).fit(df)
+    mk(2, True).fit(df)  # interrupted (dump at round 2)
+    m_res = mk(5, True).fit(df)
+    a = m_full.transform(df)["rawPrediction"]
+    b = m_res.transform(df)["rawPrediction"]
+    assert torch.allclose(a, b, rtol=1e-5, atol=1e-6)
