@@ -844,8 +844,7 @@ __global__ void line_search_eval_kernel(
     const float* __restrict__ weight,   // [N]
     const float* __restrict__ coeff,    // [D]
     int64_t n, int loss_id, float param, int want_hess) {
-  __shared__ float acc[2 + D];
-  for (int i = threadIdx.x; i < D + 2; i += blockDim.x) acc[i] = 0.0f;
+  __shared__ float wacc[4][2 + D];  // per-wave partials (256 threads = 4 waves)
   __syncthreads();
 
   float cf[D];
@@ -905,20 +904,24 @@ __global__ void line_search_eval_kernel(
     for (int d = 0; d < D; ++d) gsum[d] += __shfl_down(gsum[d], off, 64);
   }
   if ((threadIdx.x & 63) == 0) {
-    atomicAdd(&acc[0], loss_sum);
-    if (want_hess) atomicAdd(&acc[1 + D], hsum);
+    const int wv = threadIdx.x >> 6;
+    wacc[wv][0] = loss_sum;
+    wacc[wv][1 + D] = hsum;
 #pragma unroll
-    for (int d = 0; d < D; ++d) atomicAdd(&acc[1 + d], gsum[d]);
+    for (int d = 0; d < D; ++d) wacc[wv][1 + d] = gsum[d];
   }
   __syncthreads();
-  // per-block partial (no global atomics) -> deterministic fixed-order
-  // reduction in ls_reduce_kernel
+  // per-block partial, waves summed in FIXED order (no atomics anywhere)
+  // -> bitwise deterministic; final fixed-order reduction in
+  // ls_reduce_kernel
   if (threadIdx.x == 0) {
     float* row = payload + (int64_t)blockIdx.x * (2 + D);
-    row[0] = acc[0];
-    if (want_hess) row[1 + D] = acc[1 + D];
-#pragma unroll
-    for (int d = 0; d < D; ++d) row[1 + d] = acc[1 + d];
+    const int nw = (int)(blockDim.x >> 6);
+    for (int k = 0; k < 2 + D; ++k) {
+      float v = 0.0f;
+      for (int wv = 0; wv < nw; ++wv) v += wacc[wv][k];
+      row[k] = v;
+    }
   }
 }
 
