@@ -927,13 +927,13 @@ __global__ void line_search_eval_kernel(
 
 // sum the [n_blocks, 2+D] partials in block order (single block, fixed
 // order => bitwise deterministic across runs)
-__global__ void ls_reduce_kernel(float* __restrict__ out,      // [2 + D]
+__global__ void ls_reduce_kernel(float* __restrict__ out,      // [out_width]
                                  const float* __restrict__ partials,
-                                 int n_blocks, int width) {
+                                 int n_blocks, int stride, int out_width) {
   const int k = threadIdx.x;  // one thread per payload slot
-  if (k >= width) return;
+  if (k >= out_width) return;
   float s = 0.0f;
-  for (int b = 0; b < n_blocks; ++b) s += partials[(int64_t)b * width + k];
+  for (int b = 0; b < n_blocks; ++b) s += partials[(int64_t)b * stride + k];
   out[k] = s;
 }
 
@@ -974,7 +974,7 @@ void line_search_eval(torch::Tensor payload, torch::Tensor label,
 #undef LS_LAUNCH
   hipLaunchKernelGGL(ls_reduce_kernel, dim3(1), dim3(64), 0, stream,
                      payload.data_ptr<float>(), partials.data_ptr<float>(),
-                     blocks, std::min(width, (int)payload.numel()));
+                     blocks, width, std::min(width, (int)payload.numel()));
 }
 
 // ---------------------------------------------------------------------------

@@ -153,7 +153,8 @@ def test_grown_tree_gpu_matches_cpu():
 
 @pytest.mark.parametrize(
     "est",
-    ["bagging_clf", "boosting_clf", "gbm_clf", "gbm_reg", "stacking_reg"],
+    ["bagging_clf", "boosting_clf", "gbm_clf", "gbm_logloss", "gbm_reg",
+     "gbm_huber", "stacking_reg"],
 )
 def test_end_to_end_gpu_fits(est):
     import spark_ensemble_amd as sea
@@ -166,7 +167,7 @@ def test_end_to_end_gpu_fits(est):
         synthetic_regression,
     )
 
-    if est in ("gbm_reg", "stacking_reg"):
+    if est in ("gbm_reg", "gbm_huber", "stacking_reg"):
         df = synthetic_regression(50000, 32, seed=31, device=DEV)
         dft = synthetic_regression(20000, 32, seed=31, split=1, device=DEV)
     else:
@@ -179,8 +180,14 @@ def test_end_to_end_gpu_fits(est):
         m = sea.BoostingClassifier().setNumBaseLearners(5).fit(df)
     elif est == "gbm_clf":
         m = sea.GBMClassifier().setLoss("bernoulli").setNumBaseLearners(5).fit(df)
+    elif est == "gbm_logloss":
+        # vector (K-dim) L-BFGS-B line search path
+        m = sea.GBMClassifier().setLoss("logloss").setNumBaseLearners(3).fit(df)
     elif est == "gbm_reg":
         m = sea.GBMRegressor().setNumBaseLearners(5).fit(df)
+    elif est == "gbm_huber":
+        # non-smooth loss -> Brent line search on the hess-less payload
+        m = sea.GBMRegressor().setLoss("huber").setNumBaseLearners(5).fit(df)
     else:
         m = (
             sea.StackingRegressor()
