@@ -107,7 +107,10 @@ def main():
         res_label = -loss.grad_hess_fused(ylab, predictions)[0].squeeze(1)
         fr = binned.fit_frame(learner, res_label)
         model = gbm.fit_base_learner(learner, fr)
-        direction = model.predict(x).unsqueeze(1)
+        direction = getattr(model, "_train_pred", None)
+        if direction is None:
+            direction = model.predict(x)
+        direction = direction.unsqueeze(1)
         w = torch.ones(x.shape[0], device=device)
         alpha = optimize_weight_1d(
             loss, ylab, predictions, direction, w, comm, max_iter=100, tol=1e-6

@@ -925,16 +925,28 @@ __global__ void line_search_eval_kernel(
   }
 }
 
-// sum the [n_blocks, 2+D] partials in block order (single block, fixed
-// order => bitwise deterministic across runs)
+// sum the [n_blocks, stride] partials with a FIXED thread->block
+// assignment (grid of one block; lane j sums rows j, j+J, ... then lane 0
+// of each slot adds the J lane-partials in order => bitwise deterministic)
 __global__ void ls_reduce_kernel(float* __restrict__ out,      // [out_width]
                                  const float* __restrict__ partials,
                                  int n_blocks, int stride, int out_width) {
-  const int k = threadIdx.x;  // one thread per payload slot
-  if (k >= out_width) return;
+  __shared__ float lp[256];
+  const int k = threadIdx.x % stride;        // payload slot
+  const int j = threadIdx.x / stride;        // lane within slot
+  const int J = blockDim.x / stride;         // lanes per slot
   float s = 0.0f;
-  for (int b = 0; b < n_blocks; ++b) s += partials[(int64_t)b * stride + k];
-  out[k] = s;
+  if (j < J) {
+    for (int b = j; b < n_blocks; b += J)
+      s += partials[(int64_t)b * stride + k];
+  }
+  lp[threadIdx.x] = s;
+  __syncthreads();
+  if (threadIdx.x < out_width) {
+    float v = 0.0f;
+    for (int jj = 0; jj < J; ++jj) v += lp[jj * stride + threadIdx.x];
+    out[threadIdx.x] = v;
+  }
 }
 
 void line_search_eval(torch::Tensor payload, torch::Tensor label,
@@ -972,7 +984,8 @@ void line_search_eval(torch::Tensor payload, torch::Tensor label,
     case 8: LS_LAUNCH(8); break;
   }
 #undef LS_LAUNCH
-  hipLaunchKernelGGL(ls_reduce_kernel, dim3(1), dim3(64), 0, stream,
+  const int rthreads = (256 / width) * width;  // whole lanes per slot
+  hipLaunchKernelGGL(ls_reduce_kernel, dim3(1), dim3(rthreads), 0, stream,
                      payload.data_ptr<float>(), partials.data_ptr<float>(),
                      blocks, width, std::min(width, (int)payload.numel()));
 }
