@@ -196,7 +196,7 @@ def test_end_to_end_gpu_fits(est):
             .setNumFolds(3)
             .fit(df)
         )
-    if est.endswith("reg"):
+    if est in ("gbm_reg", "gbm_huber", "stacking_reg"):
         p = m.predict(dft["features"])
         ss = float(((p - dft["label"]) ** 2).mean())
         var = float(dft["label"].var())
