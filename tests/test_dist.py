@@ -57,6 +57,26 @@ out["gbm_rmse"] = float(((gbm.predict(testr["features"]) - testr["label"]) ** 2)
 bst = sea.BoostingRegressor().setNumBaseLearners(3).fit(dfr)
 out["boost_rmse"] = float(((bst.predict(testr["features"]) - testr["label"]) ** 2).mean() ** 0.5)
 
+# --- out-of-fold stacking across ranks (collective sequence must match
+#     on every rank through 2 learners x 5 folds + final refits) ---------
+from spark_ensemble_amd.models import LinearRegression
+stk = (sea.StackingRegressor()
+       .setBaseLearners([sea.GBMRegressor().setNumBaseLearners(2),
+                         sea.BaggingRegressor().setNumBaseLearners(2)])
+       .setStacker(LinearRegression())
+       .setNumFolds(3).setSeed(4).fit(dfr))
+out["stack_rmse"] = float(((stk.predict(testr["features"]) - testr["label"]) ** 2).mean() ** 0.5)
+
+# --- logistic regression: fused payload all-reduce -----------------------
+lr = sea.LogisticRegression().setMaxIter(15).fit(df)
+outp = lr.transform(test)
+out["logreg_acc"] = float((outp["prediction"] == test["label"]).float().mean())
+
+# --- evaluator all-reduce -------------------------------------------------
+from spark_ensemble_amd.tuning import MulticlassClassificationEvaluator
+ev = MulticlassClassificationEvaluator("accuracy")
+out["eval_acc"] = ev.evaluate(m.transform(df))
+
 if rank == 0:
     with open(os.environ["SEA_OUT"], "w") as f:
         json.dump(out, f)
@@ -126,3 +146,12 @@ def test_dist_gbm_quality(world2_results):
 
 def test_dist_boosting_quality(world2_results):
     assert world2_results["boost_rmse"] < 5.0
+
+
+def test_dist_stacking_quality(world2_results):
+    assert world2_results["stack_rmse"] < 1.2 * world2_results["gbm_rmse"]
+
+
+def test_dist_logreg_and_evaluator(world2_results):
+    assert world2_results["logreg_acc"] > 0.5
+    assert 0.0 <= world2_results["eval_acc"] <= 1.0
