@@ -199,7 +199,7 @@ __global__ void hist_build_kernel(
     const float* __restrict__ gh,       // [N, C]
     const int* __restrict__ row_idx,    // [M]
     const int* __restrict__ chunks,     // [2*C + n_chunks*4] (scales first)
-    int F, int B, int FG) {
+    int F, int B, int FG, int identity_rows) {
   constexpr int C = DC + NC;
   constexpr int CELLS = DC > NC ? DC : NC;
   extern __shared__ unsigned long long lds64[];  // FG * B * CELLS
@@ -223,7 +223,7 @@ __global__ void hist_build_kernel(
   const bool vec32 = (FG == 32) && (f0 + 32 <= F) && ((F & 31) == 0);
 
   for (int i = threadIdx.x; i < len; i += blockDim.x) {
-    const int r = row_idx[start + i];
+    const int r = identity_rows ? start + i : row_idx[start + i];
     const float* g = gh + (int64_t)r * C;
     // quantize once per row, pack one u64 addend per cell
     unsigned long long addend[CELLS];
@@ -327,7 +327,8 @@ __global__ void hist_decode_kernel(float* __restrict__ out,
 
 void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
                 torch::Tensor row_idx, torch::Tensor node_offsets,
-                int64_t num_bins, int64_t d_dims, torch::Tensor max_abs) {
+                int64_t num_bins, int64_t d_dims, torch::Tensor max_abs,
+                bool identity_rows) {
   CHECK_GPU(out); CHECK_GPU(bins); CHECK_GPU(gh); CHECK_GPU(row_idx);
   CHECK_CONTIG(out); CHECK_CONTIG(bins); CHECK_CONTIG(gh); CHECK_CONTIG(row_idx);
   TORCH_CHECK(!node_offsets.is_cuda(), "node_offsets stays on host");
@@ -431,7 +432,8 @@ void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
                      out.data_ptr<float>(), stage_ptr,                       \
                      bins.data_ptr<uint8_t>(),                               \
                      gh.data_ptr<float>(), row_idx.data_ptr<int>(),          \
-                     chunks.data_ptr<int>(), F, B, FG)
+                     chunks.data_ptr<int>(), F, B, FG,                       \
+                     identity_rows ? 1 : 0)
   const int key = D * 10 + NN;
   switch (key) {
     case 11: HB_LAUNCH(1, 1); break;

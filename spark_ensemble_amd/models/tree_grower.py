@@ -181,7 +181,10 @@ def grow_tree(
         # ----- histograms for this level ---------------------------------
         if hists is None:
             # root level: build everything
-            new_h = ops.hist_build(bins, gh, row_idx, offsets, B, D, gh_max)
+            new_h = ops.hist_build(
+                bins, gh, row_idx, offsets, B, D, gh_max,
+                identity_rows=(row_mask is None and bins.is_cuda),
+            )
             if comm is not None:
                 comm.all_reduce_(new_h)
             hists = new_h

@@ -82,7 +82,7 @@ def bin_features(x, edges):
     return reference.bin_features(x, edges)
 
 
-def hist_build(bins, gh, row_idx, node_offsets, num_bins, d_dims=-1, max_abs=None):
+def hist_build(bins, gh, row_idx, node_offsets, num_bins, d_dims=-1, max_abs=None, identity_rows=False):
     """Per-(node, feature, bin) channel sums.
 
     Channel contract (tree_grower.py): gh[:, :d_dims] are SIGNED gradient
@@ -113,6 +113,7 @@ def hist_build(bins, gh, row_idx, node_offsets, num_bins, d_dims=-1, max_abs=Non
                 int(num_bins),
                 int(d_dims),
                 max_abs.to(torch.float32),
+                bool(identity_rows),
             )
             return out
     return reference.hist_build(bins, gh, row_idx, node_offsets, num_bins)
