@@ -307,7 +307,26 @@ class GBMClassificationModel(ProbabilisticClassificationModel, _GBMClassifierPar
         return get_classification_loss(self.getLoss(), self._num_classes)
 
     def _margins(self, x: torch.Tensor) -> torch.Tensor:
+        from ..ensemble.utils import packed_forest_margin
+
         out = self._init.predictRaw(x)[:, : self._dim].contiguous()
+        if self._models:
+            # fast path: identity subspaces + tree stages -> one packed
+            # forest_predict kernel per class dimension
+            packed_ok = True
+            for j in range(self._dim):
+                pj = packed_forest_margin(
+                    x, [ms[j] for ms in self._models],
+                    [wts[j] for wts in self._weights],
+                    self._subspaces, x.shape[1],
+                )
+                if pj is None:
+                    packed_ok = False
+                    break
+                out[:, j] += pj
+            if packed_ok:
+                return out
+            out = self._init.predictRaw(x)[:, : self._dim].contiguous()
         for wts, sub, ms in zip(self._weights, self._subspaces, self._models):
             xs = slice_features(x, sub)
             for j, m in enumerate(ms):

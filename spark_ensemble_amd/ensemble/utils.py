@@ -62,3 +62,29 @@ def slice_features(x: torch.Tensor, indices: torch.Tensor) -> torch.Tensor:
         if bool((indices == torch.arange(x.shape[1], device=indices.device)).all()):
             return x
     return x.index_select(1, indices.to(x.device))
+
+
+def packed_forest_margin(x, models, weights, subspaces, num_features):
+    """Batched ensemble inference fast path: when every stage model is a
+    built-in tree and every subspace is identity (the GBM default,
+    subspaceRatio=1), the whole ensemble is ONE forest_predict kernel call
+    instead of a launch per stage.  Returns [N] margins or None when the
+    fast path does not apply."""
+    from ..models.tree import DecisionTreeRegressionModel
+    from ..ops import dispatch as _ops
+
+    if not models:
+        return None
+    for sub in subspaces:
+        if sub is not None and (
+            sub.numel() != num_features
+            or not bool((sub.cpu() == torch.arange(num_features)).all())
+        ):
+            return None
+    trees = []
+    for m in models:
+        if not isinstance(m, DecisionTreeRegressionModel):
+            return None
+        trees.append(m._tree)
+    w = torch.tensor([float(v) for v in weights], dtype=torch.float32)
+    return _ops.forest_predict(x, trees, w).squeeze(1)

@@ -304,8 +304,15 @@ class GBMRegressionModel(RegressionModel, _GBMRegressorParams):
     _subspaces: List[torch.Tensor] = []
 
     def predict(self, features: torch.Tensor) -> torch.Tensor:
+        from ..ensemble.utils import packed_forest_margin
+
         x = features.float()
         out = self._init.predict(x)
+        packed = packed_forest_margin(
+            x, self._models, self._weights, self._subspaces, x.shape[1]
+        )
+        if packed is not None:
+            return out + packed
         for wgt, sub, m in zip(self._weights, self._subspaces, self._models):
             out = out + wgt * m.predict(slice_features(x, sub))
         return out
