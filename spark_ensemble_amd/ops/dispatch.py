@@ -101,6 +101,25 @@ def hist_build(bins, gh, row_idx, node_offsets, num_bins, d_dims=-1, max_abs=Non
             C = gh.shape[1]
             if max_abs is None:
                 max_abs = gh.abs().amax(dim=0).cpu()
+            if C > 8:
+                # wide one-hot targets (K-class gini trees, K > 7): chunk
+                # signed channels into kernel-sized groups, each carrying
+                # the non-negative tail (hess/count) for bookkeeping
+                D = d_dims if d_dims > 0 else C - 1
+                tail = list(range(D, C))
+                parts = []
+                for s in range(0, D, 7):
+                    cols = list(range(s, min(s + 7, D))) + tail
+                    idx = torch.tensor(cols, device=gh.device)
+                    part = hist_build(
+                        bins, gh.index_select(1, idx).contiguous(), row_idx,
+                        node_offsets, num_bins, min(7, D - s),
+                        max_abs[torch.tensor(cols)], identity_rows,
+                    )
+                    parts.append(part[..., : min(7, D - s)])
+                    if s + 7 >= D:
+                        parts.append(part[..., min(7, D - s):])  # tail once
+                return torch.cat(parts, dim=-1)
             out = torch.zeros(
                 n_nodes, F, num_bins, C, dtype=torch.float32, device=bins.device
             )

@@ -363,3 +363,31 @@ def test_tree_fit_deterministic():
     r1 = m1.transform(df)["rawPrediction"]
     r2 = m2.transform(df)["rawPrediction"]
     assert torch.equal(r1, r2)
+
+
+def test_hist_build_wide_multiclass_chunked(hip, ref):
+    """K > 7 one-hot channels exceed the kernel's 8-channel cap; dispatch
+    chunks the signed channels and reassembles (letter-style 26-class
+    trees on GPU go through this path)."""
+    g = torch.Generator().manual_seed(51)
+    n, f, b, d = 20000, 9, 32, 12
+    bins = torch.randint(0, b, (n, f), generator=g, dtype=torch.uint8).to(DEV)
+    gh = torch.cat(
+        [torch.randn(n, d, generator=g), torch.rand(n, 1, generator=g)], dim=1
+    ).to(DEV)
+    rows = torch.arange(n, dtype=torch.int32).to(DEV)
+    offs = torch.tensor([0, 8000, n])
+    got = hip.hist_build(bins, gh, rows, offs, b, d)
+    want = ref.hist_build(bins.cpu(), gh.cpu(), rows.cpu(), offs, b)
+    assert got.shape == (2, f, b, d + 1)
+    assert torch.allclose(got.cpu(), want, atol=2e-2, rtol=1e-3)
+
+
+def test_wide_multiclass_tree_gpu():
+    from spark_ensemble_amd.models import DecisionTreeClassifier
+    from spark_ensemble_amd.utils.io import synthetic_classification
+
+    df = synthetic_classification(40000, 16, k=12, seed=8, device=DEV)
+    m = DecisionTreeClassifier().setMaxDepth(6).fit(df)
+    acc = float((m.transform(df)["prediction"] == df["label"]).float().mean())
+    assert acc > 1.5 / 12, acc
