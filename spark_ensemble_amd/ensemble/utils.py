@@ -66,25 +66,36 @@ def slice_features(x: torch.Tensor, indices: torch.Tensor) -> torch.Tensor:
 
 def packed_forest_margin(x, models, weights, subspaces, num_features):
     """Batched ensemble inference fast path: when every stage model is a
-    built-in tree and every subspace is identity (the GBM default,
-    subspaceRatio=1), the whole ensemble is ONE forest_predict kernel call
-    instead of a launch per stage.  Returns [N] margins or None when the
-    fast path does not apply."""
+    built-in regression tree, the whole ensemble is ONE forest_predict
+    kernel call instead of a launch (plus a feature-slice copy) per
+    stage.  Non-identity subspaces are handled by remapping each tree's
+    split-feature ids back into the ORIGINAL feature space, so the packed
+    forest walks the unsliced x.  Returns [N] margins or None when a
+    stage is not a tree."""
     from ..models.tree import DecisionTreeRegressionModel
     from ..ops import dispatch as _ops
 
     if not models:
         return None
-    for sub in subspaces:
-        if sub is not None and (
-            sub.numel() != num_features
-            or not bool((sub.cpu() == torch.arange(num_features)).all())
-        ):
-            return None
     trees = []
-    for m in models:
+    for m, sub in zip(models, subspaces):
         if not isinstance(m, DecisionTreeRegressionModel):
             return None
-        trees.append(m._tree)
+        t = m._tree
+        identity = sub is None or (
+            sub.numel() == num_features
+            and bool((sub.cpu() == torch.arange(num_features)).all())
+        )
+        if not identity:
+            remapped = getattr(m, "_tree_orig_feats", None)
+            if remapped is None:
+                feat = t["feature"].long()
+                sub_dev = sub.to(feat.device)
+                remapped = torch.where(
+                    feat >= 0, sub_dev[feat.clamp_min(0)], feat.long()
+                ).to(torch.int32)
+                m._tree_orig_feats = remapped
+            t = dict(t, feature=remapped)
+        trees.append(t)
     w = torch.tensor([float(v) for v in weights], dtype=torch.float32)
     return _ops.forest_predict(x, trees, w).squeeze(1)

@@ -108,12 +108,21 @@ class BaggingRegressionModel(RegressionModel, _BaggingRegressorParams):
         return len(self._models)
 
     def predict(self, features: torch.Tensor) -> torch.Tensor:
+        from ..ensemble.utils import packed_forest_margin
+
         x = features.float()
+        mcount = len(self._models)
+        packed = packed_forest_margin(
+            x, self._models, [1.0 / mcount] * mcount, self._subspaces,
+            x.shape[1],
+        )
+        if packed is not None:
+            return packed
         acc = None
         for sub, m in zip(self._subspaces, self._models):
             p = m.predict(slice_features(x, sub))
             acc = p if acc is None else acc + p
-        return acc / len(self._models)
+        return acc / mcount
 
     def _save_impl(self, path: str):
         persistence.save_metadata(
