@@ -140,6 +140,14 @@ def hist_build(bins, gh, row_idx, node_offsets, num_bins, d_dims=-1, max_abs=Non
 
 def split_search(hist, lam=1e-6, min_child_weight=0.0, min_instances=1.0, min_info_gain=0.0, d_dims=-1):
     if hist.is_cuda:
+        c_chk = hist.shape[3]
+        if c_chk > 8:
+            # wide one-hot channels (K > 7 gini trees): eager torch path
+            # (runs on GPU; the fused kernel caps at 8 channels)
+            return reference.split_search(
+                hist, lam, min_child_weight, min_instances, min_info_gain,
+                d_dims,
+            )
         m = _require_hip("split_argmax")
         if m is not None:
             n, f, b, c = hist.shape
