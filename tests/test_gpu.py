@@ -391,3 +391,35 @@ def test_wide_multiclass_tree_gpu():
     m = DecisionTreeClassifier().setMaxDepth(6).fit(df)
     acc = float((m.transform(df)["prediction"] == df["label"]).float().mean())
     assert acc > 1.5 / 12, acc
+
+
+def test_gpu_model_save_load_roundtrip(tmp_path):
+    """Models fit on GPU must save/load (MLlib layout) and reproduce
+    transform outputs exactly after reload."""
+    import spark_ensemble_amd as sea
+    from spark_ensemble_amd.utils.io import synthetic_classification
+
+    df = synthetic_classification(30000, 16, k=2, seed=17, device=DEV)
+    m = sea.GBMClassifier().setLoss("bernoulli").setNumBaseLearners(4).fit(df)
+    p = str(tmp_path / "m")
+    m.save(p)
+    m2 = sea.GBMClassificationModel.load(p)
+    a = m.transform(df)["rawPrediction"].cpu()
+    b = m2.transform(df)["rawPrediction"].cpu()
+    assert torch.allclose(a, b, rtol=1e-6, atol=1e-7)
+
+    bag = sea.BaggingRegressor().setNumBaseLearners(3).setSubspaceRatio(0.6).fit(
+        synthetic_regression_gpu()
+    )
+    p2 = str(tmp_path / "bag")
+    bag.save(p2)
+    bag2 = sea.BaggingRegressionModel.load(p2)
+    x = synthetic_regression_gpu()["features"]
+    assert torch.allclose(bag.predict(x).cpu(), bag2.predict(x).cpu(),
+                          rtol=1e-6, atol=1e-7)
+
+
+def synthetic_regression_gpu():
+    from spark_ensemble_amd.utils.io import synthetic_regression
+
+    return synthetic_regression(20000, 12, seed=23, device=DEV)
