@@ -217,10 +217,11 @@ __global__ void hist_build_kernel(
   for (int i = threadIdx.x; i < lds_cells; i += blockDim.x) lds64[i] = 0ull;
   __syncthreads();
 
-  // bins rows are read 16 (or 2x16) bytes at a time (uint4) when the
-  // group start is 16-aligned and inside the row
-  const bool vec16 = (FG == 16) && (f0 + 16 <= F) && ((F & 15) == 0);
-  const bool vec32 = (FG == 32) && (f0 + 32 <= F) && ((F & 31) == 0);
+  // bins rows are read FG bytes at a time in 16-B uint4 pieces when the
+  // group is 16-aligned and fully inside the row
+  const bool vec = (FG == 16 || FG == 32 || FG == 64) && (f0 + FG <= F) &&
+                   ((F % FG) == 0);
+  const int nh = FG >> 4;
 
   for (int i = threadIdx.x; i < len; i += blockDim.x) {
     const int r = identity_rows ? start + i : row_idx[start + i];
@@ -239,8 +240,7 @@ __global__ void hist_build_kernel(
       const int iv = __float2int_rn(g[DC + c] * scales[DC + c]);
       addend[c] |= (unsigned)iv;
     }
-    if (vec16 || vec32) {
-      const int nh = vec32 ? 2 : 1;
+    if (vec) {
       for (int hh = 0; hh < nh; ++hh) {
         const uint4 bv = *reinterpret_cast<const uint4*>(
             bins + (int64_t)r * F + f0 + 16 * hh);
@@ -353,13 +353,24 @@ void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
   }();
   static const int lds_budget = []() {
     const char* e = getenv("SEA_HIST_LDS");
-    return e ? atoi(e) : 65536;
+    return e ? atoi(e) : 163840;  // up to the full 160 KiB LDS per CU
   }();
-  int FG = std::max<int>(1, std::min<int>(F, lds_budget / (B * CELLS * 8)));
-  if (FG >= 32 && (F % 32) == 0 && (fg_env == 32)) FG = 32;
+  // measured (profiles/r01_hist_probe3): FG=64 + 1024 threads + 128 KiB
+  // LDS (1 block/CU, 16 waves) runs 1.9x FG=16/256 — 64-B row slices
+  // quarter the random-row fetch amplification and the wider block keeps
+  // the LDS atomic pipe fed
+  // big-LDS path measured for CELLS==1 only; CELLS>=2 keeps the proven
+  // 64-KiB / 256-thread configuration
+  const int budget = CELLS == 1 ? lds_budget : std::min(lds_budget, 65536);
+  int FG = std::max<int>(1, std::min<int>(F, budget / (B * CELLS * 8)));
+  if (CELLS == 1 && FG >= 64 && (F % 64) == 0) FG = 64;
+  else if (CELLS == 1 && FG >= 32 && (F % 32) == 0) FG = 32;
   else if (FG >= 16) FG = 16;
   else if (FG >= 8) FG = 8;
   else if (FG >= 4) FG = 4;
+  if (fg_env == 16 || fg_env == 32 || fg_env == 64)
+    FG = std::min(FG, fg_env);
+  const int threads = FG >= 64 ? 1024 : (FG >= 32 ? 512 : 256);
   const int n_groups = (int)ceil_div(F, FG);
 
   // ---- adaptive chunking: target ~resident-grid x OVERSUB blocks --------
@@ -367,14 +378,18 @@ void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
   auto offs = node_offsets.accessor<int64_t, 1>();
   int64_t total_rows = 0;
   for (int nd = 0; nd < n_nodes; ++nd) total_rows += offs[nd + 1] - offs[nd];
-  const int64_t resident = 256 * 2;  // CUs x blocks/CU at 64 KiB LDS
-  // oversubscription factor: fewer chunks amortize the flush better
-  // (probe: 48-chunk variant 1.7x the 144-chunk rate) but cap imbalance;
-  // SEA_HIST_OVERSUB overrides for experiments
-  static const double oversub = []() {
+  const size_t lds_bytes_pre = (size_t)FG * B * CELLS * 8;
+  const int blocks_per_cu =
+      std::max<int>(1, (int)(163840 / std::max<size_t>(1, lds_bytes_pre)));
+  const int64_t resident = (int64_t)256 * blocks_per_cu;
+  // oversubscription: exactly-resident grids measured fastest at 1
+  // block/CU (probe3); modest oversub amortizes imbalance otherwise
+  static const double oversub_env = []() {
     const char* e = getenv("SEA_HIST_OVERSUB");
-    return e ? atof(e) : 1.5;
+    return e ? atof(e) : 0.0;
   }();
+  const double oversub =
+      oversub_env > 0 ? oversub_env : (blocks_per_cu == 1 ? 1.0 : 1.5);
   const int64_t target_chunks = std::max<int64_t>(
       1, (int64_t)(resident * oversub) / std::max(1, n_groups));
   int64_t chunk_rows = std::max<int64_t>(
@@ -425,15 +440,21 @@ void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
   }
 
   auto stream = at::hip::getCurrentHIPStream();
-  const size_t lds_bytes = (size_t)FG * B * CELLS * 8;
+  const size_t lds_bytes = lds_bytes_pre;
 #define HB_LAUNCH(DD, NNN)                                                   \
-  hipLaunchKernelGGL((hist_build_kernel<DD, NNN>),                           \
-                     dim3(n_chunks, n_groups), dim3(256), lds_bytes, stream, \
-                     out.data_ptr<float>(), stage_ptr,                       \
-                     bins.data_ptr<uint8_t>(),                               \
-                     gh.data_ptr<float>(), row_idx.data_ptr<int>(),          \
-                     chunks.data_ptr<int>(), F, B, FG,                       \
-                     identity_rows ? 1 : 0)
+  do {                                                                       \
+    if (lds_bytes > 65536)                                                   \
+      (void)hipFuncSetAttribute(                                             \
+          reinterpret_cast<const void*>(&hist_build_kernel<DD, NNN>),        \
+          hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds_bytes);       \
+    hipLaunchKernelGGL((hist_build_kernel<DD, NNN>),                         \
+                       dim3(n_chunks, n_groups), dim3(threads), lds_bytes,   \
+                       stream, out.data_ptr<float>(), stage_ptr,             \
+                       bins.data_ptr<uint8_t>(),                             \
+                       gh.data_ptr<float>(), row_idx.data_ptr<int>(),        \
+                       chunks.data_ptr<int>(), F, B, FG,                     \
+                       identity_rows ? 1 : 0);                               \
+  } while (0)
   const int key = D * 10 + NN;
   switch (key) {
     case 11: HB_LAUNCH(1, 1); break;
