@@ -10,16 +10,16 @@ constexpr int F = 256;
 constexpr int B = 256;
 constexpr int64_t N = 10'000'000;
 
-template <int FG, int T>
+template <int FG, int T, int CELLS>
 __global__ __launch_bounds__(T) void probe_kernel(
     float* __restrict__ out, const uint8_t* __restrict__ bins,
     const float* __restrict__ gh, int rows_per_block) {
-  extern __shared__ unsigned long long lds64[];  // FG * B
+  extern __shared__ unsigned long long lds64[];  // FG * B * CELLS
   const int fg = blockIdx.y;
   const int f0 = fg * FG;
   const int64_t start = (int64_t)blockIdx.x * rows_per_block;
   const int64_t len = min((int64_t)rows_per_block, N - start);
-  for (int i = threadIdx.x; i < FG * B; i += blockDim.x) lds64[i] = 0ull;
+  for (int i = threadIdx.x; i < FG * B * CELLS; i += blockDim.x) lds64[i] = 0ull;
   __syncthreads();
   for (int64_t i = threadIdx.x; i < len; i += blockDim.x) {
     const int64_t r = start + i;
@@ -37,36 +37,39 @@ __global__ __launch_bounds__(T) void probe_kernel(
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
           const int b = (w[qq] >> (8 * j)) & 0xff;
-          atomicAdd(lds64 + ((16 * q + qq * 4 + j) * B) + b, packed);
+          unsigned long long* cell =
+              lds64 + (((16 * q + qq * 4 + j) * B) + b) * CELLS;
+#pragma unroll
+          for (int c = 0; c < CELLS; ++c) atomicAdd(cell + c, packed);
         }
     }
   }
   __syncthreads();
   float* dst = out + ((int64_t)(blockIdx.x % 32) * F + f0) * B * 2;
   for (int i = threadIdx.x; i < FG * B; i += blockDim.x) {
-    dst[2 * i] = (float)(int)(unsigned)(lds64[i] >> 32);
-    dst[2 * i + 1] = (float)(int)(unsigned)(lds64[i] & 0xFFFFFFFFull);
+    dst[2 * i] = (float)(int)(unsigned)(lds64[i * CELLS] >> 32);
+    dst[2 * i + 1] = (float)(int)(unsigned)(lds64[i * CELLS] & 0xFFFFFFFFull);
   }
 }
 
-template <int FG, int T>
+template <int FG, int T, int CELLS = 1>
 float run(const char* name, uint8_t* bins, float* gh, float* out, int n_chunks) {
   int rows_per_block = (int)((N + n_chunks - 1) / n_chunks);
   dim3 grid(n_chunks, F / FG);
-  size_t lds = (size_t)FG * B * 8;
+  size_t lds = (size_t)FG * B * CELLS * 8;
   hipError_t e = hipFuncSetAttribute(
-      reinterpret_cast<const void*>(&probe_kernel<FG, T>),
+      reinterpret_cast<const void*>(&probe_kernel<FG, T, CELLS>),
       hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
   if (e) printf("(funcattr %s) ", hipGetErrorString(e));
   hipEvent_t a, b;
   hipEventCreate(&a); hipEventCreate(&b);
-  hipLaunchKernelGGL((probe_kernel<FG, T>), grid, dim3(T), lds, 0, out, bins, gh, rows_per_block);
+  hipLaunchKernelGGL((probe_kernel<FG, T, CELLS>), grid, dim3(T), lds, 0, out, bins, gh, rows_per_block);
   hipError_t le = hipGetLastError();
   if (le) { printf("%-36s launch failed: %s\n", name, hipGetErrorString(le)); return -1; }
   hipDeviceSynchronize();
   hipEventRecord(a);
   for (int it = 0; it < 3; ++it)
-    hipLaunchKernelGGL((probe_kernel<FG, T>), grid, dim3(T), lds, 0, out, bins, gh, rows_per_block);
+    hipLaunchKernelGGL((probe_kernel<FG, T, CELLS>), grid, dim3(T), lds, 0, out, bins, gh, rows_per_block);
   hipEventRecord(b);
   hipDeviceSynchronize();
   float ms;
@@ -94,12 +97,14 @@ int main() {
   hipLaunchKernelGGL(fill_kernel, dim3(4096), dim3(256), 0, 0, bins, gh);
   HIP_CHECK(hipDeviceSynchronize());
   run<16, 256>("FG16 T256 (baseline), 768 chunks", bins, gh, out, 768 / 16);
-  run<16, 256>("FG16 T256, 48 chunks", bins, gh, out, 48);
-  run<32, 256>("FG32 T256, 48 chunks", bins, gh, out, 48);
-  run<32, 512>("FG32 T512, 48 chunks", bins, gh, out, 48);
-  run<64, 512>("FG64 T512, 64 chunks", bins, gh, out, 64);
-  run<64, 512>("FG64 T512, 256 chunks", bins, gh, out, 256);
   run<64, 1024>("FG64 T1024, 64 chunks", bins, gh, out, 64);
+  printf("-- CELLS=2 (two u64 atomics per bump; the C=3/4 shape) --\n");
+  run<16, 256, 2>("c2 FG16 T256 64KiB, 32 chunks", bins, gh, out, 32);
+  run<16, 256, 2>("c2 FG16 T256 64KiB, 128 chunks", bins, gh, out, 128);
+  run<16, 512, 2>("c2 FG16 T512 64KiB, 64 chunks", bins, gh, out, 64);
+  run<32, 1024, 2>("c2 FG32 T1024 128KiB, 32 chunks", bins, gh, out, 32);
+  run<32, 512, 2>("c2 FG32 T512 128KiB, 32 chunks", bins, gh, out, 32);
+  run<16, 1024, 2>("c2 FG16 T1024 64KiB, 64 chunks", bins, gh, out, 64);
   printf("done\n");
   return 0;
 }
