@@ -223,82 +223,47 @@ __global__ void hist_build_kernel(
                    ((F % FG) == 0);
   const int nh = FG >> 4;
 
-  // cache the per-channel scales in registers (loop-invariant)
-  float sc[C];
+  for (int i = threadIdx.x; i < len; i += blockDim.x) {
+    const int r = identity_rows ? start + i : row_idx[start + i];
+    const float* g = gh + (int64_t)r * C;
+    // quantize once per row, pack one u64 addend per cell
+    unsigned long long addend[CELLS];
 #pragma unroll
-  for (int c = 0; c < C; ++c) sc[c] = scales[c];
-
-  // 4-row ILP blocking: the row_idx -> gh/bins dependent-gather chain is
-  // latency-bound at deep (shuffled-row) levels; issuing 4 independent
-  // rows' loads per thread keeps ~4x the memory-level parallelism
-  constexpr int RB = 4;
-  for (int64_t i0 = (int64_t)threadIdx.x * RB; i0 < len;
-       i0 += (int64_t)blockDim.x * RB) {
-    int rr[RB];
-    int nr = 0;
+    for (int c = 0; c < CELLS; ++c) addend[c] = 0ull;
 #pragma unroll
-    for (int k = 0; k < RB; ++k) {
-      const int64_t i = i0 + k;
-      if (i < len) {
-        rr[k] = identity_rows ? (int)(start + i) : row_idx[start + i];
-        nr = k + 1;
-      }
+    for (int d = 0; d < DC; ++d) {
+      const int iv = __float2int_rn(g[d] * scales[d]);
+      addend[d] |= ((unsigned long long)(unsigned)iv) << 32;
     }
-    unsigned long long addend[RB][CELLS];
 #pragma unroll
-    for (int k = 0; k < RB; ++k) {
-      if (k >= nr) break;
-      const float* g = gh + (int64_t)rr[k] * C;
-#pragma unroll
-      for (int c = 0; c < CELLS; ++c) addend[k][c] = 0ull;
-#pragma unroll
-      for (int d = 0; d < DC; ++d) {
-        const int iv = __float2int_rn(g[d] * sc[d]);
-        addend[k][d] |= ((unsigned long long)(unsigned)iv) << 32;
-      }
-#pragma unroll
-      for (int c = 0; c < NC; ++c) {
-        const int iv = __float2int_rn(g[DC + c] * sc[DC + c]);
-        addend[k][c] |= (unsigned)iv;
-      }
+    for (int c = 0; c < NC; ++c) {
+      const int iv = __float2int_rn(g[DC + c] * scales[DC + c]);
+      addend[c] |= (unsigned)iv;
     }
     if (vec) {
       for (int hh = 0; hh < nh; ++hh) {
-        uint4 bv[RB];
+        const uint4 bv = *reinterpret_cast<const uint4*>(
+            bins + (int64_t)r * F + f0 + 16 * hh);
+        const unsigned w[4] = {bv.x, bv.y, bv.z, bv.w};
 #pragma unroll
-        for (int k = 0; k < RB; ++k)
-          if (k < nr)
-            bv[k] = *reinterpret_cast<const uint4*>(
-                bins + (int64_t)rr[k] * F + f0 + 16 * hh);
+        for (int q = 0; q < 4; ++q) {
 #pragma unroll
-        for (int k = 0; k < RB; ++k) {
-          if (k >= nr) break;
-          const unsigned w[4] = {bv[k].x, bv[k].y, bv[k].z, bv[k].w};
+          for (int j = 0; j < 4; ++j) {
+            const int b = (w[q] >> (8 * j)) & 0xff;
+            unsigned long long* cell =
+                lds64 + (((16 * hh + q * 4 + j) * B) + b) * CELLS;
 #pragma unroll
-          for (int q = 0; q < 4; ++q) {
-#pragma unroll
-            for (int j = 0; j < 4; ++j) {
-              const int b = (w[q] >> (8 * j)) & 0xff;
-              unsigned long long* cell =
-                  lds64 + (((16 * hh + q * 4 + j) * B) + b) * CELLS;
-#pragma unroll
-              for (int c = 0; c < CELLS; ++c)
-                atomicAdd(cell + c, addend[k][c]);
-            }
+            for (int c = 0; c < CELLS; ++c) atomicAdd(cell + c, addend[c]);
           }
         }
       }
     } else {
+      const uint8_t* br = bins + (int64_t)r * F + f0;
+      for (int f = 0; f < nf; ++f) {
+        const int b = br[f];
+        unsigned long long* cell = lds64 + ((f * B) + b) * CELLS;
 #pragma unroll
-      for (int k = 0; k < RB; ++k) {
-        if (k >= nr) break;
-        const uint8_t* br = bins + (int64_t)rr[k] * F + f0;
-        for (int f = 0; f < nf; ++f) {
-          const int b = br[f];
-          unsigned long long* cell = lds64 + ((f * B) + b) * CELLS;
-#pragma unroll
-          for (int c = 0; c < CELLS; ++c) atomicAdd(cell + c, addend[k][c]);
-        }
+        for (int c = 0; c < CELLS; ++c) atomicAdd(cell + c, addend[c]);
       }
     }
   }
