@@ -1,0 +1,53 @@
+"""Inference (serving) benchmark: batched ensemble transform throughput.
+
+Measures model.predict / predictRaw rows/sec for a trained 100-stage GBM
+and a 50-tree bagging ensemble on 10M x 256 — the packed single-kernel
+forest path (ensemble/utils.packed_forest_margin).
+"""
+import json
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, ".")
+import spark_ensemble_amd as sea
+from spark_ensemble_amd.models import DecisionTreeRegressor
+from spark_ensemble_amd.utils.io import synthetic_classification, synthetic_regression
+
+DEV = "cuda:0"
+
+
+def timeit(fn, reps=5):
+    fn(); torch.cuda.synchronize()
+    t0 = time.time()
+    for _ in range(reps):
+        fn()
+    torch.cuda.synchronize()
+    return (time.time() - t0) / reps
+
+
+def main():
+    n, f = 10_000_000, 256
+    dfc = synthetic_classification(n, f, k=2, seed=3, device=DEV, informative=48)
+    gbm = (sea.GBMClassifier().setLoss("bernoulli").setNumBaseLearners(100)
+           .setBaseLearner(DecisionTreeRegressor().setMaxDepth(8).setMaxBins(256))
+           .fit(dfc))
+    x = dfc["features"]
+    t = timeit(lambda: gbm.predictRaw(x))
+    print(json.dumps({"bench": "infer_gbm100_depth8", "rows": n,
+                      "ms": round(t * 1000, 2),
+                      "rows_per_sec": round(n / t)}))
+
+    dfr = synthetic_regression(n, f, seed=4, device=DEV)
+    bag = (sea.BaggingRegressor().setNumBaseLearners(50).setSubspaceRatio(0.5)
+           .setBaseLearner(DecisionTreeRegressor().setMaxDepth(8).setMaxBins(256))
+           .fit(dfr))
+    t = timeit(lambda: bag.predict(x))
+    print(json.dumps({"bench": "infer_bagging50_subspace", "rows": n,
+                      "ms": round(t * 1000, 2),
+                      "rows_per_sec": round(n / t)}))
+
+
+if __name__ == "__main__":
+    main()
