@@ -73,15 +73,19 @@ def synthetic_classification(
     wm = torch.randn(informative, k, generator=task)
     wi = torch.randn(1, k, generator=task)
 
-    g = torch.Generator().manual_seed(seed * 9176 + 100003 * (split + 1) + rank)
     n_local = n // world + (1 if rank < n % world else 0)
-    x = torch.randn(n_local, f, generator=g)
-    margin = x[:, :informative] @ wm
-    margin = margin + (x[:, 0] * x[:, 1]).unsqueeze(1) * wi
-    margin = margin + noise * torch.randn(n_local, k, generator=g)
+    dev = torch.device(device) if device is not None else torch.device("cpu")
+    # generate directly on the target device (bench-scale CPU randn of a
+    # 10M x 256 shard costs ~10 s; on-GPU it is milliseconds)
+    g = torch.Generator(device=dev).manual_seed(
+        seed * 9176 + 100003 * (split + 1) + rank
+    )
+    x = torch.randn(n_local, f, generator=g, device=dev)
+    wm_d, wi_d = wm.to(dev), wi.to(dev)
+    margin = x[:, :informative] @ wm_d
+    margin = margin + (x[:, 0] * x[:, 1]).unsqueeze(1) * wi_d
+    margin = margin + noise * torch.randn(n_local, k, generator=g, device=dev)
     y = margin.argmax(dim=1).float()
-    if device is not None:
-        x, y = x.to(device), y.to(device)
     return TensorFrame(features=x, label=y)
 
 
@@ -100,12 +104,13 @@ def synthetic_regression(
     informative = min(informative or min(f, 32), f)
     w = torch.randn(informative, generator=task)
 
-    g = torch.Generator().manual_seed(seed * 7919 + 100003 * (split + 1) + rank)
     n_local = n // world + (1 if rank < n % world else 0)
-    x = torch.randn(n_local, f, generator=g)
-    y = x[:, :informative] @ w
+    dev = torch.device(device) if device is not None else torch.device("cpu")
+    g = torch.Generator(device=dev).manual_seed(
+        seed * 7919 + 100003 * (split + 1) + rank
+    )
+    x = torch.randn(n_local, f, generator=g, device=dev)
+    y = x[:, :informative] @ w.to(dev)
     y = y + 0.5 * (x[:, 0] * x[:, 1]) + torch.sin(x[:, 2] * 2.0)
-    y = y + noise * torch.randn(n_local, generator=g)
-    if device is not None:
-        x, y = x.to(device), y.to(device)
+    y = y + noise * torch.randn(n_local, generator=g, device=dev)
     return TensorFrame(features=x, label=y)
