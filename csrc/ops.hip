@@ -24,12 +24,35 @@
 #include <ATen/hip/HIPContext.h>
 #include <hip/hip_runtime.h>
 
+#include <cstring>
 #include <vector>
 
 #define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be on GPU")
 #define CHECK_CONTIG(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
 
 static inline int64_t ceil_div(int64_t a, int64_t b) { return (a + b - 1) / b; }
+
+// Persistent pinned staging for the small per-launch host tables (chunk
+// lists, cursors): a pageable .to(device) blocks the HOST until the copy
+// completes, and the stream-ordered copy waits for every queued kernel —
+// serializing each tree level behind the previous one.  Pinned source +
+// non_blocking copy keeps the host running ahead.  Slots rotate per call
+// site; reuse is safe because every level ends in a host sync (the split
+// fetch) before the slot is touched again.
+static torch::Tensor h2d_async(const void* src, size_t bytes, int slot,
+                               const torch::Device& dev) {
+  static thread_local torch::Tensor pin[6];
+  auto& b = pin[slot];
+  if (!b.defined() || (size_t)b.numel() < bytes)
+    b = torch::empty({(int64_t)std::max<size_t>(bytes, 4096)},
+                     torch::TensorOptions().dtype(torch::kByte)
+                         .pinned_memory(true));
+  std::memcpy(b.data_ptr(), src, bytes);
+  auto d = torch::empty({(int64_t)bytes},
+                        torch::TensorOptions().dtype(torch::kByte).device(dev));
+  d.copy_(b.narrow(0, 0, (int64_t)bytes), /*non_blocking=*/true);
+  return d;
+}
 
 // ---------------------------------------------------------------------------
 // sample_weights: counter-based RNG (splitmix64), one state per row
@@ -425,9 +448,9 @@ void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
   }
   if ((int)chunk_v.size() == 2 * C) return;
   const int n_chunks = (int)((chunk_v.size() - 2 * C) / 4);
-  auto chunks = torch::from_blob(chunk_v.data(), {(int64_t)chunk_v.size()},
-                                 torch::kInt32)
-                    .to(bins.device(), /*non_blocking=*/false);
+  auto chunks_b = h2d_async(chunk_v.data(), chunk_v.size() * 4, 0,
+                            bins.device());
+  auto chunks = chunks_b.view(torch::kInt32);
 
   // i64 integer staging for multi-chunk nodes (order-independent flush =>
   // bitwise-deterministic histograms)
@@ -474,10 +497,9 @@ void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
   }
 #undef HB_LAUNCH
   if (!multi_nodes.empty()) {
-    auto nodes_t = torch::from_blob(multi_nodes.data(),
-                                    {(int64_t)multi_nodes.size()},
-                                    torch::kInt32)
-                       .to(bins.device(), /*non_blocking=*/false);
+    auto nodes_b = h2d_async(multi_nodes.data(), multi_nodes.size() * 4, 1,
+                             bins.device());
+    auto nodes_t = nodes_b.view(torch::kInt32);
     const int FBC = F * B * C;
     const int dblocks = (int)std::min<int64_t>(ceil_div(FBC, 256), 1024);
     hipLaunchKernelGGL(hist_decode_kernel,
@@ -600,12 +622,12 @@ void partition_rows(torch::Tensor new_rows, torch::Tensor left_counts,
     }
   }
   auto stream = at::hip::getCurrentHIPStream();
-  auto cursors = torch::from_blob(cur_v.data(), {n_nodes * 2}, torch::kInt32)
-                     .to(bins.device());
+  auto cursors_b = h2d_async(cur_v.data(), cur_v.size() * 4, 2, bins.device());
+  auto cursors = cursors_b.view(torch::kInt32);
   if (!chunk_v.empty()) {
-    auto chunks = torch::from_blob(chunk_v.data(), {(int64_t)chunk_v.size()},
-                                   torch::kInt32)
-                      .to(bins.device());
+    auto chunks_b = h2d_async(chunk_v.data(), chunk_v.size() * 4, 3,
+                              bins.device());
+    auto chunks = chunks_b.view(torch::kInt32);
     const int n_chunks = (int)(chunk_v.size() / 3);
     const size_t bit_lds = (size_t)((chunk_rows + 63) / 64) * 8;
     hipLaunchKernelGGL(partition_kernel, dim3(n_chunks), dim3(256), bit_lds,
