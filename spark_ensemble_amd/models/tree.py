@@ -142,18 +142,73 @@ class _TreeModelMixin:
         self._num_features = meta.get("numFeatures", -1)
 
 
+
+
+def _fused_fit_stats(grad, w, comm):
+    """Single fused stats pass for a tree fit prologue: ONE host sync and
+    (when distributed) one sum + one max collective replace the five
+    separate `.item()` syncs (total weight, zero-weight check, unit-weight
+    check, per-channel quantization maxima, root totals)."""
+    d = grad.shape[1]
+    n = grad.shape[0]
+    dev = grad.device
+    sums = torch.cat([
+        w.sum().reshape(1),
+        (w == 0).sum().reshape(1).to(torch.float32),
+        (w != 1).sum().reshape(1).to(torch.float32),
+        torch.tensor([float(n)], device=dev),
+        grad.sum(dim=0).reshape(-1),
+    ])
+    maxs = torch.cat([
+        grad.abs().amax(dim=0).reshape(-1) if n else torch.zeros(d, device=dev),
+        w.abs().amax().reshape(1) if n else torch.zeros(1, device=dev),
+    ])
+    if comm.is_distributed:
+        comm.all_reduce_(sums)
+        comm.all_reduce_(maxs, "max")
+    sums_c = sums.cpu()
+    maxs_c = maxs.cpu()
+    return {
+        "total_w": float(sums_c[0]),
+        "has_zero": float(sums_c[1]) > 0,
+        "all_one": float(sums_c[2]) == 0,
+        "n_global": float(sums_c[3]),
+        "gsum": sums_c[4:],
+        "gmax": maxs_c[:d],
+        "hmax": float(maxs_c[d]),
+    }
+
+
+def _grow_args(st):
+    """(hess_is_count, gh_max, root_tot) for grow_tree from fused stats;
+    root totals only valid when every row participates (no zero-weight
+    mask)."""
+    d = st["gmax"].numel()
+    if st["all_one"]:
+        gh_max = torch.cat([st["gmax"], torch.tensor([st["hmax"]])])
+        root_tot = torch.cat([st["gsum"],
+                              torch.tensor([st["total_w"]])])
+    else:
+        gh_max = torch.cat([st["gmax"], torch.tensor([st["hmax"], 1.0])])
+        root_tot = torch.cat([st["gsum"],
+                              torch.tensor([st["total_w"], st["n_global"]])])
+    return st["all_one"], gh_max, (None if st["has_zero"] else root_tot)
+
+
 class DecisionTreeRegressor(Regressor, _TreeParams):
     def _fit(self, dataset: TensorFrame) -> "DecisionTreeRegressionModel":
         x, y, w = self._extract_xyw(dataset)
         edges, bins = ensure_binned(dataset, x, self.getOrDefault("maxBins"))
         comm = get_comm()
         grad = (w * y).unsqueeze(1)
-        total_w = comm.all_reduce_scalar(float(w.sum()))
-        gp = self._grow_params(total_w)
-        mask = w > 0 if bool((w == 0).any()) else None
+        st = _fused_fit_stats(grad, w, comm)
+        gp = self._grow_params(st["total_w"])
+        mask = w > 0 if st["has_zero"] else None
+        hic, gh_max, root_tot = _grow_args(st)
         tp_out: list = []
         tree = grow_tree(bins, edges, grad, w, gp, comm, row_mask=mask,
-                         train_pred_out=tp_out)
+                         train_pred_out=tp_out, hess_is_count=hic,
+                         gh_max_in=gh_max, root_tot_in=root_tot)
         model = DecisionTreeRegressionModel()
         model._set_tree(tree, x.shape[1])
         # training-row predictions captured during growth (leaf scatter
@@ -185,10 +240,13 @@ class DecisionTreeClassifier(ProbabilisticClassifier, _TreeParams):
         onehot = torch.zeros(x.shape[0], k, dtype=torch.float32, device=x.device)
         onehot.scatter_(1, y.long().unsqueeze(1), 1.0)
         grad = onehot * w.unsqueeze(1)
-        total_w = comm.all_reduce_scalar(float(w.sum()))
-        gp = self._grow_params(total_w)
-        mask = w > 0 if bool((w == 0).any()) else None
-        tree = grow_tree(bins, edges, grad, w, gp, comm, row_mask=mask)
+        st = _fused_fit_stats(grad, w, comm)
+        gp = self._grow_params(st["total_w"])
+        mask = w > 0 if st["has_zero"] else None
+        hic, gh_max, root_tot = _grow_args(st)
+        tree = grow_tree(bins, edges, grad, w, gp, comm, row_mask=mask,
+                         hess_is_count=hic, gh_max_in=gh_max,
+                         root_tot_in=root_tot)
         model = DecisionTreeClassificationModel()
         model._set_tree(tree, x.shape[1])
         model._num_classes = k

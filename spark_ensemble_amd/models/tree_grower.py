@@ -77,6 +77,8 @@ def grow_tree(
     row_mask: Optional[torch.Tensor] = None,
     hess_is_count: Optional[bool] = None,
     train_pred_out: Optional[list] = None,
+    gh_max_in: Optional[torch.Tensor] = None,
+    root_tot_in: Optional[torch.Tensor] = None,
 ) -> Dict[str, torch.Tensor]:
     """Grow one tree; returns flat node arrays:
 
@@ -110,8 +112,12 @@ def grow_tree(
     C = gh.shape[1]
     idx_c = C - 1
     # per-channel abs maxima for the kernel's fixed-point quantization
-    # (computed ONCE per fit; one device sync amortized over all levels)
-    gh_max = gh.abs().amax(dim=0).cpu() if bins.is_cuda else None
+    # (callers pass precomputed values from their fused stats pass; the
+    # fallback costs one device sync per fit)
+    if gh_max_in is not None:
+        gh_max = gh_max_in
+    else:
+        gh_max = gh.abs().amax(dim=0).cpu() if bins.is_cuda else None
 
     if row_mask is not None:
         row_idx = row_mask.nonzero(as_tuple=True)[0].to(torch.int32)
@@ -134,14 +140,18 @@ def grow_tree(
 
     root = alloc_nodes(1)
 
-    # ----- root totals (one tiny all-reduce) -----------------------------
-    if row_idx.numel() == N:
-        root_tot = gh.sum(dim=0)
+    # ----- root totals (one tiny all-reduce; skipped when the caller's
+    # fused stats pass already reduced them) ------------------------------
+    if root_tot_in is not None:
+        root_tot = root_tot_in
     else:
-        root_tot = gh.index_select(0, row_idx.long()).sum(dim=0)
-    if comm is not None:
-        comm.all_reduce_(root_tot)
-    root_tot = root_tot.cpu()
+        if row_idx.numel() == N:
+            root_tot = gh.sum(dim=0)
+        else:
+            root_tot = gh.index_select(0, row_idx.long()).sum(dim=0)
+        if comm is not None:
+            comm.all_reduce_(root_tot)
+        root_tot = root_tot.cpu()
 
     # active level state
     node_ids = [root]
