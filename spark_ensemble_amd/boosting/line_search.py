@@ -164,7 +164,8 @@ def _newton_1d(loss, label, pred, direction, weight, comm, max_iter, tol,
                         want_hess=True)
         if comm is not None and comm.is_distributed:
             comm.all_reduce_(payload)
-        return float(payload[0]), float(payload[1]), float(payload[2])
+        pc = payload.cpu()  # one sync for all three scalars
+        return float(pc[0]), float(pc[1]), float(pc[2])
 
     a = 1.0  # natural stage weight
     blo, bhi = lo, hi

@@ -346,13 +346,26 @@ def grow_tree(
     }
     if capture and pred_rows:
         rows_cat = torch.cat(pred_rows).long()
-        lv_mat = torch.stack([v.reshape(D) for v in pred_vals]).to(device)
-        counts = torch.tensor([r.numel() for r in pred_rows], device=device)
+        lv_mat = _to_dev_async(
+            torch.stack([v.reshape(D) for v in pred_vals]), device
+        )
+        counts = _to_dev_async(
+            torch.tensor([r.numel() for r in pred_rows]), device
+        )
         vals = torch.repeat_interleave(lv_mat, counts, dim=0)
         tp = torch.zeros(N, D, dtype=torch.float32, device=device)
         tp[rows_cat] = vals
         train_pred_out.append(tp)
-    return {k: v.to(device) for k, v in tree.items()}
+    return {k: _to_dev_async(v, device) for k, v in tree.items()}
+
+
+def _to_dev_async(t: torch.Tensor, device) -> torch.Tensor:
+    """Pinned + non_blocking H2D for small host arrays: a pageable
+    .to(device) makes the host WAIT for every queued kernel before the
+    copy; async staging keeps the host running ahead."""
+    if device is None or torch.device(device).type != "cuda":
+        return t.to(device) if device is not None else t
+    return t.pin_memory().to(device, non_blocking=True)
 
 
 def _finalize_leaves(node_ids, totals, leaves, params: GrowParams, D: int):
