@@ -382,18 +382,20 @@ void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
   // LDS (1 block/CU, 16 waves) runs 1.9x FG=16/256 — 64-B row slices
   // quarter the random-row fetch amplification and the wider block keeps
   // the LDS atomic pipe fed
-  // big-LDS path measured for CELLS==1 only; CELLS>=2 keeps the proven
-  // 64-KiB / 256-thread configuration
-  const int budget = CELLS == 1 ? lds_budget : std::min(lds_budget, 65536);
+  // probe-measured optima (profiles/r01_hist_probe3): CELLS==1 ->
+  // FG=64 / 1024 threads / 128 KiB; CELLS==2 -> FG=32 / 1024 / 128 KiB;
+  // CELLS>=3 (wide multiclass) keeps the conservative 64-KiB config
+  const int budget = CELLS <= 2 ? lds_budget : std::min(lds_budget, 65536);
   int FG = std::max<int>(1, std::min<int>(F, budget / (B * CELLS * 8)));
   if (CELLS == 1 && FG >= 64 && (F % 64) == 0) FG = 64;
-  else if (CELLS == 1 && FG >= 32 && (F % 32) == 0) FG = 32;
+  else if (CELLS <= 2 && FG >= 32 && (F % 32) == 0) FG = 32;
   else if (FG >= 16) FG = 16;
   else if (FG >= 8) FG = 8;
   else if (FG >= 4) FG = 4;
   if (fg_env == 16 || fg_env == 32 || fg_env == 64)
     FG = std::min(FG, fg_env);
-  const int threads = FG >= 64 ? 1024 : (FG >= 32 ? 512 : 256);
+  const int threads =
+      (FG >= 64 || (CELLS == 2 && FG >= 32)) ? 1024 : (FG >= 32 ? 512 : 256);
   const int n_groups = (int)ceil_div(F, FG);
 
   // ---- adaptive chunking: target ~resident-grid x OVERSUB blocks --------
