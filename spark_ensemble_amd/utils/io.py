@@ -114,3 +114,53 @@ def synthetic_regression(
     y = y + 0.5 * (x[:, 0] * x[:, 1]) + torch.sin(x[:, 2] * 2.0)
     y = y + noise * torch.randn(n_local, generator=g, device=dev)
     return TensorFrame(features=x, label=y)
+
+
+def from_pandas(df, features_cols=None, label_col="label", weight_col=None,
+                device=None) -> TensorFrame:
+    """Build a TensorFrame from a pandas DataFrame (the on-ramp for users
+    coming from Spark DataFrames).  ``features_cols`` defaults to every
+    numeric column except the label/weight."""
+    import pandas as pd  # noqa: F401
+
+    if features_cols is None:
+        features_cols = [
+            c for c in df.columns
+            if c not in (label_col, weight_col)
+            and np.issubdtype(df[c].dtype, np.number)
+        ]
+    x = torch.from_numpy(
+        np.ascontiguousarray(df[features_cols].to_numpy(dtype=np.float32))
+    )
+    cols = {"features": x}
+    if label_col in df.columns:
+        cols["label"] = torch.from_numpy(
+            df[label_col].to_numpy(dtype=np.float32)
+        )
+    if weight_col and weight_col in df.columns:
+        cols["weight"] = torch.from_numpy(
+            df[weight_col].to_numpy(dtype=np.float32)
+        )
+    fr = TensorFrame(cols)
+    return fr.to(device) if device is not None else fr
+
+
+def read_parquet(path: str, features_cols=None, label_col="label",
+                 weight_col=None, device=None) -> TensorFrame:
+    """Load a parquet file into a TensorFrame (pyarrow)."""
+    import pyarrow.parquet as pq
+
+    table = pq.read_table(path)
+    return from_pandas(
+        table.to_pandas(), features_cols, label_col, weight_col, device
+    )
+
+
+def read_csv(path: str, features_cols=None, label_col="label",
+             weight_col=None, device=None, **kw) -> TensorFrame:
+    """Load a CSV file into a TensorFrame (pandas reader kwargs pass
+    through)."""
+    import pandas as pd
+
+    return from_pandas(pd.read_csv(path, **kw), features_cols, label_col,
+                       weight_col, device)

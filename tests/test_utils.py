@@ -77,3 +77,37 @@ def test_slice_features_gather():
     assert out.tolist() == [[0.0, 2.0], [4.0, 6.0], [8.0, 10.0]]
     # identity short-circuit
     assert slice_features(x, torch.arange(4)) is x
+
+
+def test_pandas_parquet_csv_interop(tmp_path):
+    import numpy as np
+    import pandas as pd
+    import torch
+
+    import spark_ensemble_amd as sea
+    from spark_ensemble_amd.utils.io import from_pandas, read_csv, read_parquet
+
+    rng = np.random.default_rng(3)
+    df = pd.DataFrame({
+        "f0": rng.normal(size=200), "f1": rng.normal(size=200),
+        "label": rng.integers(0, 2, size=200).astype(float),
+        "name": ["x"] * 200,  # non-numeric column must be ignored
+    })
+    fr = from_pandas(df)
+    assert fr["features"].shape == (200, 2)
+    assert fr["label"].dtype == torch.float32
+
+    df.drop(columns=["name"]).to_csv(tmp_path / "d.csv", index=False)
+    fr2 = read_csv(str(tmp_path / "d.csv"))
+    assert torch.allclose(fr2["features"], fr["features"], atol=1e-6)
+
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    pq.write_table(pa.Table.from_pandas(df.drop(columns=["name"])),
+                   tmp_path / "d.parquet")
+    fr3 = read_parquet(str(tmp_path / "d.parquet"))
+    assert torch.allclose(fr3["features"], fr["features"], atol=1e-6)
+
+    # and a model fits straight off it
+    m = sea.GBMClassifier().setNumBaseLearners(2).fit(fr)
+    assert "prediction" in m.transform(fr)
