@@ -120,6 +120,14 @@ class BaggingClassificationModel(
     def numModels(self):
         return len(self._models)
 
+    @property
+    def featureImportances(self):
+        from ..ensemble.utils import ensemble_feature_importances
+
+        return ensemble_feature_importances(
+            self._models, [1.0] * len(self._models), self._subspaces, self._num_features
+        )
+
     def predictRaw(self, features: torch.Tensor) -> torch.Tensor:
         x = features.float()
         k = self._num_classes

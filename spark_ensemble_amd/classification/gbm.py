@@ -360,6 +360,16 @@ class GBMClassificationModel(ProbabilisticClassificationModel, _GBMClassifierPar
     def numModels(self) -> int:
         return len(self._models)
 
+    @property
+    def featureImportances(self):
+        from ..ensemble.utils import ensemble_feature_importances
+
+        ms, ws, subs = [], [], []
+        for mrow, wrow, sub in zip(self._models, self._weights, self._subspaces):
+            for m, w in zip(mrow, wrow):
+                ms.append(m); ws.append(w); subs.append(sub)
+        return ensemble_feature_importances(ms, ws, subs, self._num_features)
+
     # -- persistence: two-level model-$i-$k nesting (reference :621-634) ---
     def _save_impl(self, path: str):
         persistence.save_metadata(

@@ -99,3 +99,20 @@ def packed_forest_margin(x, models, weights, subspaces, num_features):
         trees.append(t)
     w = torch.tensor([float(v) for v in weights], dtype=torch.float32)
     return _ops.forest_predict(x, trees, w).squeeze(1)
+
+
+def ensemble_feature_importances(models, weights, subspaces, num_features):
+    """Weighted, subspace-mapped aggregate of member featureImportances
+    (normalized to sum 1; zeros when no member exposes importances)."""
+    agg = torch.zeros(num_features, dtype=torch.float64)
+    for m, w, sub in zip(models, weights, subspaces):
+        fi = getattr(m, "featureImportances", None)
+        if fi is None:
+            continue
+        fi = fi.double() * abs(float(w))
+        if sub is None or fi.numel() == num_features:
+            agg[: fi.numel()] += fi
+        else:
+            agg.index_add_(0, sub.cpu().long(), fi)
+    t = float(agg.sum())
+    return (agg / t if t > 0 else agg).float()

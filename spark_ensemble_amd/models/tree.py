@@ -99,6 +99,15 @@ class _TreeParams(Params):
 class _TreeModelMixin:
     """Shared storage/persistence for fitted tree models."""
 
+    @property
+    def featureImportances(self):
+        """Gain-based normalized feature importances (MLlib
+        ``featureImportances`` analog): per-feature sums of split gains."""
+        fi = self._tree.get("feature_importance")
+        if fi is None:
+            fi = torch.zeros(self._num_features)
+        return fi.cpu()
+
     def _set_tree(self, tree: dict, num_features: int):
         self._tree = tree
         self._num_features = num_features

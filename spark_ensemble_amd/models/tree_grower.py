@@ -139,6 +139,9 @@ def grow_tree(
         return start
 
     root = alloc_nodes(1)
+    # gain-based feature importances (MLlib featureImportances analog):
+    # sum of split gains per feature, normalized at the end
+    fi = torch.zeros(F, dtype=torch.float64)
 
     # ----- root totals (one tiny all-reduce; skipped when the caller's
     # fused stats pass already reduced them) ------------------------------
@@ -275,6 +278,7 @@ def grow_tree(
                 t = int(b_cpu[i])
                 nid = node_ids[i]
                 feats[nid] = f
+                fi[f] += float(gain_cpu[i])
                 thrs[nid] = float(edges_cpu[f, t])
                 cid = alloc_nodes(2)
                 lefts[nid] = cid
@@ -338,11 +342,15 @@ def grow_tree(
     for i, lv in enumerate(leaves):
         if lv is not None:
             leaf_value[i] = lv
+    fi_tot = float(fi.sum())
     tree = {
         "feature": torch.tensor(feats, dtype=torch.int32),
         "threshold": torch.tensor(thrs, dtype=torch.float32),
         "left_child": torch.tensor(lefts, dtype=torch.int32),
         "leaf_value": leaf_value,
+        "feature_importance": (fi / fi_tot if fi_tot > 0 else fi).to(
+            torch.float32
+        ),
     }
     if capture and pred_rows:
         rows_cat = torch.cat(pred_rows).long()

@@ -107,6 +107,14 @@ class BaggingRegressionModel(RegressionModel, _BaggingRegressorParams):
     def numModels(self):
         return len(self._models)
 
+    @property
+    def featureImportances(self):
+        from ..ensemble.utils import ensemble_feature_importances
+
+        return ensemble_feature_importances(
+            self._models, [1.0] * len(self._models), self._subspaces, self._num_features
+        )
+
     def predict(self, features: torch.Tensor) -> torch.Tensor:
         from ..ensemble.utils import packed_forest_margin
 

@@ -321,6 +321,14 @@ class GBMRegressionModel(RegressionModel, _GBMRegressorParams):
     def numModels(self) -> int:
         return len(self._models)
 
+    @property
+    def featureImportances(self):
+        from ..ensemble.utils import ensemble_feature_importances
+
+        return ensemble_feature_importances(
+            self._models, self._weights, self._subspaces, self._num_features
+        )
+
     # -- persistence (reference GBMRegressor.scala:563-605 layout) ---------
     def _save_impl(self, path: str):
         persistence.save_metadata(

@@ -83,3 +83,32 @@ def test_tree_roundtrip_persistence(tmp_path, clf_frame):
     out2 = m2.transform(clf_frame)
     assert torch.equal(out1["prediction"], out2["prediction"])
     assert torch.allclose(out1["probability"], out2["probability"])
+
+
+def test_feature_importances():
+    import torch
+
+    import spark_ensemble_amd as sea
+    from spark_ensemble_amd.utils.io import synthetic_regression
+
+    # only feature 0 is informative -> importances concentrate there
+    g = torch.Generator().manual_seed(5)
+    x = torch.randn(4000, 6, generator=g)
+    y = 3.0 * x[:, 0] + 0.01 * torch.randn(4000, generator=g)
+    from spark_ensemble_amd.frame import TensorFrame
+
+    df = TensorFrame(features=x, label=y)
+    t = sea.DecisionTreeRegressor().setMaxDepth(4).fit(df)
+    fi = t.featureImportances
+    assert abs(float(fi.sum()) - 1.0) < 1e-5
+    assert float(fi[0]) > 0.9, fi
+
+    gbm = sea.GBMRegressor().setNumBaseLearners(3).fit(df)
+    fig = gbm.featureImportances
+    assert abs(float(fig.sum()) - 1.0) < 1e-5
+    assert float(fig[0]) > 0.9, fig
+
+    bag = sea.BaggingRegressor().setNumBaseLearners(4).setSubspaceRatio(0.8).fit(df)
+    fib = bag.featureImportances
+    assert abs(float(fib.sum()) - 1.0) < 1e-4
+    assert int(fib.argmax()) == 0
