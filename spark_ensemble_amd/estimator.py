@@ -21,7 +21,7 @@ import torch
 
 from . import persistence
 from .frame import TensorFrame
-from .params import Params, ParamValidators
+from .params import Params
 
 
 class Identifiable(Params):

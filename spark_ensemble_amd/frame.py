@@ -19,7 +19,7 @@ Conventions:
 
 from __future__ import annotations
 
-from typing import Dict, Iterator, Optional
+from typing import Dict, Optional
 
 import torch
 

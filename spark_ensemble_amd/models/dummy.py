@@ -10,9 +10,6 @@ RCCL scalar all-reduce (utils/stats.py).
 
 from __future__ import annotations
 
-import os
-from typing import Optional
-
 import torch
 
 from .. import persistence

@@ -16,8 +16,6 @@ reference's treeAggregate at GBMClassifier.scala:423-427.
 
 from __future__ import annotations
 
-from typing import Optional
-
 import numpy as np
 import torch
 
