@@ -167,6 +167,15 @@ def grow_tree(
 
     edges_cpu = edges.cpu()
 
+    import os as _os
+    import time as _time
+    _prof = _os.environ.get("SEA_GROW_PROF") == "1"
+    _t = {}
+
+    def _tick(name, t0):
+        if _prof:
+            _t[name] = _t.get(name, 0.0) + (_time.perf_counter() - t0)
+
     # leaf-wise (rows, value) capture for train_pred_out
     pred_rows: List[torch.Tensor] = []
     pred_vals: List[torch.Tensor] = []
@@ -191,6 +200,7 @@ def grow_tree(
             _capture_leaves(range(n_active), offsets, row_idx, node_ids)
             break
 
+        _t0 = _time.perf_counter() if _prof else 0.0
         # ----- histograms for this level ---------------------------------
         if hists is None:
             # root level: build everything
@@ -234,6 +244,8 @@ def grow_tree(
                     child_hists[j] = hists[parent_of[j]] - child_hists[sib]
             hists = child_hists
 
+        _tick("hist_launch", _t0)
+        _t0 = _time.perf_counter() if _prof else 0.0
         # ----- split decision (identical on every rank) ------------------
         gain, feat, b, left_stats = ops.split_search(
             hists,
@@ -250,11 +262,15 @@ def grow_tree(
         if bins.is_cuda:
             pr_async = ops.partition_rows_async(bins, row_idx, offsets, feat, b)
 
+        _tick("split+part_launch", _t0)
+        _t0 = _time.perf_counter() if _prof else 0.0
         gain_cpu = gain.cpu()
         feat_cpu = feat.cpu()
         b_cpu = b.cpu()
         left_stats = left_stats.cpu()
 
+        _tick("sync_fetch", _t0)
+        _t0 = _time.perf_counter() if _prof else 0.0
         do_split = torch.isfinite(gain_cpu)
         ns_idx = (~do_split).nonzero(as_tuple=True)[0]
         if ns_idx.numel():
@@ -286,6 +302,8 @@ def grow_tree(
             else:
                 child_ids.append(None)
 
+        _tick("leaf+record", _t0)
+        _t0 = _time.perf_counter() if _prof else 0.0
         # partition rows of splitting nodes
         if pr_async is not None:
             new_rows, new_offs, _ = ops.partition_rows_finish(*pr_async)
@@ -296,6 +314,8 @@ def grow_tree(
             )
         offs_list = new_offs.tolist()
 
+        _tick("part_finish", _t0)
+        _t0 = _time.perf_counter() if _prof else 0.0
         # ----- next level bookkeeping ------------------------------------
         next_nodes: List[int] = []
         next_off: List[int] = [0]
@@ -335,7 +355,10 @@ def grow_tree(
         hists = hists.index_select(0, hist_keep)
         parent_of = nb_parent
         built_mask = nb_mask
+        _tick("bookkeeping", _t0)
 
+    if _prof:
+        print("[grow prof]", {k: round(v * 1000, 2) for k, v in _t.items()})
     # assemble arrays
     n_nodes = len(feats)
     leaf_value = torch.zeros(n_nodes, D, dtype=torch.float32)
