@@ -255,6 +255,8 @@ def grow_tree(
             params.min_info_gain,
             d_dims=D,
         )
+        _tick("split_launch", _t0)
+        _t0 = _time.perf_counter() if _prof else 0.0
         # overlap: on GPU launch the partition straight off the device
         # split outputs, BEFORE the host fetch below drains the stream —
         # one device sync per level instead of two
@@ -262,7 +264,7 @@ def grow_tree(
         if bins.is_cuda:
             pr_async = ops.partition_rows_async(bins, row_idx, offsets, feat, b)
 
-        _tick("split+part_launch", _t0)
+        _tick("part_launch", _t0)
         _t0 = _time.perf_counter() if _prof else 0.0
         gain_cpu = gain.cpu()
         feat_cpu = feat.cpu()
