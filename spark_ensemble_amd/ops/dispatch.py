@@ -179,21 +179,32 @@ def partition_rows_async(bins, row_idx, node_offsets, feat, thr):
     """Launch the partition kernel WITHOUT syncing (feat/thr may be live
     device tensors straight from split_argmax); call
     :func:`partition_rows_finish` after the caller's own device sync."""
+    import time as _time
+
+    _prof = os.environ.get("SEA_GROW_PROF") == "1"
+    _t0 = _time.perf_counter() if _prof else 0.0
     m = _require_hip("partition_rows")
     n = node_offsets.numel() - 1
     offs_cpu = node_offsets.to(torch.int64).cpu()
     row_idx = row_idx.to(torch.int32)
     new_rows = torch.empty_like(row_idx)
     left_counts = torch.zeros(n, dtype=torch.int32, device=bins.device)
+    f32 = feat.to(torch.int32).to(bins.device)
+    t32 = thr.to(torch.int32).to(bins.device)
+    if _prof:
+        _t1 = _time.perf_counter()
     m.partition_rows(
         new_rows,
         left_counts,
         bins,
         row_idx,
         offs_cpu,
-        feat.to(torch.int32).to(bins.device),
-        thr.to(torch.int32).to(bins.device),
+        f32,
+        t32,
     )
+    if _prof:
+        _t2 = _time.perf_counter()
+        print(f"[part prof] prep={(_t1-_t0)*1000:.2f} native={(_t2-_t1)*1000:.2f}")
     return new_rows, left_counts, offs_cpu
 
 
