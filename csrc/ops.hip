@@ -24,6 +24,7 @@
 #include <ATen/hip/HIPContext.h>
 #include <hip/hip_runtime.h>
 
+#include <chrono>
 #include <cstring>
 #include <vector>
 
@@ -624,12 +625,16 @@ void partition_rows(torch::Tensor new_rows, torch::Tensor left_counts,
     }
   }
   auto stream = at::hip::getCurrentHIPStream();
+  static const bool part_dbg = getenv("SEA_PART_DEBUG") != nullptr;
+  auto tp0 = std::chrono::steady_clock::now();
   auto cursors_b = h2d_async(cur_v.data(), cur_v.size() * 4, 2, bins.device());
   auto cursors = cursors_b.view(torch::kInt32);
+  auto tp1 = std::chrono::steady_clock::now();
   if (!chunk_v.empty()) {
     auto chunks_b = h2d_async(chunk_v.data(), chunk_v.size() * 4, 3,
                               bins.device());
     auto chunks = chunks_b.view(torch::kInt32);
+    auto tp2 = std::chrono::steady_clock::now();
     const int n_chunks = (int)(chunk_v.size() / 3);
     const size_t bit_lds = (size_t)((chunk_rows + 63) / 64) * 8;
     hipLaunchKernelGGL(partition_kernel, dim3(n_chunks), dim3(256), bit_lds,
@@ -638,6 +643,15 @@ void partition_rows(torch::Tensor new_rows, torch::Tensor left_counts,
                        bins.data_ptr<uint8_t>(), row_idx.data_ptr<int>(),
                        chunks.data_ptr<int>(), feat.data_ptr<int>(),
                        thr.data_ptr<int>(), F);
+    if (part_dbg) {
+      auto tp3 = std::chrono::steady_clock::now();
+      auto us = [](auto a, auto b) {
+        return std::chrono::duration_cast<std::chrono::microseconds>(b - a)
+            .count();
+      };
+      fprintf(stderr, "[part c++] cursors=%ldus chunks=%ldus launch=%ldus\n",
+              (long)us(tp0, tp1), (long)us(tp1, tp2), (long)us(tp2, tp3));
+    }
   }
   auto lcur = cursors.view({n_nodes, 2}).select(1, 0);
   auto seg_start =
