@@ -51,7 +51,11 @@ static torch::Tensor h2d_async(const void* src, size_t bytes, int slot,
   std::memcpy(b.data_ptr(), src, bytes);
   auto d = torch::empty({(int64_t)bytes},
                         torch::TensorOptions().dtype(torch::kByte).device(dev));
-  d.copy_(b.narrow(0, 0, (int64_t)bytes), /*non_blocking=*/true);
+  // raw async copy on the current stream: torch's copy_(non_blocking)
+  // proved to still block the host behind queued kernels here
+  auto stream = at::hip::getCurrentHIPStream();
+  (void)hipMemcpyAsync(d.data_ptr(), b.data_ptr(), bytes,
+                       hipMemcpyHostToDevice, stream);
   return d;
 }
 
