@@ -607,6 +607,7 @@ void partition_rows(torch::Tensor new_rows, torch::Tensor left_counts,
                     torch::Tensor bins, torch::Tensor row_idx,
                     torch::Tensor node_offsets, torch::Tensor feat,
                     torch::Tensor thr) {
+  auto tpA = std::chrono::steady_clock::now();
   CHECK_GPU(new_rows); CHECK_GPU(bins); CHECK_GPU(row_idx);
   CHECK_GPU(feat); CHECK_GPU(thr); CHECK_GPU(left_counts);
   const int F = (int)bins.size(1);
@@ -653,8 +654,10 @@ void partition_rows(torch::Tensor new_rows, torch::Tensor left_counts,
         return std::chrono::duration_cast<std::chrono::microseconds>(b - a)
             .count();
       };
-      fprintf(stderr, "[part c++] cursors=%ldus chunks=%ldus launch=%ldus\n",
-              (long)us(tp0, tp1), (long)us(tp1, tp2), (long)us(tp2, tp3));
+      fprintf(stderr,
+              "[part c++] entry=%ldus cursors=%ldus chunks=%ldus launch=%ldus\n",
+              (long)us(tpA, tp0), (long)us(tp0, tp1), (long)us(tp1, tp2),
+              (long)us(tp2, tp3));
     }
   }
   auto lcur = cursors.view({n_nodes, 2}).select(1, 0);
