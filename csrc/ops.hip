@@ -661,11 +661,13 @@ void partition_rows(torch::Tensor new_rows, torch::Tensor left_counts,
     }
   }
   auto lcur = cursors.view({n_nodes, 2}).select(1, 0);
-  auto seg_start =
-      torch::from_blob(cur_v.data(), {n_nodes, 2}, torch::kInt32)
-          .select(1, 0)
-          .clone()
-          .to(bins.device());
+  // NOTE: this upload was a pageable .to(device) — a SYNCHRONOUS copy that
+  // stalled the host ~1 ms/level behind the queued hist/split kernels
+  // (found via SEA_PART_DEBUG bracketing); pinned async staging fixes it
+  std::vector<int> seg_v(n_nodes);
+  for (int nd = 0; nd < n_nodes; ++nd) seg_v[nd] = (int)offs[nd];
+  auto seg_b = h2d_async(seg_v.data(), seg_v.size() * 4, 4, bins.device());
+  auto seg_start = seg_b.view(torch::kInt32);
   left_counts.copy_(lcur - seg_start);
 }
 
