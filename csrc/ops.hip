@@ -1260,14 +1260,7 @@ void logreg_loss_grad(torch::Tensor payload, torch::Tensor x, torch::Tensor y,
                       torch::Tensor w, torch::Tensor wmat, bool has_bias);
 bool logreg_fused_supported(int64_t F, int64_t K);
 
-void noop_bench(torch::Tensor a, torch::Tensor b, torch::Tensor c,
-                torch::Tensor d, torch::Tensor e, torch::Tensor f,
-                torch::Tensor g) {
-  (void)a; (void)b; (void)c; (void)d; (void)e; (void)f; (void)g;
-}
-
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-  m.def("noop_bench", &noop_bench);
   m.def("logreg_loss_grad", &logreg_loss_grad,
         "single-pass fused logistic loss+gradient");
   m.def("logreg_fused_supported", &logreg_fused_supported);
