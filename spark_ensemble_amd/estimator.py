@@ -93,7 +93,12 @@ class _Reader:
 
 
 class Estimator(PipelineStage):
-    def fit(self, dataset: TensorFrame, params: Optional[dict] = None) -> "Model":
+    def fit(self, dataset: TensorFrame, params: Optional[dict] = None):
+        """Fit on ``dataset``. ``params`` may be a dict of overrides (one
+        model returned) or a LIST of dicts (one model per map — the Spark
+        ``fit(dataset, paramMaps)`` overload)."""
+        if isinstance(params, (list, tuple)):
+            return [self.fit(dataset, pm) for pm in params]
         est = self.copy(params) if params else self
         return est._fit(dataset)
 

@@ -62,3 +62,15 @@ def test_explain_params():
 def test_unknown_param_raises():
     with pytest.raises(AttributeError):
         GBMRegressor().set("bogus", 1)
+
+
+def test_fit_with_param_map_list():
+    import spark_ensemble_amd as sea
+    from spark_ensemble_amd.utils.io import synthetic_regression
+
+    df = synthetic_regression(300, 6, seed=2)
+    models = sea.GBMRegressor().fit(
+        df, [{"numBaseLearners": 1}, {"numBaseLearners": 3}]
+    )
+    assert len(models) == 2
+    assert models[0].numModels == 1 and models[1].numModels == 3
