@@ -423,3 +423,25 @@ def synthetic_regression_gpu():
     from spark_ensemble_amd.utils.io import synthetic_regression
 
     return synthetic_regression(20000, 12, seed=23, device=DEV)
+
+
+def test_gpu_checkpoint_resume(tmp_path):
+    """Resume a half-trained GBM on GPU: saved (CPU) stage models must
+    replay cleanly against device tensors."""
+    import spark_ensemble_amd as sea
+    from spark_ensemble_amd.utils.io import synthetic_regression
+
+    df = synthetic_regression(50000, 16, seed=19, device=DEV)
+
+    def mk(k, ck):
+        e = sea.GBMRegressor().setNumBaseLearners(k).setSeed(4)
+        if ck:
+            e.setCheckpointInterval(2).setCheckpointDir(str(tmp_path / "gk"))
+        return e
+
+    m_full = mk(4, False).fit(df)
+    mk(2, True).fit(df)
+    m_res = mk(4, True).fit(df)
+    a = m_full.predict(df["features"])
+    b = m_res.predict(df["features"])
+    assert torch.allclose(a, b, rtol=1e-4, atol=1e-5)
