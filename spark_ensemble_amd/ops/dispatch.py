@@ -250,10 +250,12 @@ def _forest_predict_hip(m, x, trees, weights):
     # pack all trees into one node-array arena so a single kernel launch
     # walks every (row, tree) pair
     dev = x.device
-    feats = torch.cat([t["feature"].to(torch.int32) for t in trees]).to(dev)
-    thrs = torch.cat([t["threshold"].to(torch.float32) for t in trees]).to(dev)
-    lefts = torch.cat([t["left_child"].to(torch.int32) for t in trees]).to(dev)
-    leaves = torch.cat([t["leaf_value"].to(torch.float32) for t in trees]).to(dev)
+    # per-tree moves BEFORE the cat: a resumed ensemble mixes CPU-loaded
+    # checkpoint stages with device-fitted ones
+    feats = torch.cat([t["feature"].to(dev, torch.int32) for t in trees])
+    thrs = torch.cat([t["threshold"].to(dev, torch.float32) for t in trees])
+    lefts = torch.cat([t["left_child"].to(dev, torch.int32) for t in trees])
+    leaves = torch.cat([t["leaf_value"].to(dev, torch.float32) for t in trees])
     sizes = torch.tensor([t["feature"].numel() for t in trees], dtype=torch.int64)
     offsets = torch.cat([torch.zeros(1, dtype=torch.int64), sizes.cumsum(0)])[:-1]
     offsets = offsets.to(torch.int32).to(dev)
