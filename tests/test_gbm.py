@@ -195,3 +195,16 @@ def test_gbm_init_strategies(reg_frame):
     for strat in ("constant", "zero", "base"):
         gbm = GBMRegressor().setInitStrategy(strat).setNumBaseLearners(2).fit(reg_frame)
         assert torch.isfinite(gbm.predict(reg_frame["features"])).all()
+
+
+def test_gbm_beats_boosting(clf_frame, clf_frame_test):
+    """Reference GBMClassifierSuite.scala:51-87 also asserts GBM beats
+    plain AdaBoost on the same budget."""
+    import spark_ensemble_amd as sea
+
+    gbm = sea.GBMClassifier().setNumBaseLearners(10).setSeed(3).fit(clf_frame)
+    bst = sea.BoostingClassifier().setNumBaseLearners(10).setSeed(3).fit(clf_frame)
+    y = clf_frame_test["label"]
+    acc_g = float((gbm.transform(clf_frame_test)["prediction"] == y).float().mean())
+    acc_b = float((bst.transform(clf_frame_test)["prediction"] == y).float().mean())
+    assert acc_g >= acc_b - 0.02, (acc_g, acc_b)
