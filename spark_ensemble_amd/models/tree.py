@@ -100,6 +100,33 @@ class _TreeModelMixin:
     """Shared storage/persistence for fitted tree models."""
 
     @property
+    def toDebugString(self) -> str:
+        """Human-readable tree dump (MLlib ``toDebugString`` analog)."""
+        t = {k: v.cpu() for k, v in self._tree.items()}
+        feat, thr, left = t["feature"], t["threshold"], t["left_child"]
+        leaf = t["leaf_value"]
+        lines = [
+            f"{type(self).__name__} of depth {self.depth}, "
+            f"{feat.numel()} nodes"
+        ]
+
+        def rec(nid: int, indent: str):
+            f = int(feat[nid])
+            if f < 0:
+                vals = [round(float(v), 6) for v in leaf[nid]]
+                val = vals[0] if len(vals) == 1 else vals
+                lines.append(f"{indent}Predict: {val}")
+                return
+            th = float(thr[nid])
+            lines.append(f"{indent}If (feature {f} <= {th:.6g})")
+            rec(int(left[nid]), indent + " ")
+            lines.append(f"{indent}Else (feature {f} > {th:.6g})")
+            rec(int(left[nid]) + 1, indent + " ")
+
+        rec(0, " ")
+        return "\n".join(lines)
+
+    @property
     def featureImportances(self):
         """Gain-based normalized feature importances (MLlib
         ``featureImportances`` analog): per-feature sums of split gains."""

@@ -113,3 +113,21 @@ def test_feature_importances():
     fib = bag.featureImportances
     assert abs(float(fib.sum()) - 1.0) < 1e-4
     assert int(fib.argmax()) == 0
+
+
+def test_to_debug_string():
+    import torch
+
+    import spark_ensemble_amd as sea
+    from spark_ensemble_amd.frame import TensorFrame
+
+    g = torch.Generator().manual_seed(9)
+    x = torch.randn(500, 3, generator=g)
+    y = (x[:, 1] > 0.2).float()
+    m = sea.DecisionTreeClassifier().setMaxDepth(2).fit(
+        TensorFrame(features=x, label=y)
+    )
+    s = m.toDebugString
+    assert "If (feature 1 <=" in s
+    assert "Predict:" in s
+    assert s.count("Else") == s.count("If")
