@@ -100,6 +100,10 @@ class _TreeModelMixin:
     """Shared storage/persistence for fitted tree models."""
 
     @property
+    def numNodes(self) -> int:
+        return int(self._tree["feature"].numel())
+
+    @property
     def toDebugString(self) -> str:
         """Human-readable tree dump (MLlib ``toDebugString`` analog)."""
         t = {k: v.cpu() for k, v in self._tree.items()}

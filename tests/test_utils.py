@@ -111,3 +111,27 @@ def test_pandas_parquet_csv_interop(tmp_path):
     # and a model fits straight off it
     m = sea.GBMClassifier().setNumBaseLearners(2).fit(fr)
     assert "prediction" in m.transform(fr)
+
+
+def test_libsvm_roundtrip(tmp_path):
+    import torch
+
+    from spark_ensemble_amd.utils.io import load_libsvm
+
+    p = tmp_path / "d.svm"
+    p.write_text(
+        "1 1:0.5 3:2.0\n"
+        "-1 2:1.5\n"
+        "1 1:-1.0 2:0.25 3:4.0\n"
+    )
+    fr = load_libsvm(str(p))
+    assert fr["features"].shape == (3, 3)
+    # +-1 labels normalize to {0, 1}
+    assert set(fr["label"].tolist()) == {0.0, 1.0}
+    assert float(fr["features"][0, 0]) == 0.5
+    assert float(fr["features"][1, 1]) == 1.5
+    assert float(fr["features"][2, 2]) == 4.0
+    # zero-filled absent entries
+    assert float(fr["features"][0, 1]) == 0.0
+    fr4 = load_libsvm(str(p), num_features=5)
+    assert fr4["features"].shape == (3, 5)
