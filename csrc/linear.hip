@@ -75,66 +75,86 @@ __global__ __launch_bounds__(256) void logreg_loss_grad_kernel(
   const int64_t wave_id = (int64_t)blockIdx.x * (blockDim.x >> 6) + wave;
   const int64_t n_waves = (int64_t)gridDim.x * (blockDim.x >> 6);
 
-  for (int64_t r = wave_id; r < n; r += n_waves) {
-    const float* xr = x + r * F;
-    float xf[NC][4];
-    float acc[K];
+  // two rows per iteration: the 6-step shfl_xor margin reductions of the
+  // two rows interleave (independent chains), halving exposed shuffle
+  // latency per row
+  constexpr int RR = 2;
+  for (int64_t r0 = wave_id * RR; r0 < n; r0 += n_waves * RR) {
+    const int nr = (r0 + 1 < n) ? RR : 1;
+    float xf[RR][NC][4];
+    float acc[RR][K];
 #pragma unroll
-    for (int k = 0; k < K; ++k) acc[k] = 0.0f;
+    for (int u = 0; u < RR; ++u)
 #pragma unroll
-    for (int j = 0; j < NC; ++j) {
-      const int f0 = j * 256 + lane * 4;
-      if (f0 + 3 < F) {
-        const float4 v = *reinterpret_cast<const float4*>(xr + f0);
-        xf[j][0] = v.x; xf[j][1] = v.y; xf[j][2] = v.z; xf[j][3] = v.w;
-      } else {
+      for (int k = 0; k < K; ++k) acc[u][k] = 0.0f;
 #pragma unroll
-        for (int c = 0; c < 4; ++c)
-          xf[j][c] = (f0 + c < F) ? xr[f0 + c] : 0.0f;
-      }
+    for (int u = 0; u < RR; ++u) {
+      if (u >= nr) break;
+      const float* xr = x + (r0 + u) * F;
 #pragma unroll
-      for (int c = 0; c < 4; ++c) {
-        const int f = f0 + c;
-        if (f < F) {
+      for (int j = 0; j < NC; ++j) {
+        const int f0 = j * 256 + lane * 4;
+        if (f0 + 3 < F) {
+          const float4 v = *reinterpret_cast<const float4*>(xr + f0);
+          xf[u][j][0] = v.x; xf[u][j][1] = v.y;
+          xf[u][j][2] = v.z; xf[u][j][3] = v.w;
+        } else {
 #pragma unroll
-          for (int k = 0; k < K; ++k)
-            acc[k] = fmaf(xf[j][c], wlds[k * F + f], acc[k]);
+          for (int c = 0; c < 4; ++c)
+            xf[u][j][c] = (f0 + c < F) ? xr[f0 + c] : 0.0f;
+        }
+#pragma unroll
+        for (int c = 0; c < 4; ++c) {
+          const int f = f0 + c;
+          if (f < F) {
+#pragma unroll
+            for (int k = 0; k < K; ++k)
+              acc[u][k] = fmaf(xf[u][j][c], wlds[k * F + f], acc[u][k]);
+          }
         }
       }
     }
-    // cross-lane reduce: every lane ends with the full margin
+    // cross-lane reduce both rows' margins with interleaved chains
 #pragma unroll
-    for (int k = 0; k < K; ++k) {
+    for (int off = 32; off > 0; off >>= 1)
 #pragma unroll
-      for (int off = 32; off > 0; off >>= 1)
-        acc[k] += __shfl_xor(acc[k], off, 64);
-      acc[k] += bias[k];
-    }
-    // softmax loss + gradient scale (identical in all lanes)
-    float m = acc[0];
-#pragma unroll
-    for (int k = 1; k < K; ++k) m = fmaxf(m, acc[k]);
-    float se = 0.0f;
-#pragma unroll
-    for (int k = 0; k < K; ++k) se += __expf(acc[k] - m);
-    const float logz = m + __logf(se);
-    const int yr = y[r];
-    const float wr = w[r];
-    if (lane == 0) loss_acc += wr * (logz - acc[yr]);
-    float g[K];
-#pragma unroll
-    for (int k = 0; k < K; ++k) {
-      g[k] = (__expf(acc[k] - logz) - (k == yr ? 1.0f : 0.0f)) * wr;
-      if (lane == 0) gbias[k] += g[k];
-    }
-    // gradient outer-product accumulate
-#pragma unroll
-    for (int j = 0; j < NC; ++j)
-#pragma unroll
-      for (int c = 0; c < 4; ++c)
+      for (int u = 0; u < RR; ++u)
 #pragma unroll
         for (int k = 0; k < K; ++k)
-          greg[j * 4 + c][k] = fmaf(xf[j][c], g[k], greg[j * 4 + c][k]);
+          acc[u][k] += __shfl_xor(acc[u][k], off, 64);
+#pragma unroll
+    for (int u = 0; u < RR; ++u) {
+      if (u >= nr) break;
+      const int64_t r = r0 + u;
+#pragma unroll
+      for (int k = 0; k < K; ++k) acc[u][k] += bias[k];
+      // softmax loss + gradient scale (identical in all lanes)
+      float m = acc[u][0];
+#pragma unroll
+      for (int k = 1; k < K; ++k) m = fmaxf(m, acc[u][k]);
+      float se = 0.0f;
+#pragma unroll
+      for (int k = 0; k < K; ++k) se += __expf(acc[u][k] - m);
+      const float logz = m + __logf(se);
+      const int yr = y[r];
+      const float wr = w[r];
+      if (lane == 0) loss_acc += wr * (logz - acc[u][yr]);
+      float g[K];
+#pragma unroll
+      for (int k = 0; k < K; ++k) {
+        g[k] = (__expf(acc[u][k] - logz) - (k == yr ? 1.0f : 0.0f)) * wr;
+        if (lane == 0) gbias[k] += g[k];
+      }
+      // gradient outer-product accumulate
+#pragma unroll
+      for (int j = 0; j < NC; ++j)
+#pragma unroll
+        for (int c = 0; c < 4; ++c)
+#pragma unroll
+          for (int k = 0; k < K; ++k)
+            greg[j * 4 + c][k] =
+                fmaf(xf[u][j][c], g[k], greg[j * 4 + c][k]);
+    }
   }
 
   // flush: one atomic per (feature, class) per wave
