@@ -12,13 +12,18 @@ from spark_ensemble_amd.ops import dispatch as ops  # noqa: E402
 
 
 def bench(label, bins, gh, rows, offs, B, iters=3):
+    # per-channel maxima precomputed ONCE, as the tree grower does per fit
+    # (leaving it to the dispatch default re-runs a strided 10M-row amax
+    # per call and used to inflate the C=3 numbers ~2.4x)
+    max_abs = gh.abs().amax(dim=0).cpu()
+    d = max(1, gh.shape[1] - 2)
     torch.cuda.synchronize()
     # warmup
-    out = ops.hist_build(bins, gh, rows, offs, B)
+    out = ops.hist_build(bins, gh, rows, offs, B, d, max_abs)
     torch.cuda.synchronize()
     t0 = time.time()
     for _ in range(iters):
-        out = ops.hist_build(bins, gh, rows, offs, B)
+        out = ops.hist_build(bins, gh, rows, offs, B, d, max_abs)
     torch.cuda.synchronize()
     dt = (time.time() - t0) / iters * 1000
     n_rows = rows.numel()
