@@ -33,10 +33,11 @@ def _is_identity(indices: Optional[torch.Tensor], num_features: int) -> bool:
 class BinnedDataset:
     def __init__(self, x: torch.Tensor, source_frame: Optional[TensorFrame] = None):
         self.x = x
+        self._frame = source_frame
         self._by_bins: Dict[int, Tuple[torch.Tensor, torch.Tensor]] = {}
         # seed from any bins already cached on the source frame
         if source_frame is not None:
-            for mb in (256, 64, 32):
+            for mb in (256, 128, 64, 32):
                 hit = source_frame.cache_get("bins", x, mb)
                 if hit is not None:
                     self._by_bins[mb] = hit
@@ -52,6 +53,13 @@ class BinnedDataset:
                 comm.broadcast_(edges, src=0)
             bins = ops.bin_features(self.x, edges)
             self._by_bins[max_bins] = (edges, bins)
+            # write back to the source frame so REPEATED fits over the same
+            # resident features (CV folds excluded — different tensors) skip
+            # the ~2 s binning; without this every fresh fit re-paid it in
+            # its first round
+            if self._frame is not None:
+                self._frame.cache_put("bins", self.x, max_bins,
+                                      self._by_bins[max_bins])
         return self._by_bins[max_bins]
 
     def sliced_features(self, indices: Optional[torch.Tensor]) -> torch.Tensor:
