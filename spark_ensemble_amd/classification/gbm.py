@@ -180,6 +180,7 @@ class GBMClassifier(ProbabilisticClassifier, _GBMClassifierParams):
             i = r0
             instr.log_named_value("resumed_from_round", r0)
 
+        instr.timers["setup_ms"] = instr.elapsed_ms()
         while i < k_stages and v < self.getOrDefault("numRounds"):
             idx = subspaces[i]
             xs = binned.sliced_features(idx)
