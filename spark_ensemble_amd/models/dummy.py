@@ -131,8 +131,13 @@ class DummyClassifier(ProbabilisticClassifier, _DummyClassifierParams):
             prob = torch.full((k,), 1.0 / k)
             raw = prob.log()
         elif strategy == "prior":
-            counts = torch.zeros(k, dtype=torch.float64, device=x.device)
-            counts.index_add_(0, y.long(), w.double())
+            # per-class masked sums, NOT index_add_: a 10M-row f64
+            # index_add onto k cells is a CAS-loop pileup on gfx950
+            # (~2.2 s measured for k=2); k reduction kernels are ~ms
+            yl = y.long()
+            counts = torch.stack(
+                [(w * (yl == c).to(w.dtype)).sum().double() for c in range(k)]
+            ).to(x.device)
             comm.all_reduce_(counts)
             prob = (counts / counts.sum()).float().cpu()
             raw = prob.clamp_min(1e-300).log()

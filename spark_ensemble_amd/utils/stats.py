@@ -66,7 +66,10 @@ def dist_quantile(
         idx = ((v - lo) / width).floor().clamp_(0, bins - 1).long()
         inside = (v >= lo) & (v <= hi)
         hist = torch.zeros(bins, dtype=torch.float64, device=v.device)
-        hist.index_add_(0, idx[inside], w[inside].double())
+        # f32 accumulation: f64 global atomics CAS-loop on gfx950
+        hist = hist.float()
+        hist.index_add_(0, idx[inside], w[inside].float())
+        hist = hist.double()
         comm.all_reduce_(hist)
         cum = below + hist.cumsum(0)
         sel = int((cum >= target).to(torch.int8).argmax())
