@@ -204,8 +204,17 @@ class PredictionModel(Model, _PredictorParams):
     def numFeatures(self) -> int:
         return getattr(self, "_num_features", -1)
 
+    def _check_features(self, x: torch.Tensor):
+        nf = self.numFeatures
+        if nf > 0 and x.shape[1] != nf:
+            raise ValueError(
+                f"{type(self).__name__} was trained on {nf} features but "
+                f"received {x.shape[1]}"
+            )
+
     def transform(self, dataset: TensorFrame) -> TensorFrame:
         x = dataset[self.getFeaturesCol()].float()
+        self._check_features(x)
         return dataset.withColumn(self.getPredictionCol(), self.predict(x))
 
 
@@ -269,6 +278,7 @@ class ClassificationModel(PredictionModel, _ClassifierParams):
 
     def transform(self, dataset: TensorFrame) -> TensorFrame:
         x = dataset[self.getFeaturesCol()].float()
+        self._check_features(x)
         raw = self.predictRaw(x)
         out = dataset.withColumn(self.getRawPredictionCol(), raw)
         out = out.withColumn(self.getPredictionCol(), raw.argmax(dim=1).float())
@@ -287,6 +297,7 @@ class ProbabilisticClassificationModel(
 
     def transform(self, dataset: TensorFrame) -> TensorFrame:
         x = dataset[self.getFeaturesCol()].float()
+        self._check_features(x)
         raw = self.predictRaw(x)
         prob = self.raw2probabilityInPlace(raw.clone())
         out = dataset.withColumn(self.getRawPredictionCol(), raw)

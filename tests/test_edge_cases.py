@@ -66,3 +66,14 @@ def test_more_bins_than_distinct_values():
     )
     p = m.predict(x)
     assert torch.allclose(p, y, atol=1e-5)
+
+
+def test_feature_width_mismatch_raises():
+    import pytest
+
+    df = TensorFrame(features=torch.randn(100, 5),
+                     label=torch.randint(0, 2, (100,)).float())
+    m = sea.GBMClassifier().setNumBaseLearners(2).fit(df)
+    bad = TensorFrame(features=torch.randn(10, 7))
+    with pytest.raises(ValueError, match="trained on 5 features"):
+        m.transform(bad)
