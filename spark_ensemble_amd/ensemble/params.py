@@ -82,11 +82,16 @@ class HasBaseLearner(FitsBaseLearners):
 
     # -- persistence (learner/ subdir; ensembleParams.scala:87-103) --------
     def _save_learner(self, path: str):
-        self.getBaseLearner().save(os.path.join(path, "learner"), overwrite=True)
+        # baseLearner may be unset (the built-in tree default applies);
+        # estimator saves then simply omit the learner/ subdir
+        lr = self.getOrNone("baseLearner")
+        if lr is not None:
+            lr.save(os.path.join(path, "learner"), overwrite=True)
 
     @staticmethod
     def _load_learner(path: str):
-        return persistence.load_instance(os.path.join(path, "learner"))
+        p = os.path.join(path, "learner")
+        return persistence.load_instance(p) if os.path.isdir(p) else None
 
 
 class HasBaseLearners(FitsBaseLearners):

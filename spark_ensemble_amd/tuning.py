@@ -38,7 +38,8 @@ from .parallel import get_comm
 
 class Pipeline(Estimator):
     """``Pipeline(stages=[t1, est]).fit(df)`` fits estimator stages in
-    order, transforming the running frame through each fitted stage."""
+    order, transforming the running frame through each fitted stage.
+    Persists with the Spark layout: ``stage-<i>/`` nested saves."""
 
     def __init__(self, uid=None, stages: Optional[Sequence] = None):
         super().__init__(uid)
@@ -69,6 +70,24 @@ class Pipeline(Estimator):
         pm._stages = fitted
         return pm
 
+    def _save_impl(self, path: str):
+        from . import persistence
+        import os
+
+        persistence.save_metadata(self, path,
+                                  extra={"numStages": len(self._stages)})
+        for i, st in enumerate(self._stages):
+            st.save(os.path.join(path, f"stage-{i}"), overwrite=True)
+
+    def _load_extra(self, path: str, meta: dict):
+        from . import persistence
+        import os
+
+        self._stages = [
+            persistence.load_instance(os.path.join(path, f"stage-{i}"))
+            for i in range(int(meta.get("numStages", 0)))
+        ]
+
 
 class PipelineModel(Model):
     _stages: List[Transformer]
@@ -82,6 +101,24 @@ class PipelineModel(Model):
         for s in self._stages:
             df = s.transform(df)
         return df
+
+    def _save_impl(self, path: str):
+        from . import persistence
+        import os
+
+        persistence.save_metadata(self, path,
+                                  extra={"numStages": len(self._stages)})
+        for i, st in enumerate(self._stages):
+            st.save(os.path.join(path, f"stage-{i}"), overwrite=True)
+
+    def _load_extra(self, path: str, meta: dict):
+        from . import persistence
+        import os
+
+        self._stages = [
+            persistence.load_instance(os.path.join(path, f"stage-{i}"))
+            for i in range(int(meta.get("numStages", 0)))
+        ]
 
 
 # ---------------------------------------------------------------------------

@@ -98,3 +98,23 @@ def test_train_validation_split():
     assert len(m.avgMetrics) == 2
     # 4 boosting rounds must beat 1 on held-out rmse for this easy data
     assert m.bestIndex == 1, m.avgMetrics
+
+
+def test_pipeline_save_load(tmp_path):
+    from spark_ensemble_amd.tuning import Pipeline, PipelineModel
+
+    df = synthetic_classification(400, 8, k=2, seed=4)
+    pm = Pipeline(stages=[sea.GBMClassifier().setNumBaseLearners(2)]).fit(df)
+    p = str(tmp_path / "pipe")
+    pm.save(p)
+    pm2 = PipelineModel.load(p)
+    a = pm.transform(df)["rawPrediction"]
+    b = pm2.transform(df)["rawPrediction"]
+    assert torch.allclose(a, b, rtol=1e-6, atol=1e-7)
+
+    # estimator pipeline round-trips too
+    est = Pipeline(stages=[sea.GBMRegressor().setNumBaseLearners(2)])
+    pe = str(tmp_path / "pipe_est")
+    est.save(pe)
+    est2 = Pipeline.load(pe)
+    assert len(est2.getStages()) == 1
