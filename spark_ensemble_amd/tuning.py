@@ -333,6 +333,29 @@ class CrossValidatorModel(Model):
     def transform(self, dataset: TensorFrame) -> TensorFrame:
         return self.bestModel.transform(dataset)
 
+    def _save_impl(self, path: str):
+        import os
+
+        from . import persistence
+
+        persistence.save_metadata(
+            self, path,
+            extra={"avgMetrics": list(map(float, self.avgMetrics)),
+                   "bestIndex": int(self.bestIndex)},
+        )
+        self.bestModel.save(os.path.join(path, "bestModel"), overwrite=True)
+
+    def _load_extra(self, path: str, meta: dict):
+        import os
+
+        from . import persistence
+
+        self.avgMetrics = meta.get("avgMetrics", [])
+        self.bestIndex = int(meta.get("bestIndex", 0))
+        self.bestModel = persistence.load_instance(
+            os.path.join(path, "bestModel")
+        )
+
 
 class TrainValidationSplit(CrossValidator):
     """Single split selection (Spark's TrainValidationSplit)."""

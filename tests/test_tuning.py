@@ -118,3 +118,24 @@ def test_pipeline_save_load(tmp_path):
     est.save(pe)
     est2 = Pipeline.load(pe)
     assert len(est2.getStages()) == 1
+
+
+def test_cross_validator_model_save_load(tmp_path):
+    from spark_ensemble_amd.tuning import CrossValidator, CrossValidatorModel
+
+    df = synthetic_regression(400, 6, seed=8)
+    cv = CrossValidator(
+        estimator=sea.GBMRegressor().setSeed(1),
+        estimatorParamMaps=[{"numBaseLearners": 1}, {"numBaseLearners": 3}],
+        evaluator=RegressionEvaluator("rmse"),
+        numFolds=2, seed=3,
+    )
+    m = cv.fit(df)
+    p = str(tmp_path / "cv")
+    m.save(p)
+    m2 = CrossValidatorModel.load(p)
+    assert m2.bestIndex == m.bestIndex
+    assert m2.avgMetrics == [float(v) for v in m.avgMetrics]
+    a = m.transform(df)["prediction"]
+    b = m2.transform(df)["prediction"]
+    assert torch.allclose(a, b, rtol=1e-6, atol=1e-7)
