@@ -74,3 +74,32 @@ def test_fit_with_param_map_list():
     )
     assert len(models) == 2
     assert models[0].numModels == 1 and models[1].numModels == 3
+
+
+def test_mllib_metadata_layout(tmp_path):
+    """Pin the on-disk contract (reference §3.4 layout): one JSON line in
+    metadata/part-00000 with class/uid/paramMap, nested model-$i dirs."""
+    import json
+    import os
+
+    import spark_ensemble_amd as sea
+    from spark_ensemble_amd.utils.io import synthetic_regression
+
+    df = synthetic_regression(200, 5, seed=1)
+    m = sea.BaggingRegressor().setNumBaseLearners(2).fit(df)
+    p = tmp_path / "m"
+    m.save(str(p))
+
+    meta_file = p / "metadata" / "part-00000"
+    assert meta_file.exists()
+    lines = meta_file.read_text().strip().split("\n")
+    assert len(lines) == 1
+    meta = json.loads(lines[0])
+    assert meta["class"].endswith("BaggingRegressionModel")
+    assert "uid" in meta and "paramMap" in meta
+    assert meta["numModels"] == 2
+    assert (p / "model-0" / "metadata" / "part-00000").exists()
+    assert (p / "model-1").is_dir()
+    assert (p / "data-0" / "part-00000").exists()
+    row = json.loads((p / "data-0" / "part-00000").read_text().strip())
+    assert "subspace" in row
