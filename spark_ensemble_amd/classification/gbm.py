@@ -12,7 +12,10 @@ Re-creates reference classification/GBMClassifier.scala:219-496:
   * newton pseudo-residuals with per-dim hessian floor 1e-2 and weights
     0.5 * h_j / sum(h_j) * w (:337-375),
   * dim-D line search by L-BFGS-B with bounds [0, inf) started at
-    ones(dim) (:413-431),
+    ones(dim) (:413-431); scalar-dim smooth losses (bernoulli,
+    exponential) use the safeguarded-Newton search instead of Brent
+    (same minimizer, fewer full-data evaluations — see
+    boosting/line_search.py),
   * patience early stop identical to GBMRegressor (:451-479),
   * model predictRaw = init raw + sum_i sum_j w_ij f_ij(slice(x)); binary
     dim-1 maps s -> (-s, s) (:567-589); raw2probability delegates to the
