@@ -95,7 +95,12 @@ class BoostingClassifier(ProbabilisticClassifier, _BoostingClassifierParams):
         from ..utils import checkpoint as ckpt
 
         ckpt_dir = self.getCheckpointDir()
-        resumed = ckpt.load_round_state(ckpt_dir)
+        ck_fp = (
+            ckpt.fingerprint(self, x.shape[0], x.shape[1], y, w)
+            if ckpt_dir else None
+        )
+        ck_saved: set = set()
+        resumed = ckpt.load_round_state(ckpt_dir, ck_fp)
         replay = resumed[1][:k] if resumed else []
         if replay:
             instr.log_named_value("resumed_from_round", len(replay))
@@ -157,10 +162,12 @@ class BoostingClassifier(ProbabilisticClassifier, _BoostingClassifierParams):
             interval = self.getCheckpointInterval()
             if (ckpt_dir and i >= len(replay) and interval > 0
                     and (i + 1) % interval == 0):
-                ckpt.save_round_state(ckpt_dir, i + 1, models, est_weights)
+                ckpt.save_round_state(ckpt_dir, i + 1, models, est_weights,
+                                      fingerprint=ck_fp, _saved_dirs=ck_saved)
             i += 1
 
         instr.finish()
+        ckpt.clear(ckpt_dir)  # resume state is crash recovery only
         model = BoostingClassificationModel()
         model._models = models
         model._weights = est_weights

@@ -164,7 +164,11 @@ class GBMClassifier(ProbabilisticClassifier, _GBMClassifierParams):
         from ..utils import checkpoint as ckpt
 
         ckpt_dir = self.getCheckpointDir()
-        resumed = ckpt.load_round_state(ckpt_dir)
+        ck_fp = (
+            ckpt.fingerprint(self, n, num_features, y, w) if ckpt_dir else None
+        )
+        ck_saved: set = set()
+        resumed = ckpt.load_round_state(ckpt_dir, ck_fp)
         if resumed:
             r0, models, weights, extra = resumed
             r0 = min(r0, k_stages)
@@ -271,11 +275,14 @@ class GBMClassifier(ProbabilisticClassifier, _GBMClassifierParams):
                 ckpt.save_round_state(
                     ckpt_dir, i + 1, models, weights,
                     extra={"best_err": best_err, "v": v},
+                    fingerprint=ck_fp, _saved_dirs=ck_saved,
                 )
             i += 1
 
         keep = i - v
         instr.finish()
+        # resume state is crash recovery only; a completed fit clears it
+        ckpt.clear(ckpt_dir)
         model = GBMClassificationModel()
         model._init = init
         model._models = models[:keep]

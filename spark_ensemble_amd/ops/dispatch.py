@@ -104,21 +104,29 @@ def hist_build(bins, gh, row_idx, node_offsets, num_bins, d_dims=-1, max_abs=Non
             if C > 8:
                 # wide one-hot targets (K-class gini trees, K > 7): chunk
                 # signed channels into kernel-sized groups, each carrying
-                # the non-negative tail (hess/count) for bookkeeping
+                # the non-negative tail (hess/count) for bookkeeping.
+                # Chunk size is 8 - len(tail) so every recursive call has at
+                # most 8 channels and takes the direct kernel path (a fixed
+                # chunk of 7 would re-enter this branch forever when the
+                # tail has 2 channels, e.g. weighted multiclass).
                 D = d_dims if d_dims > 0 else C - 1
                 tail = list(range(D, C))
+                step = 8 - len(tail)
+                assert 1 <= step <= 7, f"hist_build tail too wide: {len(tail)}"
                 parts = []
-                for s in range(0, D, 7):
-                    cols = list(range(s, min(s + 7, D))) + tail
+                for s in range(0, D, step):
+                    w = min(step, D - s)
+                    cols = list(range(s, s + w)) + tail
                     idx = torch.tensor(cols, device=gh.device)
                     part = hist_build(
                         bins, gh.index_select(1, idx).contiguous(), row_idx,
-                        node_offsets, num_bins, min(7, D - s),
+                        node_offsets, num_bins, w,
                         max_abs[torch.tensor(cols)], identity_rows,
                     )
-                    parts.append(part[..., : min(7, D - s)])
-                    if s + 7 >= D:
-                        parts.append(part[..., min(7, D - s):])  # tail once
+                    assert part.shape[-1] <= 8  # recursion must have bottomed out
+                    parts.append(part[..., :w])
+                    if s + step >= D:
+                        parts.append(part[..., w:])  # tail once
                 return torch.cat(parts, dim=-1)
             out = torch.zeros(
                 n_nodes, F, num_bins, C, dtype=torch.float32, device=bins.device

@@ -181,7 +181,11 @@ class GBMRegressor(Regressor, _GBMRegressorParams):
         from ..utils import checkpoint as ckpt
 
         ckpt_dir = self.getCheckpointDir()
-        resumed = ckpt.load_round_state(ckpt_dir)
+        ck_fp = (
+            ckpt.fingerprint(self, n, num_features, y, w) if ckpt_dir else None
+        )
+        ck_saved: set = set()
+        resumed = ckpt.load_round_state(ckpt_dir, ck_fp)
         if resumed:
             r0, models, weights, extra = resumed
             r0 = min(r0, k_stages)
@@ -278,9 +282,13 @@ class GBMRegressor(Regressor, _GBMRegressorParams):
                 ckpt.save_round_state(
                     ckpt_dir, i + 1, models, weights,
                     extra={"best_err": best_err, "v": v},
+                    fingerprint=ck_fp, _saved_dirs=ck_saved,
                 )
             i += 1
         instr.finish()
+        # a completed fit owns no resume state (resume is crash recovery
+        # only — checkpointInterval must never change the fitted result)
+        ckpt.clear(ckpt_dir)
 
         keep = i - v
         model = GBMRegressionModel()

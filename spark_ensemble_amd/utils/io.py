@@ -66,6 +66,13 @@ def synthetic_classification(
     row draws from the same task, so train (split=0) and test (split=1)
     frames share a distribution.  Learnable by trees and by logistic
     regression, not trivially separable.
+
+    NOTE: rows are drawn from a DEVICE-LOCAL generator (CPU
+    and GPU torch RNG engines differ), so the same ``seed`` gives
+    different rows on CPU vs GPU.  The distribution (and hence every
+    statistical-quality assertion) is device-independent; only row-exact
+    CPU-vs-GPU comparisons must generate on one device and ``.to()`` the
+    other — which is what the parity tests in tests/test_gpu.py do.
     """
     rank, world = shard or (0, 1)
     task = torch.Generator().manual_seed(seed * 9176 + 4242)

@@ -65,11 +65,18 @@ def dist_quantile(
             break
         idx = ((v - lo) / width).floor().clamp_(0, bins - 1).long()
         inside = (v >= lo) & (v <= hi)
+        # f32 index_add_ (f64 global atomics CAS-loop on gfx950), but
+        # chunked so no single f32 accumulation exceeds ~2^22 increments
+        # (past 2^24 a bin would silently drop unit adds — ADVICE r01),
+        # then summed into f64 before the all-reduce.
+        idx_in = idx[inside]
+        w_in = w[inside].float()
         hist = torch.zeros(bins, dtype=torch.float64, device=v.device)
-        # f32 accumulation: f64 global atomics CAS-loop on gfx950
-        hist = hist.float()
-        hist.index_add_(0, idx[inside], w[inside].float())
-        hist = hist.double()
+        chunk = 1 << 22
+        for s in range(0, max(idx_in.numel(), 1), chunk):
+            part = torch.zeros(bins, dtype=torch.float32, device=v.device)
+            part.index_add_(0, idx_in[s:s + chunk], w_in[s:s + chunk])
+            hist += part.double()
         comm.all_reduce_(hist)
         cum = below + hist.cumsum(0)
         sel = int((cum >= target).to(torch.int8).argmax())
