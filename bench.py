@@ -99,13 +99,18 @@ def main():
 
     def timed_fit(loss: str, rounds: int) -> float:
         """Barrier+sync-bracketed wall time of one public fit() call,
-        MAX over ranks."""
+        MAX over ranks.  Each call gets a FRESH TensorFrame over the same
+        tensors: the frame-level bins cache must not leak a previous
+        fit's binning into this fit's timed region."""
+        from spark_ensemble_amd.frame import TensorFrame
+
+        fit_df = TensorFrame(features=df["features"], label=df["label"])
         est = make_est(loss, rounds)
         comm.barrier()
         if device.type == "cuda":
             torch.cuda.synchronize()
         t_start = time.time()
-        est.fit(df)
+        est.fit(fit_df)
         if device.type == "cuda":
             torch.cuda.synchronize()
         comm.barrier()
