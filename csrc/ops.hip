@@ -760,6 +760,156 @@ void forest_predict(torch::Tensor out, torch::Tensor x, torch::Tensor feat,
 }
 
 // ---------------------------------------------------------------------------
+// forest_predict2: LDS-staged tree-group-tiled forest inference.
+//
+//   The v1 kernel above walks ALL T trees per row with every node fetch a
+//   dependent global load (L1-thrashing: 64 lanes x ~800 scattered node +
+//   feature reads) — measured 72.7M rows/s for a 100-tree depth-8 forest
+//   (docs/performance.md r01).  v2 restructures serving the MI355X way:
+//
+//   * nodes are packed host-side into ONE u64 per node:
+//       low 16  = feature (s16, -1 leaf)   | mid 16 = left-child (s16,
+//       sibling right = left+1)            | high 32 = f32 bits of the
+//       threshold — or, for D == 1, the WEIGHTED leaf value when feat < 0
+//       (leaf tensors never touched in-kernel for scalar outputs).
+//   * trees are tiled into groups of <= 16 Ki nodes (128 KiB); each block
+//     stages its group's packed nodes in LDS once (ds_read_b64 is 256
+//     B/clk/CU vs ~200-900 cyc global latency per dependent node hop),
+//     then grid-strides over rows: per (row, group) all walks hit LDS.
+//   * one f32 atomicAdd per (row, channel, group) merges group partials —
+//     different rows never contend.
+//
+//   grid (row_chunks, n_groups) x 1024 threads; 1 block/CU at 128 KiB LDS
+//   = 16 waves covering the ds_read dependent-latency chain.
+//   Reference semantics: the model transform loops at reference
+//   BaggingRegressor.scala:221-228, GBMClassifier.scala:567-589.
+// ---------------------------------------------------------------------------
+
+template <bool D1>
+__global__ void forest_predict2_kernel(
+    float* __restrict__ out,                       // [N, D] (+=, pre-zeroed)
+    const float* __restrict__ x,                   // [N, F]
+    const unsigned long long* __restrict__ nodes,  // [total] packed
+    const float* __restrict__ leaf,                // [total, D] (D > 1)
+    const int* __restrict__ tree_off,              // [T] global node base
+    const float* __restrict__ w,                   // [T]
+    const int* __restrict__ groups,                // [G, 4]
+    int64_t n, int F, int D) {
+  extern __shared__ unsigned long long tlds[];  // group nodes
+  const int* grp = groups + blockIdx.y * 4;
+  const int first_tree = grp[0];
+  const int n_trees = grp[1];
+  const int node_base = grp[2];
+  const int n_nodes = grp[3];
+  for (int i = threadIdx.x; i < n_nodes; i += blockDim.x)
+    tlds[i] = nodes[node_base + i];
+  __syncthreads();
+
+  int64_t r = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  const int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; r < n; r += stride) {
+    const float* xr = x + r * F;
+    float acc[8];
+    if (!D1)
+#pragma unroll
+      for (int d = 0; d < 8; ++d) acc[d] = 0.0f;
+    float acc1 = 0.0f;
+    // 4 trees walked per lane concurrently: each hop is a dependent
+    // LDS-node + scattered-x load chain (~50 + ~200 cyc); interleaving
+    // four independent chains quadruples the memory-level parallelism
+    // the wave exposes (fully unrolled -> no runtime-indexed locals)
+    for (int t = 0; t < n_trees; t += 4) {
+      const int nt4 = min(4, n_trees - t);
+      int toff[4], node[4], f[4];
+      unsigned long long nd[4];
+#pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        if (k < nt4) {
+          toff[k] = tree_off[first_tree + t + k] - node_base;
+          node[k] = toff[k];
+          nd[k] = tlds[node[k]];
+          f[k] = (short)(nd[k] & 0xFFFFu);
+        } else {
+          f[k] = -1;
+        }
+      }
+      bool alive = true;
+      while (alive) {
+        alive = false;
+#pragma unroll
+        for (int k = 0; k < 4; ++k) {
+          if (f[k] >= 0) {
+            const float xv = xr[f[k]];
+            const float thr = __uint_as_float((unsigned)(nd[k] >> 32));
+            const int left = (short)((nd[k] >> 16) & 0xFFFFu);
+            node[k] = toff[k] + left + (xv <= thr ? 0 : 1);
+            nd[k] = tlds[node[k]];
+            f[k] = (short)(nd[k] & 0xFFFFu);
+            alive |= (f[k] >= 0);
+          }
+        }
+      }
+#pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        if (k < nt4) {
+          if (D1) {
+            acc1 += __uint_as_float((unsigned)(nd[k] >> 32));
+          } else {
+            const float wt = w[first_tree + t + k];
+            const float* lv = leaf + (int64_t)(node_base + node[k]) * D;
+            for (int d = 0; d < D; ++d) acc[d] += wt * lv[d];
+          }
+        }
+      }
+    }
+    if (D1) {
+      atomicAdd(out + r, acc1);
+    } else {
+      float* o = out + r * D;
+      for (int d = 0; d < D; ++d)
+        if (acc[d] != 0.0f) atomicAdd(o + d, acc[d]);
+    }
+  }
+}
+
+void forest_predict2(torch::Tensor out, torch::Tensor x, torch::Tensor nodes,
+                     torch::Tensor leaf, torch::Tensor tree_off,
+                     torch::Tensor w, torch::Tensor groups, int64_t D,
+                     int64_t max_group_nodes) {
+  CHECK_GPU(out); CHECK_GPU(x); CHECK_GPU(nodes); CHECK_GPU(groups);
+  CHECK_CONTIG(out); CHECK_CONTIG(x); CHECK_CONTIG(nodes);
+  TORCH_CHECK(D <= 8, "forest_predict2: D <= 8");
+  int64_t n = x.size(0);
+  int F = (int)x.size(1);
+  const int G = (int)groups.size(0);
+  const size_t lds = (size_t)max_group_nodes * 8;
+  TORCH_CHECK(lds <= 131072, "forest_predict2: group too big for LDS");
+  auto stream = at::hip::getCurrentHIPStream();
+  const int threads = 1024;
+  int rblocks = (int)std::min<int64_t>(ceil_div(n, threads), 8192);
+  // fill the chip even for few groups
+  rblocks = std::max(rblocks, (int)std::min<int64_t>(
+                                  ceil_div(512, (int64_t)G), ceil_div(n, 64)));
+#define FP2_LAUNCH(DD1)                                                       \
+  do {                                                                        \
+    if (lds > 65536)                                                          \
+      (void)hipFuncSetAttribute(                                              \
+          reinterpret_cast<const void*>(&forest_predict2_kernel<DD1>),        \
+          hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);              \
+    hipLaunchKernelGGL((forest_predict2_kernel<DD1>), dim3(rblocks, G),       \
+                       dim3(threads), lds, stream, out.data_ptr<float>(),     \
+                       x.data_ptr<float>(),                                   \
+                       (const unsigned long long*)nodes.data_ptr<int64_t>(),  \
+                       leaf.numel() ? leaf.data_ptr<float>() : nullptr,       \
+                       tree_off.data_ptr<int>(), w.data_ptr<float>(),         \
+                       groups.data_ptr<int>(), n, F, (int)D);                 \
+  } while (0)
+  if (D == 1) FP2_LAUNCH(true);
+  else FP2_LAUNCH(false);
+#undef FP2_LAUNCH
+}
+
+// ---------------------------------------------------------------------------
 // fused GBM loss kernels
 //   loss ids match spark_ensemble_amd/boosting/losses.py LOSS_IDS
 // ---------------------------------------------------------------------------
@@ -1272,6 +1422,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("partition_rows", &partition_rows, "block-aggregated node partition");
   m.def("tree_predict", &tree_predict, "single-tree batched predict");
   m.def("forest_predict", &forest_predict, "packed-forest weighted predict");
+  m.def("forest_predict2", &forest_predict2,
+        "LDS-staged tree-group-tiled forest predict");
   m.def("grad_hess", &grad_hess, "fused per-row loss gradient/hessian");
   m.def("line_search_eval", &line_search_eval, "fused loss+grad line-search payload");
 }

@@ -323,13 +323,16 @@ class GBMClassificationModel(ProbabilisticClassificationModel, _GBMClassifierPar
         out = self._init.predictRaw(x)[:, : self._dim].contiguous()
         if self._models:
             # fast path: identity subspaces + tree stages -> one packed
-            # forest_predict kernel per class dimension
+            # forest_predict kernel per class dimension (arena cached
+            # across transform calls)
+            caches = self.__dict__.setdefault("_pack_caches", {})
             packed_ok = True
             for j in range(self._dim):
                 pj = packed_forest_margin(
                     x, [ms[j] for ms in self._models],
                     [wts[j] for wts in self._weights],
                     self._subspaces, x.shape[1],
+                    cache=caches.setdefault(j, {}),
                 )
                 if pj is None:
                     packed_ok = False

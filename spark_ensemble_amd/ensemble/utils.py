@@ -64,14 +64,16 @@ def slice_features(x: torch.Tensor, indices: torch.Tensor) -> torch.Tensor:
     return x.index_select(1, indices.to(x.device))
 
 
-def packed_forest_margin(x, models, weights, subspaces, num_features):
+def packed_forest_margin(x, models, weights, subspaces, num_features,
+                         cache=None):
     """Batched ensemble inference fast path: when every stage model is a
     built-in regression tree, the whole ensemble is ONE forest_predict
     kernel call instead of a launch (plus a feature-slice copy) per
     stage.  Non-identity subspaces are handled by remapping each tree's
     split-feature ids back into the ORIGINAL feature space, so the packed
     forest walks the unsliced x.  Returns [N] margins or None when a
-    stage is not a tree."""
+    stage is not a tree.  ``cache``: caller-owned dict holding the packed
+    arena across transform calls (models are immutable post-fit)."""
     from ..models.tree import DecisionTreeRegressionModel
     from ..ops import dispatch as _ops
 
@@ -98,7 +100,7 @@ def packed_forest_margin(x, models, weights, subspaces, num_features):
             t = dict(t, feature=remapped)
         trees.append(t)
     w = torch.tensor([float(v) for v in weights], dtype=torch.float32)
-    return _ops.forest_predict(x, trees, w).squeeze(1)
+    return _ops.forest_predict(x, trees, w, cache=cache).squeeze(1)
 
 
 def ensemble_feature_importances(models, weights, subspaces, num_features):

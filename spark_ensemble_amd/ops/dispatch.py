@@ -246,35 +246,109 @@ def tree_predict(x, feature, threshold, left_child, leaf_value, max_depth):
     return reference.tree_predict(x, feature, threshold, left_child, leaf_value, max_depth)
 
 
-def forest_predict(x, trees, weights=None, max_depth=64):
+def forest_predict(x, trees, weights=None, max_depth=64, cache=None):
+    """Packed-forest ensemble inference.
+
+    ``cache``: an optional caller-owned dict; the packed node arena is
+    stored there under key "pack" so repeated transform calls on the same
+    (immutable, post-fit) tree list skip the per-call packing.
+    """
     if x.is_cuda and trees:
         m = _require_hip("forest_predict")
         if m is not None:
-            return _forest_predict_hip(m, x, trees, weights)
+            pack = cache.get("pack") if cache is not None else None
+            if pack is None or pack["T"] != len(trees) or pack["dev"] != x.device:
+                pack = _pack_forest(trees, weights, x.device)
+                if cache is not None:
+                    cache["pack"] = pack
+            return _forest_predict_packed(m, x, pack)
     return reference.forest_predict(x, trees, weights, max_depth)
 
 
-def _forest_predict_hip(m, x, trees, weights):
-    # pack all trees into one node-array arena so a single kernel launch
-    # walks every (row, tree) pair
-    dev = x.device
+_FP2_GROUP_NODES = 16384  # 128 KiB of packed 8-B nodes per LDS group
+
+
+def _pack_forest(trees, weights, dev):
+    """Pack a tree list into the arena tensors both kernel paths consume.
+    Done once per (model, device) — callers cache the result."""
     # per-tree moves BEFORE the cat: a resumed ensemble mixes CPU-loaded
     # checkpoint stages with device-fitted ones
     feats = torch.cat([t["feature"].to(dev, torch.int32) for t in trees])
     thrs = torch.cat([t["threshold"].to(dev, torch.float32) for t in trees])
     lefts = torch.cat([t["left_child"].to(dev, torch.int32) for t in trees])
-    leaves = torch.cat([t["leaf_value"].to(dev, torch.float32) for t in trees])
-    sizes = torch.tensor([t["feature"].numel() for t in trees], dtype=torch.int64)
-    offsets = torch.cat([torch.zeros(1, dtype=torch.int64), sizes.cumsum(0)])[:-1]
-    offsets = offsets.to(torch.int32).to(dev)
+    leaves = torch.cat(
+        [t["leaf_value"].to(dev, torch.float32) for t in trees]
+    ).contiguous()
+    sizes = [t["feature"].numel() for t in trees]
+    sizes_t = torch.tensor(sizes, dtype=torch.int64)
+    offsets = torch.cat([torch.zeros(1, dtype=torch.int64), sizes_t.cumsum(0)])[:-1]
+    offsets32 = offsets.to(torch.int32).to(dev)
     D = trees[0]["leaf_value"].shape[1]
     T = len(trees)
     if weights is None:
         w = torch.ones(T, dtype=torch.float32, device=dev)
     else:
         w = weights.to(torch.float32).to(dev)
-    out = torch.zeros(x.shape[0], D, dtype=torch.float32, device=dev)
-    m.forest_predict(out, x.contiguous(), feats, thrs, lefts, leaves.contiguous(), offsets, w, D)
+
+    pack = {
+        "T": T, "D": D, "dev": dev, "feats": feats, "thrs": thrs,
+        "lefts": lefts, "leaves": leaves, "offsets32": offsets32, "w": w,
+        "v2": False,
+    }
+    fits_v2 = D <= 8 and max(sizes) <= _FP2_GROUP_NODES
+    if not fits_v2:
+        return pack
+
+    # ---- v2: pack (feat s16 | left s16 | payload f32-bits) into one u64
+    # per node; for D == 1 a leaf's payload IS its weighted leaf value so
+    # the kernel never touches the leaf tensor (csrc forest_predict2)
+    if D == 1:
+        tree_id = torch.repeat_interleave(
+            torch.arange(T, device=dev), sizes_t.to(dev)
+        )
+        payload = torch.where(feats < 0, w[tree_id] * leaves[:, 0], thrs)
+    else:
+        payload = thrs
+    node64 = (
+        (feats.to(torch.int64) & 0xFFFF)
+        | ((lefts.to(torch.int64) & 0xFFFF) << 16)
+        | ((payload.view(torch.int32).to(torch.int64) & 0xFFFFFFFF) << 32)
+    ).contiguous()
+
+    # greedy contiguous tree groups under the LDS node budget
+    groups = []
+    off_list = offsets.tolist()
+    first = 0
+    acc = 0
+    max_nodes = 0
+    for t in range(T):
+        if acc + sizes[t] > _FP2_GROUP_NODES and acc > 0:
+            groups.append([first, t - first, off_list[first], acc])
+            max_nodes = max(max_nodes, acc)
+            first, acc = t, 0
+        acc += sizes[t]
+    groups.append([first, T - first, off_list[first], acc])
+    max_nodes = max(max_nodes, acc)
+    pack.update(
+        v2=True,
+        node64=node64,
+        groups_t=torch.tensor(groups, dtype=torch.int32, device=dev),
+        max_nodes=max_nodes,
+    )
+    return pack
+
+
+def _forest_predict_packed(m, x, pack):
+    D = pack["D"]
+    out = torch.zeros(x.shape[0], D, dtype=torch.float32, device=x.device)
+    if pack["v2"] and x.shape[1] < 32768 and hasattr(m, "forest_predict2"):
+        m.forest_predict2(out, x.contiguous(), pack["node64"], pack["leaves"],
+                          pack["offsets32"], pack["w"], pack["groups_t"], D,
+                          pack["max_nodes"])
+    else:
+        m.forest_predict(out, x.contiguous(), pack["feats"], pack["thrs"],
+                         pack["lefts"], pack["leaves"], pack["offsets32"],
+                         pack["w"], D)
     return out
 
 

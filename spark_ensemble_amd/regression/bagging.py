@@ -123,6 +123,7 @@ class BaggingRegressionModel(RegressionModel, _BaggingRegressorParams):
         packed = packed_forest_margin(
             x, self._models, [1.0 / mcount] * mcount, self._subspaces,
             x.shape[1],
+            cache=self.__dict__.setdefault("_pack_cache", {}),
         )
         if packed is not None:
             return packed

@@ -321,7 +321,8 @@ class GBMRegressionModel(RegressionModel, _GBMRegressorParams):
         x = features.float()
         out = self._init.predict(x)
         packed = packed_forest_margin(
-            x, self._models, self._weights, self._subspaces, x.shape[1]
+            x, self._models, self._weights, self._subspaces, x.shape[1],
+            cache=self.__dict__.setdefault("_pack_cache", {}),
         )
         if packed is not None:
             return out + packed

@@ -1,0 +1,319 @@
+"""Dataset-based statistical-quality suites on the reference's own data
+(vendored from /root/reference/data: adult 32,560-row binary, cpusmall
+8,191-row regression, letter 14,999-row 26-class — the exact files every
+reference suite loads, e.g. GBMClassifierSuite.scala:53-58).
+
+Ports the reference's dominant test genre off synthetic
+data onto the real datasets:
+  * ensemble beats its base learner (BaggingClassifierSuite.scala:48-78,
+    GBMClassifierSuite.scala:51-87, BaggingRegressorSuite.scala:48-75)
+  * ensemble beats its best member (BaggingClassifierSuite.scala:80-112)
+  * diversity among bagged models (BaggingClassifierSuite.scala:114-155)
+  * SAMME ~= SAMME.R within +-0.02 (BoostingClassifierSuite.scala:93-124)
+  * cross-library anchors: the reference asserts against MLlib
+    RandomForest / GBTClassifier (BaggingRegressorSuite.scala:48-75,
+    GBMClassifierSuite.scala:89-146); sklearn plays that role here.
+
+Measured anchor accuracies are recorded in docs/datasets.md.
+"""
+
+import os
+
+import pytest
+import torch
+
+import spark_ensemble_amd as sea
+from spark_ensemble_amd.frame import TensorFrame
+from spark_ensemble_amd.models import (
+    DecisionTreeClassifier,
+    DecisionTreeRegressor,
+)
+from spark_ensemble_amd.utils.io import load_libsvm
+
+DATA = os.path.join(os.path.dirname(__file__), "..", "data")
+
+
+def _split(df: TensorFrame, seed: int = 5, frac: float = 0.75):
+    n = df["features"].shape[0]
+    g = torch.Generator().manual_seed(seed)
+    perm = torch.randperm(n, generator=g)
+    k = int(n * frac)
+    tr, te = perm[:k], perm[k:]
+    cols_tr = {c: df[c][tr] for c in df.columns}
+    cols_te = {c: df[c][te] for c in df.columns}
+    return TensorFrame(cols_tr), TensorFrame(cols_te)
+
+
+@pytest.fixture(scope="module")
+def adult():
+    df = load_libsvm(os.path.join(DATA, "adult", "adult.svm"))
+    return _split(df)
+
+
+@pytest.fixture(scope="module")
+def cpusmall():
+    df = load_libsvm(os.path.join(DATA, "cpusmall", "cpusmall.svm"))
+    return _split(df)
+
+
+@pytest.fixture(scope="module")
+def letter():
+    df = load_libsvm(os.path.join(DATA, "letter", "letter.svm"))
+    # labels are 1..26 -> 0-based class ids
+    cols = {c: df[c] for c in df.columns}
+    cols["label"] = cols["label"] - 1.0
+    return _split(TensorFrame(cols))
+
+
+def _acc(model, te):
+    return float(
+        (model.transform(te)["prediction"] == te["label"]).float().mean()
+    )
+
+
+def _mse(model, te):
+    p = model.predict(te["features"])
+    return float(((p - te["label"]) ** 2).mean())
+
+
+# ---------------------------------------------------------------------------
+# adult (binary classification)
+# ---------------------------------------------------------------------------
+
+
+def test_adult_bagging_beats_base(adult):
+    tr, te = adult
+    base = DecisionTreeClassifier().setMaxDepth(5)
+    single_acc = _acc(base.fit(tr), te)
+    bag = (
+        sea.BaggingClassifier()
+        .setBaseLearner(DecisionTreeClassifier().setMaxDepth(5))
+        .setNumBaseLearners(10)
+        .setSubsampleRatio(0.7)
+        .setSubspaceRatio(0.7)
+        .setReplacement(True)
+        .setSeed(3)
+        .fit(tr)
+    )
+    bag_acc = _acc(bag, te)
+    # vs the base learner trained on everything
+    assert bag_acc >= single_acc - 0.005, (bag_acc, single_acc)
+
+
+def test_letter_bagging_beats_best_member(letter):
+    """Reference BaggingClassifierSuite.scala:80-112 runs this on LETTER
+    with 20 learners, replacement, 0.8/0.8 — where ensembling gains are
+    decisive; mirrored here."""
+    tr, te = letter
+    bag = (
+        sea.BaggingClassifier()
+        .setBaseLearner(DecisionTreeClassifier().setMaxDepth(8))
+        .setNumBaseLearners(20)
+        .setSubsampleRatio(0.8)
+        .setSubspaceRatio(0.8)
+        .setReplacement(True)
+        .setSeed(0)
+        .fit(tr)
+    )
+    bag_acc = _acc(bag, te)
+    member_accs = []
+    for m, sub in zip(bag._models, bag._subspaces):
+        xs = te["features"].index_select(1, sub)
+        member_accs.append(
+            float((m.transform(TensorFrame(features=xs, label=te["label"]))
+                   ["prediction"] == te["label"]).float().mean())
+        )
+    assert bag_acc > max(member_accs), (bag_acc, max(member_accs))
+
+
+def test_adult_bagging_diversity(adult):
+    """Bagged members must disagree (reference
+    BaggingClassifierSuite.scala:114-155: pairwise agreement < 0.85)."""
+    tr, te = adult
+    bag = (
+        sea.BaggingClassifier()
+        .setBaseLearner(DecisionTreeClassifier().setMaxDepth(5))
+        .setNumBaseLearners(6)
+        .setSubsampleRatio(0.5)
+        .setSubspaceRatio(0.3)
+        .setReplacement(True)
+        .setSeed(11)
+        .fit(tr)
+    )
+    preds = []
+    for m, sub in zip(bag._models, bag._subspaces):
+        xs = te["features"].index_select(1, sub)
+        preds.append(m.transform(TensorFrame(features=xs))["prediction"])
+    agrees = []
+    for i in range(len(preds)):
+        for j in range(i + 1, len(preds)):
+            agrees.append(float((preds[i] == preds[j]).float().mean()))
+    assert min(agrees) < 0.85, agrees
+
+
+def test_adult_samme_matches_samme_r(adult):
+    """|acc(SAMME) - acc(SAMME.R)| <= 0.02 (reference
+    BoostingClassifierSuite.scala:93-124)."""
+    tr, te = adult
+    accs = {}
+    for algo in ("discrete", "real"):
+        m = (
+            sea.BoostingClassifier()
+            .setBaseLearner(DecisionTreeClassifier().setMaxDepth(3))
+            .setAlgorithm(algo)
+            .setNumBaseLearners(10)
+            .setSeed(7)
+            .fit(tr)
+        )
+        accs[algo] = _acc(m, te)
+    assert abs(accs["discrete"] - accs["real"]) <= 0.02, accs
+
+
+def test_adult_gbm_beats_tree_and_boosting(adult):
+    """GBM > plain tree and >= AdaBoost on adult (reference
+    GBMClassifierSuite.scala:51-87)."""
+    tr, te = adult
+    tree_acc = _acc(DecisionTreeClassifier().setMaxDepth(5).fit(tr), te)
+    boost_acc = _acc(
+        sea.BoostingClassifier()
+        .setBaseLearner(DecisionTreeClassifier().setMaxDepth(3))
+        .setNumBaseLearners(8)
+        .setSeed(2)
+        .fit(tr),
+        te,
+    )
+    gbm_acc = _acc(
+        sea.GBMClassifier()
+        .setBaseLearner(DecisionTreeRegressor().setMaxDepth(5))
+        .setLoss("bernoulli")
+        .setNumBaseLearners(15)
+        .setLearningRate(0.3)
+        .setSeed(2)
+        .fit(tr),
+        te,
+    )
+    assert gbm_acc > tree_acc, (gbm_acc, tree_acc)
+    assert gbm_acc >= boost_acc - 0.005, (gbm_acc, boost_acc)
+
+
+def test_adult_gbm_close_to_sklearn_gbt(adult):
+    """Cross-library anchor: within +-0.05 of sklearn's
+    GradientBoostingClassifier — the analog of the reference's
+    'within +-0.05 of MLlib GBTClassifier'
+    (GBMClassifierSuite.scala:142-144)."""
+    sklearn = pytest.importorskip("sklearn.ensemble")
+    tr, te = adult
+    ref = sklearn.GradientBoostingClassifier(
+        n_estimators=15, max_depth=5, learning_rate=0.3, random_state=0
+    ).fit(tr["features"].numpy(), tr["label"].numpy())
+    ref_acc = float(ref.score(te["features"].numpy(), te["label"].numpy()))
+    gbm_acc = _acc(
+        sea.GBMClassifier()
+        .setBaseLearner(DecisionTreeRegressor().setMaxDepth(5))
+        .setLoss("bernoulli")
+        .setNumBaseLearners(15)
+        .setLearningRate(0.3)
+        .setSeed(2)
+        .fit(tr),
+        te,
+    )
+    assert abs(gbm_acc - ref_acc) <= 0.05, (gbm_acc, ref_acc)
+
+
+# ---------------------------------------------------------------------------
+# cpusmall (regression)
+# ---------------------------------------------------------------------------
+
+
+def test_cpusmall_bagging_beats_base(cpusmall):
+    tr, te = cpusmall
+    single = _mse(DecisionTreeRegressor().setMaxDepth(6).fit(tr), te)
+    bag = (
+        sea.BaggingRegressor()
+        .setBaseLearner(DecisionTreeRegressor().setMaxDepth(6))
+        .setNumBaseLearners(15)
+        .setSubsampleRatio(0.7)
+        .setReplacement(True)
+        .setSeed(5)
+        .fit(tr)
+    )
+    assert _mse(bag, te) < single, (_mse(bag, te), single)
+
+
+def test_cpusmall_bagging_close_to_sklearn_rf(cpusmall):
+    """Cross-library anchor: the reference asserts its bagging beats
+    MLlib RandomForest (BaggingRegressorSuite.scala:48-75); here the
+    anchor is sklearn RandomForestRegressor at matched size."""
+    ensemble = pytest.importorskip("sklearn.ensemble")
+    tr, te = cpusmall
+    rf = ensemble.RandomForestRegressor(
+        n_estimators=15, max_depth=6, random_state=0
+    ).fit(tr["features"].numpy(), tr["label"].numpy())
+    rf_mse = float(
+        ((torch.from_numpy(rf.predict(te["features"].numpy())).float()
+          - te["label"]) ** 2).mean()
+    )
+    bag = (
+        sea.BaggingRegressor()
+        .setBaseLearner(DecisionTreeRegressor().setMaxDepth(6))
+        .setNumBaseLearners(15)
+        .setSubsampleRatio(0.8)
+        .setReplacement(True)
+        .setSeed(5)
+        .fit(tr)
+    )
+    ours = _mse(bag, te)
+    assert ours <= rf_mse * 1.25, (ours, rf_mse)
+
+
+def test_cpusmall_gbm_beats_base(cpusmall):
+    tr, te = cpusmall
+    single = _mse(DecisionTreeRegressor().setMaxDepth(5).fit(tr), te)
+    gbm = (
+        sea.GBMRegressor()
+        .setBaseLearner(DecisionTreeRegressor().setMaxDepth(5))
+        .setNumBaseLearners(15)
+        .setLearningRate(0.3)
+        .setSeed(5)
+        .fit(tr)
+    )
+    assert _mse(gbm, te) < single, (_mse(gbm, te), single)
+
+
+# ---------------------------------------------------------------------------
+# letter (26-class)
+# ---------------------------------------------------------------------------
+
+
+def test_letter_bagging_beats_single_tree(letter):
+    tr, te = letter
+    single_acc = _acc(DecisionTreeClassifier().setMaxDepth(8).fit(tr), te)
+    bag = (
+        sea.BaggingClassifier()
+        .setBaseLearner(DecisionTreeClassifier().setMaxDepth(8))
+        .setNumBaseLearners(8)
+        .setSubsampleRatio(0.8)
+        .setReplacement(True)
+        .setVotingStrategy("soft")
+        .setSeed(9)
+        .fit(tr)
+    )
+    bag_acc = _acc(bag, te)
+    assert bag_acc > single_acc, (bag_acc, single_acc)
+
+
+def test_letter_gbm_multiclass_learns(letter):
+    """26-class logloss GBM (dim=K => 26 trees/round) must clearly beat
+    the class prior after 2 rounds (the letter-shape multiclass path)."""
+    tr, te = letter
+    gbm = (
+        sea.GBMClassifier()
+        .setBaseLearner(DecisionTreeRegressor().setMaxDepth(5))
+        .setLoss("logloss")
+        .setNumBaseLearners(2)
+        .setLearningRate(0.5)
+        .setSeed(3)
+        .fit(tr)
+    )
+    acc = _acc(gbm, te)
+    assert acc > 0.3, acc  # prior is ~1/26 = 0.038

@@ -425,23 +425,107 @@ def synthetic_regression_gpu():
     return synthetic_regression(20000, 12, seed=23, device=DEV)
 
 
-def test_gpu_checkpoint_resume(tmp_path):
-    """Resume a half-trained GBM on GPU: saved (CPU) stage models must
-    replay cleanly against device tensors."""
+def test_gpu_checkpoint_resume(tmp_path, monkeypatch):
+    """Resume a CRASHED GBM fit on GPU: saved (CPU) stage models must
+    replay cleanly against device tensors.  (A completed fit clears its
+    own checkpoint, so the crash is injected mid-fit.)"""
+    import os
+
     import spark_ensemble_amd as sea
+    from spark_ensemble_amd.utils.instrumentation import Instrumentation
     from spark_ensemble_amd.utils.io import synthetic_regression
 
     df = synthetic_regression(50000, 16, seed=19, device=DEV)
 
-    def mk(k, ck):
-        e = sea.GBMRegressor().setNumBaseLearners(k).setSeed(4)
+    def mk(ck):
+        e = sea.GBMRegressor().setNumBaseLearners(4).setSeed(4)
         if ck:
             e.setCheckpointInterval(2).setCheckpointDir(str(tmp_path / "gk"))
         return e
 
-    m_full = mk(4, False).fit(df)
-    mk(2, True).fit(df)
-    m_res = mk(4, True).fit(df)
+    m_full = mk(False).fit(df)
+
+    class Crash(Exception):
+        pass
+
+    orig = Instrumentation.log_round
+
+    def patched(self, i, **kw):
+        orig(self, i, **kw)
+        if i >= 2:
+            raise Crash()
+
+    monkeypatch.setattr(Instrumentation, "log_round", patched)
+    with pytest.raises(Crash):
+        mk(True).fit(df)
+    monkeypatch.undo()
+    assert os.path.exists(tmp_path / "gk" / "state.json")
+
+    m_res = mk(True).fit(df)
     a = m_full.predict(df["features"])
     b = m_res.predict(df["features"])
     assert torch.allclose(a, b, rtol=1e-4, atol=1e-5)
+    assert not os.path.exists(tmp_path / "gk" / "state.json")
+
+
+def _rand_complete_tree(depth, F, D, gen):
+    """A random but VALID complete binary tree in flat node-array form
+    (BFS order: node i's children at 2i+1, 2i+2)."""
+    n_internal = 2 ** depth - 1
+    n_nodes = 2 ** (depth + 1) - 1
+    feature = torch.full((n_nodes,), -1, dtype=torch.int32)
+    feature[:n_internal] = torch.randint(0, F, (n_internal,), generator=gen)
+    threshold = torch.zeros(n_nodes)
+    threshold[:n_internal] = torch.randn(n_internal, generator=gen)
+    left = torch.full((n_nodes,), -1, dtype=torch.int32)
+    left[:n_internal] = 2 * torch.arange(n_internal, dtype=torch.int32) + 1
+    leaf = torch.randn(n_nodes, D, generator=gen)
+    return {"feature": feature, "threshold": threshold,
+            "left_child": left, "leaf_value": leaf}
+
+
+@pytest.mark.parametrize("depth,T,D", [(9, 40, 1), (6, 30, 3)])
+def test_forest_predict2_multigroup_matches_reference(hip, ref, depth, T, D):
+    """The LDS-staged v2 serving kernel: deep trees spanning MULTIPLE
+    16Ki-node LDS groups (depth 9 x 40 trees = 40,920 nodes = 3 groups)
+    and the D > 1 global-leaf path must match the torch reference."""
+    g = torch.Generator().manual_seed(77)
+    F = 20
+    x = torch.randn(30000, F, generator=g).to(DEV)
+    trees = [_rand_complete_tree(depth, F, D, g) for _ in range(T)]
+    w = torch.rand(T, generator=g) + 0.5
+    got = hip.forest_predict(x, [
+        {k: v.to(DEV) for k, v in t.items()} for t in trees
+    ], w.to(DEV))
+    want = ref.forest_predict(x.cpu(), trees, w)
+    assert torch.allclose(got.cpu(), want, rtol=1e-4, atol=1e-4)
+
+
+def test_hist_build_weighted_wide_multiclass(hip, ref):
+    """weighted K >= 7 one-hot channels (2-channel
+    nonneg tail) previously recursed forever in the C > 8 chunking."""
+    g = torch.Generator().manual_seed(61)
+    n, f, b, d = 15000, 9, 32, 9
+    bins = torch.randint(0, b, (n, f), generator=g, dtype=torch.uint8).to(DEV)
+    # d signed grads + hess + count (non-unit weights -> 2-wide tail)
+    gh = torch.cat(
+        [torch.randn(n, d, generator=g), torch.rand(n, 2, generator=g)], dim=1
+    ).to(DEV)
+    rows = torch.arange(n, dtype=torch.int32).to(DEV)
+    offs = torch.tensor([0, 6000, n])
+    got = hip.hist_build(bins, gh, rows, offs, b, d)
+    want = ref.hist_build(bins.cpu(), gh.cpu(), rows.cpu(), offs, b)
+    assert got.shape == (2, f, b, d + 2)
+    assert torch.allclose(got.cpu(), want, atol=2e-2, rtol=1e-3)
+
+
+def test_weighted_wide_multiclass_tree_gpu():
+    """End-to-end: a weighted 8-class gini tree (BoostingClassifier's
+    reweighted rounds hit exactly this shape)."""
+    import spark_ensemble_amd as sea
+    from spark_ensemble_amd.utils.io import synthetic_classification
+
+    df = synthetic_classification(30000, 12, k=8, seed=13, device=DEV)
+    m = sea.BoostingClassifier().setNumBaseLearners(3).fit(df)
+    acc = float((m.transform(df)["prediction"] == df["label"]).float().mean())
+    assert acc > 1.5 / 8, acc
