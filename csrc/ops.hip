@@ -785,10 +785,17 @@ void forest_predict(torch::Tensor out, torch::Tensor x, torch::Tensor feat,
 //   BaggingRegressor.scala:221-228, GBMClassifier.scala:567-589.
 // ---------------------------------------------------------------------------
 
-template <bool D1>
+// BINNED == false: x is [N, F] f32, payload high-32 = f32 threshold bits.
+// BINNED == true:  x is [N, F] u8 rank-transformed rows (each feature
+//   value replaced by its lower-bound rank within the forest's OWN sorted
+//   per-feature threshold set — an EXACT transform: x <= thr  <=>
+//   rank(x) <= rank(thr)), payload high-32 = the threshold's rank.
+//   Rows shrink 4x (u8 vs f32), so the divergent per-lane gathers stop
+//   thrashing L1 (256 B rows: 4 cache lines instead of 16).
+template <bool D1, bool BINNED>
 __global__ void forest_predict2_kernel(
     float* __restrict__ out,                       // [N, D] (+=, pre-zeroed)
-    const float* __restrict__ x,                   // [N, F]
+    const void* __restrict__ xv_,                  // [N, F] f32 | u8
     const unsigned long long* __restrict__ nodes,  // [total] packed
     const float* __restrict__ leaf,                // [total, D] (D > 1)
     const int* __restrict__ tree_off,              // [T] global node base
@@ -808,58 +815,36 @@ __global__ void forest_predict2_kernel(
   int64_t r = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
   const int64_t stride = gridDim.x * (int64_t)blockDim.x;
   for (; r < n; r += stride) {
-    const float* xr = x + r * F;
+    const float* xr = BINNED ? nullptr : (const float*)xv_ + r * F;
+    const uint8_t* br = BINNED ? (const uint8_t*)xv_ + r * F : nullptr;
     float acc[8];
     if (!D1)
 #pragma unroll
       for (int d = 0; d < 8; ++d) acc[d] = 0.0f;
     float acc1 = 0.0f;
-    // 4 trees walked per lane concurrently: each hop is a dependent
-    // LDS-node + scattered-x load chain (~50 + ~200 cyc); interleaving
-    // four independent chains quadruples the memory-level parallelism
-    // the wave exposes (fully unrolled -> no runtime-indexed locals)
-    for (int t = 0; t < n_trees; t += 4) {
-      const int nt4 = min(4, n_trees - t);
-      int toff[4], node[4], f[4];
-      unsigned long long nd[4];
-#pragma unroll
-      for (int k = 0; k < 4; ++k) {
-        if (k < nt4) {
-          toff[k] = tree_off[first_tree + t + k] - node_base;
-          node[k] = toff[k];
-          nd[k] = tlds[node[k]];
-          f[k] = (short)(nd[k] & 0xFFFFu);
+    for (int t = 0; t < n_trees; ++t) {
+      const int toff = tree_off[first_tree + t] - node_base;
+      int node = toff;
+      unsigned long long nd = tlds[node];
+      int f = (short)(nd & 0xFFFFu);
+      while (f >= 0) {
+        const int left = (short)((nd >> 16) & 0xFFFFu);
+        bool go_left;
+        if (BINNED) {
+          go_left = br[f] <= (unsigned)(nd >> 32);
         } else {
-          f[k] = -1;
+          go_left = xr[f] <= __uint_as_float((unsigned)(nd >> 32));
         }
+        node = toff + left + (go_left ? 0 : 1);
+        nd = tlds[node];
+        f = (short)(nd & 0xFFFFu);
       }
-      bool alive = true;
-      while (alive) {
-        alive = false;
-#pragma unroll
-        for (int k = 0; k < 4; ++k) {
-          if (f[k] >= 0) {
-            const float xv = xr[f[k]];
-            const float thr = __uint_as_float((unsigned)(nd[k] >> 32));
-            const int left = (short)((nd[k] >> 16) & 0xFFFFu);
-            node[k] = toff[k] + left + (xv <= thr ? 0 : 1);
-            nd[k] = tlds[node[k]];
-            f[k] = (short)(nd[k] & 0xFFFFu);
-            alive |= (f[k] >= 0);
-          }
-        }
-      }
-#pragma unroll
-      for (int k = 0; k < 4; ++k) {
-        if (k < nt4) {
-          if (D1) {
-            acc1 += __uint_as_float((unsigned)(nd[k] >> 32));
-          } else {
-            const float wt = w[first_tree + t + k];
-            const float* lv = leaf + (int64_t)(node_base + node[k]) * D;
-            for (int d = 0; d < D; ++d) acc[d] += wt * lv[d];
-          }
-        }
+      if (D1) {
+        acc1 += __uint_as_float((unsigned)(nd >> 32));  // weighted leaf
+      } else {
+        const float wt = w[first_tree + t];
+        const float* lv = leaf + (int64_t)(node_base + node) * D;
+        for (int d = 0; d < D; ++d) acc[d] += wt * lv[d];
       }
     }
     if (D1) {
@@ -879,6 +864,7 @@ void forest_predict2(torch::Tensor out, torch::Tensor x, torch::Tensor nodes,
   CHECK_GPU(out); CHECK_GPU(x); CHECK_GPU(nodes); CHECK_GPU(groups);
   CHECK_CONTIG(out); CHECK_CONTIG(x); CHECK_CONTIG(nodes);
   TORCH_CHECK(D <= 8, "forest_predict2: D <= 8");
+  const bool binned = x.scalar_type() == torch::kUInt8;
   int64_t n = x.size(0);
   int F = (int)x.size(1);
   const int G = (int)groups.size(0);
@@ -890,22 +876,24 @@ void forest_predict2(torch::Tensor out, torch::Tensor x, torch::Tensor nodes,
   // fill the chip even for few groups
   rblocks = std::max(rblocks, (int)std::min<int64_t>(
                                   ceil_div(512, (int64_t)G), ceil_div(n, 64)));
-#define FP2_LAUNCH(DD1)                                                       \
+#define FP2_LAUNCH(DD1, BB)                                                   \
   do {                                                                        \
     if (lds > 65536)                                                          \
       (void)hipFuncSetAttribute(                                              \
-          reinterpret_cast<const void*>(&forest_predict2_kernel<DD1>),        \
+          reinterpret_cast<const void*>(&forest_predict2_kernel<DD1, BB>),    \
           hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);              \
-    hipLaunchKernelGGL((forest_predict2_kernel<DD1>), dim3(rblocks, G),       \
+    hipLaunchKernelGGL((forest_predict2_kernel<DD1, BB>), dim3(rblocks, G),   \
                        dim3(threads), lds, stream, out.data_ptr<float>(),     \
-                       x.data_ptr<float>(),                                   \
+                       x.data_ptr(),                                          \
                        (const unsigned long long*)nodes.data_ptr<int64_t>(),  \
                        leaf.numel() ? leaf.data_ptr<float>() : nullptr,       \
                        tree_off.data_ptr<int>(), w.data_ptr<float>(),         \
                        groups.data_ptr<int>(), n, F, (int)D);                 \
   } while (0)
-  if (D == 1) FP2_LAUNCH(true);
-  else FP2_LAUNCH(false);
+  if (D == 1 && binned) FP2_LAUNCH(true, true);
+  else if (D == 1) FP2_LAUNCH(true, false);
+  else if (binned) FP2_LAUNCH(false, true);
+  else FP2_LAUNCH(false, false);
 #undef FP2_LAUNCH
 }
 

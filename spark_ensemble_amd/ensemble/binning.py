@@ -20,6 +20,32 @@ from ..ops import dispatch as ops
 from .utils import slice_features
 
 
+def apply_categorical_edges_(edges: torch.Tensor, categorical) -> torch.Tensor:
+    """Overwrite quantile cut points with IDENTITY cut points for declared
+    categorical features (reference Utils.getFeaturesMetadata semantics,
+    Utils.scala:42-61: base learners must see categorical info — here the
+    bin id IS the category id, so every split threshold is an exact
+    category boundary).  edges[f] = [0, 1, ..., card-2, +BIG...]; the bin
+    rule 'first edge >= v' then maps category k -> bin k.  Unseen ids
+    > card-1 fall in the top bin."""
+    if not categorical:
+        return edges
+    B1 = edges.shape[1]
+    for f, card in categorical.items():
+        if card - 1 > B1:
+            raise ValueError(
+                f"categorical feature {f} has {card} categories but maxBins "
+                f"allows only {B1 + 1}; raise maxBins"
+            )
+        row = torch.full((B1,), 3.0e38, dtype=edges.dtype, device=edges.device)
+        if card > 1:
+            row[: card - 1] = torch.arange(
+                card - 1, dtype=edges.dtype, device=edges.device
+            )
+        edges[f] = row
+    return edges
+
+
 def _is_identity(indices: Optional[torch.Tensor], num_features: int) -> bool:
     if indices is None:
         return True
@@ -45,6 +71,8 @@ class BinnedDataset:
     def get(self, max_bins: int) -> Tuple[torch.Tensor, torch.Tensor]:
         if max_bins not in self._by_bins:
             edges = ops.quantile_bins(self.x, max_bins)
+            if self._frame is not None:
+                apply_categorical_edges_(edges, self._frame.categorical)
             from ..parallel import get_comm
 
             comm = get_comm()

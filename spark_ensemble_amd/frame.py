@@ -49,10 +49,31 @@ class TensorFrame:
         # can never alias stale entries.  Propagated (same dict object)
         # through column-level transforms.
         self._cache: Dict = {}
+        # Per-feature metadata (reference Utils.getFeaturesMetadata,
+        # Utils.scala:42-61 — AttributeGroup column metadata carrying
+        # categorical info): {feature_index: cardinality}.  Categorical
+        # features hold integer category ids 0..card-1 in the float
+        # features tensor (Spark ML's indexed-category convention) and
+        # are binned by IDENTITY instead of quantiles, so tree splits land
+        # on exact category-id boundaries (ordinal-categorical handling;
+        # propagation through subspace slices is free because the sliced
+        # edge rows travel with the feature — ensemble/binning.py).
+        self._categorical: Dict[int, int] = {}
 
     def _with_cache_of(self, other: "TensorFrame") -> "TensorFrame":
         self._cache = other._cache
+        self._categorical = other._categorical
         return self
+
+    # -- feature metadata -------------------------------------------------
+    def set_categorical(self, categorical: Dict[int, int]) -> "TensorFrame":
+        """Declare categorical features: {feature_index: num_categories}."""
+        self._categorical = {int(k): int(v) for k, v in categorical.items()}
+        return self
+
+    @property
+    def categorical(self) -> Dict[int, int]:
+        return self._categorical
 
     def cache_get(self, kind: str, src: torch.Tensor, key):
         ent = self._cache.get(kind)

@@ -56,6 +56,10 @@ def ensure_binned(frame, x: torch.Tensor, max_bins: int):
     if cached is not None:
         return cached
     edges = ops.quantile_bins(x, max_bins)
+    if frame is not None and frame.categorical:
+        from ..ensemble.binning import apply_categorical_edges_
+
+        apply_categorical_edges_(edges, frame.categorical)
     from ..parallel import get_comm
 
     comm = get_comm()

@@ -335,16 +335,96 @@ def _pack_forest(trees, weights, dev):
         groups_t=torch.tensor(groups, dtype=torch.int32, device=dev),
         max_nodes=max_nodes,
     )
+    _pack_forest_binned(pack, feats, thrs, payload)
     return pack
+
+
+def _pack_forest_binned(pack, feats, thrs, payload):
+    """Rank-transform serving structures (EXACT): collect each feature's
+    sorted unique threshold set; a node's threshold becomes its RANK in
+    that set, and at predict time every row is rank-transformed once via
+    the bin_features kernel (lower_bound against the padded set), after
+    which  x <= thr  <=>  rank(x) <= rank(thr)  on u8 bins — rows shrink
+    4x for the divergent walk gathers.  Disabled when any feature carries
+    > 255 distinct thresholds (u8 overflow)."""
+    dev = pack["dev"]
+    mask = feats >= 0
+    m_cnt = int(mask.sum())
+    if m_cnt == 0:
+        return
+    f_int = feats[mask].long()
+    t_val = thrs[mask]
+    f_max = int(f_int.max()) + 1
+    # sort by (feature, threshold)
+    order = torch.argsort(t_val)
+    f_s, t_s = f_int[order], t_val[order]
+    order2 = torch.argsort(f_s, stable=True)
+    p = order[order2]
+    f_s, t_s = f_s[order2], t_s[order2]
+    keep = torch.ones_like(f_s, dtype=torch.bool)
+    keep[1:] = (f_s[1:] != f_s[:-1]) | (t_s[1:] != t_s[:-1])
+    uidx = keep.long().cumsum(0) - 1
+    fu, tu = f_s[keep], t_s[keep]
+    counts = torch.bincount(fu, minlength=f_max)
+    maxcnt = int(counts.max())
+    if maxcnt > 255:
+        return
+    offs = torch.zeros(f_max + 1, dtype=torch.long, device=dev)
+    offs[1:] = counts.cumsum(0)
+    rank_sorted = uidx - offs[f_s]
+    rank_node = torch.empty(m_cnt, dtype=torch.long, device=dev)
+    rank_node[p] = rank_sorted
+    rank_full = torch.zeros(feats.numel(), dtype=torch.int64, device=dev)
+    rank_full[mask] = rank_node
+    payload_bits = torch.where(
+        mask,
+        rank_full,
+        payload.view(torch.int32).to(torch.int64) & 0xFFFFFFFF,
+    )
+    node64b = (
+        (feats.to(torch.int64) & 0xFFFF)
+        | ((pack["lefts"].to(torch.int64) & 0xFFFF) << 16)
+        | (payload_bits << 32)
+    ).contiguous()
+    pack.update(
+        binned_ok=True, node64b=node64b, bin_fu=fu, bin_tu=tu,
+        bin_offs=offs, bin_maxcnt=max(maxcnt, 1), bin_fmax=f_max,
+    )
+
+
+def _binned_edges(pack, F):
+    """[F, maxcnt] padded per-feature threshold sets for bin_features
+    (pad = +3e38 so lower_bound is unaffected); cached per F."""
+    cached = pack.get("bedges")
+    if cached is not None and cached.shape[0] == F:
+        return cached
+    dev = pack["dev"]
+    mc = pack["bin_maxcnt"]
+    edges = torch.full((F, mc), 3.0e38, dtype=torch.float32, device=dev)
+    fu, tu, offs = pack["bin_fu"], pack["bin_tu"], pack["bin_offs"]
+    col = torch.arange(fu.numel(), device=dev) - offs[fu]
+    edges[fu, col] = tu
+    pack["bedges"] = edges
+    return edges
 
 
 def _forest_predict_packed(m, x, pack):
     D = pack["D"]
     out = torch.zeros(x.shape[0], D, dtype=torch.float32, device=x.device)
     if pack["v2"] and x.shape[1] < 32768 and hasattr(m, "forest_predict2"):
-        m.forest_predict2(out, x.contiguous(), pack["node64"], pack["leaves"],
-                          pack["offsets32"], pack["w"], pack["groups_t"], D,
-                          pack["max_nodes"])
+        xc = x.contiguous()
+        if (pack.get("binned_ok") and pack["bin_fmax"] <= x.shape[1]
+                and os.environ.get("SEA_SERVE_RAW") != "1"):
+            edges = _binned_edges(pack, x.shape[1])
+            xb = torch.empty(x.shape, dtype=torch.uint8, device=x.device)
+            m.bin_features(xb, xc, edges)
+            m.forest_predict2(out, xb, pack["node64b"], pack["leaves"],
+                              pack["offsets32"], pack["w"], pack["groups_t"],
+                              D, pack["max_nodes"])
+        else:
+            m.forest_predict2(out, xc, pack["node64"], pack["leaves"],
+                              pack["offsets32"], pack["w"], pack["groups_t"],
+                              D, pack["max_nodes"])
     else:
         m.forest_predict(out, x.contiguous(), pack["feats"], pack["thrs"],
                          pack["lefts"], pack["leaves"], pack["offsets32"],

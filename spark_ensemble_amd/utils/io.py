@@ -127,18 +127,32 @@ def from_pandas(df, features_cols=None, label_col="label", weight_col=None,
                 device=None) -> TensorFrame:
     """Build a TensorFrame from a pandas DataFrame (the on-ramp for users
     coming from Spark DataFrames).  ``features_cols`` defaults to every
-    numeric column except the label/weight."""
-    import pandas as pd  # noqa: F401
+    numeric or categorical column except the label/weight.  pandas
+    ``category`` columns are factorized to 0-based codes and recorded as
+    categorical feature metadata (reference Utils.getFeaturesMetadata,
+    Utils.scala:42-61), so tree splits land on exact category ids."""
+    import pandas as pd
+
+    def _usable(c):
+        return isinstance(df[c].dtype, pd.CategoricalDtype) or np.issubdtype(
+            df[c].dtype, np.number
+        )
 
     if features_cols is None:
         features_cols = [
             c for c in df.columns
-            if c not in (label_col, weight_col)
-            and np.issubdtype(df[c].dtype, np.number)
+            if c not in (label_col, weight_col) and _usable(c)
         ]
-    x = torch.from_numpy(
-        np.ascontiguousarray(df[features_cols].to_numpy(dtype=np.float32))
-    )
+    mats = []
+    categorical = {}
+    for j, c in enumerate(features_cols):
+        if isinstance(df[c].dtype, pd.CategoricalDtype):
+            codes = df[c].cat.codes.to_numpy()
+            categorical[j] = len(df[c].cat.categories)
+            mats.append(codes.astype(np.float32))
+        else:
+            mats.append(df[c].to_numpy(dtype=np.float32))
+    x = torch.from_numpy(np.ascontiguousarray(np.stack(mats, axis=1)))
     cols = {"features": x}
     if label_col in df.columns:
         cols["label"] = torch.from_numpy(
@@ -149,6 +163,8 @@ def from_pandas(df, features_cols=None, label_col="label", weight_col=None,
             df[weight_col].to_numpy(dtype=np.float32)
         )
     fr = TensorFrame(cols)
+    if categorical:
+        fr.set_categorical(categorical)
     return fr.to(device) if device is not None else fr
 
 

@@ -54,11 +54,19 @@ def main():
     m = dispatch._load_hip()
     assert m is not None
 
-    # v2 packed + cached
+    # v2 packed + cached (binned rank-transform mode when available)
     cache = {}
     t_first = timeit(lambda: dispatch.forest_predict(x, trees, w, cache=cache), reps=1)
     t2 = timeit(lambda: dispatch.forest_predict(x, trees, w, cache=cache))
     out2 = dispatch.forest_predict(x, trees, w, cache=cache)
+
+    # v2 raw-float mode
+    import os
+    os.environ["SEA_SERVE_RAW"] = "1"
+    t2raw = timeit(lambda: dispatch.forest_predict(x, trees, w, cache=cache))
+    out2raw = dispatch.forest_predict(x, trees, w, cache=cache)
+    del os.environ["SEA_SERVE_RAW"]
+    err_raw = float((out2raw - out2).abs().max())
 
     # v1 via the packed arena fallback entry
     pack = cache["pack"]
@@ -74,12 +82,15 @@ def main():
 
     print(json.dumps({
         "probe": "serve", "rows": n, "trees": T, "depth": depth,
-        "v2_ms": round(t2 * 1000, 2),
+        "v2_binned_ms": round(t2 * 1000, 2),
         "v2_first_call_ms": round(t_first * 1000, 2),
+        "v2_raw_ms": round(t2raw * 1000, 2),
         "v1_ms": round(t1 * 1000, 2),
-        "v2_rows_per_sec": round(n / t2),
+        "v2_binned_rows_per_sec": round(n / t2),
+        "v2_raw_rows_per_sec": round(n / t2raw),
         "v1_rows_per_sec": round(n / t1),
-        "max_abs_diff": err,
+        "binned_vs_raw_max_diff": err_raw,
+        "v1_vs_v2_max_diff": err,
     }))
 
 
