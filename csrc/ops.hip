@@ -213,10 +213,16 @@ void bin_features(torch::Tensor out, torch::Tensor x, torch::Tensor edges) {
 //   |sum| < 2^30 per chunk per bin (no field overflow, low->high carries
 //   impossible since low fields are non-negative and bounded).
 //
-//   chunks: int32 [n_chunks, 4] = (node, start, len, single_chunk_flag);
-//   quantization scales ride in the same upload (one H2D per launch).
+//   chunks: int32 [n_chunks, 5] = (node, start, len, single_chunk_flag,
+//   col0); quantization scales ride in the same upload (one H2D per
+//   launch).  ``col0`` is the node's base column in a gh matrix of row
+//   stride CH — 0/CH==C for single-tree builds; a FOREST build (K trees
+//   or K classes grown level-synchronously, reference
+//   GBMClassifier.scala:377-411 parallel futures) interleaves per-tree
+//   channel groups [g_t, h_t(, cnt_t)] and sets col0 = tree * C, so one
+//   launch histograms every active node of every tree.
 //   Flush converts back to f32 into out [n_nodes, F, B, C]; single-chunk
-//   nodes use plain stores, multi-chunk nodes f32 global atomics.
+//   nodes use plain stores, multi-chunk nodes integer staging atomics.
 // ---------------------------------------------------------------------------
 
 template <int DC, int NC>
@@ -224,15 +230,15 @@ __global__ void hist_build_kernel(
     float* __restrict__ out,            // [n_nodes, F, B, C]
     long long* __restrict__ stage,      // [n_nodes, F, B, C] i64 (multi-chunk)
     const uint8_t* __restrict__ bins,   // [N, F]
-    const float* __restrict__ gh,       // [N, C]
+    const float* __restrict__ gh,       // [N, CH]
     const int* __restrict__ row_idx,    // [M]
-    const int* __restrict__ chunks,     // [2*C + n_chunks*4] (scales first)
-    int F, int B, int FG, int identity_rows) {
+    const int* __restrict__ chunks,     // [2*C + n_chunks*5] (scales first)
+    int F, int B, int FG, int CH, int identity_rows) {
   constexpr int C = DC + NC;
   constexpr int CELLS = DC > NC ? DC : NC;
   extern __shared__ unsigned long long lds64[];  // FG * B * CELLS
   const float* scales = reinterpret_cast<const float*>(chunks);
-  const int* chk = chunks + 2 * C + blockIdx.x * 4;
+  const int* chk = chunks + 2 * C + blockIdx.x * 5;
   const int fg = blockIdx.y;
   const int f0 = fg * FG;
   const int nf = min(FG, F - f0);
@@ -240,6 +246,7 @@ __global__ void hist_build_kernel(
   const int start = chk[1];
   const int len = chk[2];
   const int single = chk[3];
+  const int col0 = chk[4];
 
   const int lds_cells = FG * B * CELLS;
   for (int i = threadIdx.x; i < lds_cells; i += blockDim.x) lds64[i] = 0ull;
@@ -253,7 +260,7 @@ __global__ void hist_build_kernel(
 
   for (int i = threadIdx.x; i < len; i += blockDim.x) {
     const int r = identity_rows ? start + i : row_idx[start + i];
-    const float* g = gh + (int64_t)r * C;
+    const float* g = gh + (int64_t)r * CH + col0;
     // quantize once per row, pack one u64 addend per cell
     unsigned long long addend[CELLS];
 #pragma unroll
@@ -356,14 +363,21 @@ __global__ void hist_decode_kernel(float* __restrict__ out,
 void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
                 torch::Tensor row_idx, torch::Tensor node_offsets,
                 int64_t num_bins, int64_t d_dims, torch::Tensor max_abs,
-                bool identity_rows) {
+                bool identity_rows, torch::Tensor node_col0,
+                int64_t c_per_node) {
   CHECK_GPU(out); CHECK_GPU(bins); CHECK_GPU(gh); CHECK_GPU(row_idx);
   CHECK_CONTIG(out); CHECK_CONTIG(bins); CHECK_CONTIG(gh); CHECK_CONTIG(row_idx);
   TORCH_CHECK(!node_offsets.is_cuda(), "node_offsets stays on host");
   TORCH_CHECK(!max_abs.is_cuda(), "max_abs stays on host");
   const int F = (int)bins.size(1);
   const int B = (int)num_bins;
-  const int C = (int)gh.size(1);
+  const int CH = (int)gh.size(1);  // gh row stride
+  // channels PER NODE: gh width for single-tree builds; explicit for
+  // forest builds where gh interleaves per-tree channel groups
+  const int C = c_per_node > 0 ? (int)c_per_node : CH;
+  const bool forest = node_col0.numel() > 0;
+  TORCH_CHECK(!forest || node_col0.numel() == node_offsets.numel() - 1,
+              "node_col0 must have one entry per node");
   const int D = d_dims > 0 ? (int)d_dims : (C >= 3 ? C - 2 : C - 1);
   const int NN = C - D;
   TORCH_CHECK(C <= 8, "gh channels capped at 8 (grad dims + hess + count)");
@@ -441,6 +455,7 @@ void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
     s = std::min(s, (double)(1u << 30) / (double)m);
     scales_f[c] = (float)s;
   }
+  const int* col0_p = forest ? node_col0.data_ptr<int>() : nullptr;
   std::vector<int> multi_nodes;
   for (int nd = 0; nd < n_nodes; ++nd) {
     int64_t s = offs[nd], e = offs[nd + 1];
@@ -451,10 +466,11 @@ void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
       chunk_v.push_back((int)c);
       chunk_v.push_back((int)std::min<int64_t>(chunk_rows, e - c));
       chunk_v.push_back(single);
+      chunk_v.push_back(col0_p ? col0_p[nd] : 0);
     }
   }
   if ((int)chunk_v.size() == 2 * C) return;
-  const int n_chunks = (int)((chunk_v.size() - 2 * C) / 4);
+  const int n_chunks = (int)((chunk_v.size() - 2 * C) / 5);
   auto chunks_b = h2d_async(chunk_v.data(), chunk_v.size() * 4, 0,
                             bins.device());
   auto chunks = chunks_b.view(torch::kInt32);
@@ -482,7 +498,7 @@ void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
                        stream, out.data_ptr<float>(), stage_ptr,             \
                        bins.data_ptr<uint8_t>(),                             \
                        gh.data_ptr<float>(), row_idx.data_ptr<int>(),        \
-                       chunks.data_ptr<int>(), F, B, FG,                     \
+                       chunks.data_ptr<int>(), F, B, FG, CH,                 \
                        identity_rows ? 1 : 0);                               \
   } while (0)
   const int key = D * 10 + NN;

@@ -75,19 +75,28 @@ class BaggingClassifier(ProbabilisticClassifier, _BaggingClassifierParams):
         subspaces = [
             subspace(self.getSubspaceRatio(), num_features, seed + i) for i in range(k)
         ]
-        models = []
-        for i in range(k):
-            bag_w = self.sample_weights(
-                self.getReplacement(),
-                self.getSubsampleRatio(),
-                n,
-                seed + i,
-                x.device,
-                w,
-                comm.rank,
-            )
-            fr = binned.fit_frame(learner, y, bag_w, subspaces[i])
-            models.append(self.fit_base_learner(learner, fr, weight_col="weight"))
+        if learner is not None and learner.hasParam("maxBins"):
+            binned.get(int(learner.getOrDefault("maxBins")))  # pre-warm once
+
+        def fit_one(i):
+            def task():
+                bag_w = self.sample_weights(
+                    self.getReplacement(),
+                    self.getSubsampleRatio(),
+                    n,
+                    seed + i,
+                    x.device,
+                    w,
+                    comm.rank,
+                )
+                fr = binned.fit_frame(learner, y, bag_w, subspaces[i])
+                return self.fit_base_learner(learner, fr, weight_col="weight")
+            return task
+
+        from ..parallel.streams import parallel_fits
+
+        models = parallel_fits([fit_one(i) for i in range(k)],
+                               self.getParallelism())
 
         model = BaggingClassificationModel()
         model._models = models

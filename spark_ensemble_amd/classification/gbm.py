@@ -4,9 +4,10 @@ Re-creates reference classification/GBMClassifier.scala:219-496:
   * losses {logloss, exponential, bernoulli} (default logloss); LogLoss has
     dim = K, so each round fits K base regressors on per-class
     pseudo-residuals (reference fits them in parallel Futures :377-411;
-    here they are fitted sequentially — each tree fit is itself a fully
-    GPU-parallel histogram pass, so there is no idle hardware to recover
-    the way the reference's driver threads do),
+    here, for the built-in tree learner on full bags, all K trees grow
+    LEVEL-SYNCHRONOUSLY in fused kernel launches — one histogram build and
+    one RCCL all-reduce per level for the whole round,
+    tree_grower.grow_forest; generic base learners fit sequentially),
   * init {prior, uniform}; binary + dim-1 prior -> constant log-odds model
     (reference :275-283),
   * newton pseudo-residuals with per-dim hessian floor 1e-2 and weights
@@ -32,7 +33,7 @@ we keep the mathematically consistent orientation.
 from __future__ import annotations
 
 import os
-from typing import List
+from typing import List, Optional
 
 import numpy as np
 import torch
@@ -188,6 +189,18 @@ class GBMClassifier(ProbabilisticClassifier, _GBMClassifierParams):
             instr.log_named_value("resumed_from_round", r0)
 
         instr.timers["setup_ms"] = instr.elapsed_ms()
+
+        # fused-round eligibility that depends only on params (checked
+        # once); the per-round weight positivity check needs a sync and
+        # is done lazily inside _can_fuse_round
+        from ..models.tree import DecisionTreeRegressor
+
+        self._fuse_static_ok = (
+            type(learner) is DecisionTreeRegressor
+            and learner.getOrDefault("minWeightFractionPerNode") == 0.0
+        )
+        self._fuse_w_ok: Optional[bool] = None
+
         while i < k_stages and v < self.getOrDefault("numRounds"):
             idx = subspaces[i]
             xs = binned.sliced_features(idx)
@@ -215,23 +228,49 @@ class GBMClassifier(ProbabilisticClassifier, _GBMClassifierParams):
                 res_label = -g
                 res_weight = bag_w.unsqueeze(1).expand(-1, dim)
 
-            # K per-class base-regressor fits (reference parallel Futures
-            # :377-411; here sequential per class — each fit is itself a
-            # fully GPU-parallel histogram pass)
-            imodels = []
-            for j in range(dim):
-                fr = binned.fit_frame(
-                    learner, res_label[:, j].contiguous(),
-                    res_weight[:, j].contiguous(), idx, xs,
+            # K per-class base-regressor fits.  The reference runs these
+            # as driver-side parallel futures (:377-411); here, when the
+            # base learner is the built-in histogram tree, ALL K trees
+            # grow level-synchronously in fused kernel launches (ONE
+            # histogram build + ONE all-reduce per level for the whole
+            # round — tree_grower.grow_forest).  Generic base learners
+            # keep the sequential per-class path.
+            directions = None
+            if dim > 1 and self._can_fuse_round(learner, res_weight):
+                from ..models.tree import fit_tree_forest
+
+                f_edges, f_bins = binned.sliced_binned(
+                    idx, learner.getOrDefault("maxBins")
                 )
-                imodels.append(self.fit_base_learner(learner, fr, weight_col="weight"))
+                imodels, directions = fit_tree_forest(
+                    learner, f_edges, f_bins, res_label, res_weight, comm
+                )
+            else:
+                from ..parallel.streams import parallel_fits
 
-            def _dir(m):
-                tp = getattr(m, "_train_pred", None)
-                return tp if tp is not None and tp.shape[0] == xs.shape[0] \
-                    else m.predict(xs)
+                def class_task(j):
+                    def task():
+                        fr = binned.fit_frame(
+                            learner, res_label[:, j].contiguous(),
+                            res_weight[:, j].contiguous(), idx, xs,
+                        )
+                        return self.fit_base_learner(
+                            learner, fr, weight_col="weight"
+                        )
+                    return task
 
-            directions = torch.stack([_dir(m) for m in imodels], dim=1)
+                imodels = parallel_fits(
+                    [class_task(j) for j in range(dim)],
+                    self.getOrDefault("parallelism"), warm_first=True,
+                )
+
+            if directions is None:
+                def _dir(m):
+                    tp = getattr(m, "_train_pred", None)
+                    return tp if tp is not None and tp.shape[0] == xs.shape[0] \
+                        else m.predict(xs)
+
+                directions = torch.stack([_dir(m) for m in imodels], dim=1)
 
             if optimized:
                 if dim == 1:
@@ -298,6 +337,19 @@ class GBMClassifier(ProbabilisticClassifier, _GBMClassifierParams):
         ):
             model.set(p, self.getOrDefault(p))
         return model
+
+    def _can_fuse_round(self, learner, res_weight) -> bool:
+        """Fused K-tree rounds need: built-in tree learner, no per-tree
+        weight thresholds, and strictly positive weights on every row
+        (the fused grower has no zero-weight row mask).  Sub-sampling
+        introduces zeros, so only the full-bag configs fuse."""
+        if not getattr(self, "_fuse_static_ok", False):
+            return False
+        if self.getSubsampleRatio() < 1.0 or self.getReplacement():
+            return False
+        if self._fuse_w_ok is None:
+            self._fuse_w_ok = bool((res_weight > 0).all())
+        return self._fuse_w_ok
 
     def _save_impl(self, path: str):
         persistence.save_metadata(self, path)

@@ -83,30 +83,50 @@ class StackingClassifier(Predictor, _StackingClassifierParams):
         wcol = _check_weight_support(learners, stacker, self.getWeightCol())
         use_w = w if wcol else torch.ones_like(w)
 
+        from ..parallel.streams import parallel_fits
+
+        # shared derived-data cache: features binned once for every
+        # fold/final fit (see regression/stacking.py)
+        shared = TensorFrame(features=x, label=y, weight=use_w)
+
         if self.getOrDefault("inSample"):
-            models = [
-                self.fit_base_learner(
-                    lr,
-                    TensorFrame(features=x, label=y, weight=use_w),
-                    weight_col="weight",
-                )
-                for lr in learners
-            ]
+            models = parallel_fits(
+                [
+                    (lambda lr=lr: self.fit_base_learner(
+                        lr, shared, weight_col="weight"))
+                    for lr in learners
+                ],
+                self.getParallelism(), warm_first=True,
+            )
             meta = _meta_features_clf(models, x, method)
         else:
             num_folds = self.getOrDefault("numFolds")
             g = torch.Generator().manual_seed(self.getOrDefault("seed"))
             fold = torch.randint(0, num_folds, (n,), generator=g).to(x.device)
-            meta_cols: List[torch.Tensor] = []
-            for lr in learners:
-                col = None
-                for f in range(num_folds):
+
+            def fold_task(lr, f):
+                def task():
                     wmask = use_w * (fold != f).float()
-                    m = self.fit_base_learner(
-                        lr,
-                        TensorFrame(features=x, label=y, weight=wmask),
+                    return self.fit_base_learner(
+                        lr, shared.withColumn("weight", wmask),
                         weight_col="weight",
                     )
+                return task
+
+            tasks = [fold_task(lr, f)
+                     for lr in learners for f in range(num_folds)]
+            tasks += [
+                (lambda lr=lr: self.fit_base_learner(
+                    lr, shared, weight_col="weight"))
+                for lr in learners
+            ]
+            fitted = parallel_fits(tasks, self.getParallelism(),
+                                   warm_first=True)
+            meta_cols: List[torch.Tensor] = []
+            for mi, lr in enumerate(learners):
+                col = None
+                for f in range(num_folds):
+                    m = fitted[mi * num_folds + f]
                     sel = fold == f
                     part = _model_meta(m, x[sel], method)
                     if col is None:
@@ -116,14 +136,7 @@ class StackingClassifier(Predictor, _StackingClassifierParams):
                     col[sel] = part
                 meta_cols.append(col)
             meta = torch.cat(meta_cols, dim=1)
-            models = [
-                self.fit_base_learner(
-                    lr,
-                    TensorFrame(features=x, label=y, weight=use_w),
-                    weight_col="weight",
-                )
-                for lr in learners
-            ]
+            models = fitted[len(learners) * num_folds:]
 
         stack = self.fit_base_learner(
             stacker,

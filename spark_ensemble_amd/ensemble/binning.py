@@ -95,6 +95,18 @@ class BinnedDataset:
             return self.x
         return slice_features(self.x, indices)
 
+    def sliced_binned(self, indices: Optional[torch.Tensor], max_bins: int):
+        """(edges, bins) for a feature subspace — the pre-binned inputs a
+        fused forest fit consumes directly (fit_tree_forest)."""
+        edges, bins = self.get(max_bins)
+        if _is_identity(indices, self.x.shape[1]):
+            return edges, bins
+        idx_dev = indices.to(bins.device)
+        return (
+            edges.index_select(0, idx_dev).contiguous(),
+            bins.index_select(1, idx_dev).contiguous(),
+        )
+
     def fit_frame(
         self,
         learner,

@@ -270,6 +270,51 @@ class DecisionTreeRegressionModel(_TreeModelMixin, RegressionModel, _TreeParams)
         return self._predict_values(features.float()).squeeze(1)
 
 
+def fit_tree_forest(learner, edges, bins, labels, weights, comm=None):
+    """T independent ``DecisionTreeRegressor`` fits as ONE fused forest
+    grow (tree_grower.grow_forest): the MI355X form of the reference's
+    per-class / per-learner fit futures.  ``labels`` [N, T]; ``weights``
+    [N] shared or [N, T] per-tree, all strictly positive (callers
+    fall back to sequential fits when zero weights exist — the fused
+    grower has no row mask).  Returns (models, train_pred [N, T]).
+
+    Caller contract: ``learner`` is a plain DecisionTreeRegressor with
+    minWeightFractionPerNode == 0 (per-tree total weights would otherwise
+    need per-tree thresholds)."""
+    from ..parallel import get_comm as _gc
+    from .tree_grower import grow_forest
+
+    comm = comm or _gc()
+    w2 = weights if weights.dim() == 2 else weights.unsqueeze(1)
+    grads = (labels * w2).contiguous()
+    # one fused stats sync: slot-wise quantization maxima + unit check
+    stats = torch.stack([
+        grads.abs().max(),
+        weights.abs().max(),
+        (weights != 1).sum().to(torch.float32),
+    ])
+    if comm.is_distributed:
+        comm.all_reduce_(stats, "max")  # max of the unit-check is fine: >0 anywhere means not-unit
+    stats_c = stats.cpu()
+    hic = float(stats_c[2]) == 0.0
+    gh_max = torch.tensor(
+        [float(stats_c[0]), float(stats_c[1])] + ([] if hic else [1.0])
+    )
+    gp = learner._grow_params(1.0)  # minWeightFraction==0 ⇒ total unused
+    tp_out: list = []
+    trees = grow_forest(bins, edges, grads, weights, gp, comm,
+                        hess_is_count=hic, train_pred_out=tp_out,
+                        gh_max_in=gh_max)
+    models = []
+    for t, tree in enumerate(trees):
+        m = DecisionTreeRegressionModel()
+        m._set_tree(tree, bins.shape[1])
+        m._train_pred = tp_out[0][:, t]
+        m._copy_cols_from(learner)
+        models.append(m)
+    return models, tp_out[0]
+
+
 class DecisionTreeClassifier(ProbabilisticClassifier, _TreeParams):
     def _fit(self, dataset: TensorFrame) -> "DecisionTreeClassificationModel":
         x, y, w = self._extract_xyw(dataset)

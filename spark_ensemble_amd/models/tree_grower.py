@@ -396,6 +396,326 @@ def grow_tree(
     return {k: _to_dev_async(v, device) for k, v in tree.items()}
 
 
+MAX_FUSED_TREES = 32  # bounds the [T*N] row-index arena per fused batch
+
+
+def grow_forest(
+    bins: torch.Tensor,  # [N, F] uint8
+    edges: torch.Tensor,  # [F, B-1] f32
+    grads: torch.Tensor,  # [N, T] f32 per-tree weighted targets
+    hess: torch.Tensor,  # [N] shared or [N, T] per-tree weights
+    params: GrowParams,
+    comm: Optional[Comm] = None,
+    hess_is_count: Optional[bool] = None,
+    train_pred_out: Optional[list] = None,
+    gh_max_in: Optional[torch.Tensor] = None,
+) -> List[Dict[str, torch.Tensor]]:
+    """Grow T single-output trees LEVEL-SYNCHRONOUSLY in fused launches.
+
+    The MI355X replacement for the reference's driver-side parallel fits
+    (the K per-class futures of one GBM round, GBMClassifier.scala:377-411,
+    and the per-learner futures of Bagging, BaggingRegressor.scala:145-166,
+    when every member sees the identity subspace): every active node of
+    every tree lands in ONE hist_build launch per level (per-node channel
+    selection via gh column offsets), one split_argmax, one partition, and
+    — critically for multi-GPU — ONE histogram all-reduce per level
+    instead of T.
+
+    grads[:, t] is tree t's weighted regression target (w_t * y_t), hess
+    its weight column (shared tensor [N] when all trees reweight rows the
+    same way).  Returns T tree dicts (same schema as grow_tree); when
+    ``train_pred_out`` is a list, appends the [N, T] training-row
+    predictions (leaf scatter).
+    """
+    device = bins.device
+    N, F = bins.shape
+    T = grads.shape[1]
+    B = params.max_bins
+
+    if T > MAX_FUSED_TREES:
+        out: List[Dict[str, torch.Tensor]] = []
+        preds = [] if train_pred_out is not None else None
+        for s in range(0, T, MAX_FUSED_TREES):
+            sl = slice(s, min(s + MAX_FUSED_TREES, T))
+            h_sl = hess if hess.dim() == 1 else hess[:, sl].contiguous()
+            sub_pred = [] if preds is not None else None
+            out.extend(grow_forest(
+                bins, edges, grads[:, sl].contiguous(), h_sl, params, comm,
+                hess_is_count, sub_pred, gh_max_in,
+            ))
+            if preds is not None:
+                preds.append(sub_pred[0])
+        if preds is not None:
+            train_pred_out.append(torch.cat(preds, dim=1))
+        return out
+
+    h_shared = hess.dim() == 1
+    if hess_is_count is None:
+        hess_is_count = bool((hess == 1).all())
+    C = 2 if hess_is_count else 3
+    parts = [grads, hess.unsqueeze(1).expand(N, T) if h_shared else hess]
+    if C == 3:
+        parts.append(torch.ones(N, T, dtype=torch.float32, device=device))
+    gh = torch.stack(parts, dim=2).reshape(N, T * C).contiguous()
+
+    if gh_max_in is not None:
+        gh_max = gh_max_in  # slot-wise [C] host tensor
+    elif bins.is_cuda:
+        gm = [float(grads.abs().max()), float(hess.max())]
+        if C == 3:
+            gm.append(1.0)
+        gh_max = torch.tensor(gm)
+    else:
+        gh_max = None
+
+    assert T * N < 2**31, "fused forest row arena exceeds int32"
+    row_idx = torch.arange(N, dtype=torch.int32, device=device).repeat(T)
+
+    # per-tree flat node arrays
+    feats = [[] for _ in range(T)]
+    thrs = [[] for _ in range(T)]
+    lefts = [[] for _ in range(T)]
+    leaves = [[] for _ in range(T)]
+    fi = torch.zeros(T, F, dtype=torch.float64)
+
+    def alloc_nodes(t: int, k: int) -> int:
+        start = len(feats[t])
+        feats[t].extend([-1] * k)
+        thrs[t].extend([0.0] * k)
+        lefts[t].extend([-1] * k)
+        leaves[t].extend([None] * k)
+        return start
+
+    # root totals: one fused reduction (+ one all-reduce) for all trees
+    g_sum = grads.sum(dim=0)  # [T]
+    if h_shared:
+        h_sum = hess.sum().reshape(1).expand(T)
+    else:
+        h_sum = hess.sum(dim=0)
+    cols = [g_sum, h_sum]
+    if C == 3:
+        cols.append(torch.full((T,), float(N), device=device))
+    root_tot = torch.stack(cols, dim=1)  # [T, C] device
+    if comm is not None:
+        comm.all_reduce_(root_tot)
+    totals = root_tot.cpu()
+
+    # active level state (across all trees)
+    node_ids: List[Tuple[int, int]] = [(t, alloc_nodes(t, 1)) for t in range(T)]
+    offsets = torch.arange(0, (T + 1) * N, N, dtype=torch.int64)
+    hists: Optional[torch.Tensor] = None
+    parent_of: List[int] = []
+    built_mask: List[bool] = []
+
+    edges_cpu = edges.cpu()
+    idx_c = C - 1
+
+    pred_rows: List[List[torch.Tensor]] = [[] for _ in range(T)]
+    pred_vals: List[List[float]] = [[] for _ in range(T)]
+    capture = train_pred_out is not None
+
+    def _capture_leaves(act_idx, offs, ridx):
+        if not capture:
+            return
+        ol = offs.tolist() if isinstance(offs, torch.Tensor) else offs
+        for i in act_idx:
+            s0, e0 = int(ol[i]), int(ol[i + 1])
+            if e0 > s0:
+                t, nid = node_ids[i]
+                pred_rows[t].append(ridx[s0:e0])
+                pred_vals[t].append(float(leaves[t][nid]))
+
+    def _finalize(act_idx, tots):
+        tt = tots.cpu() if isinstance(tots, torch.Tensor) else tots
+        for j, i in enumerate(act_idx):
+            t, nid = node_ids[i]
+            g = float(tt[j, 0])
+            h = float(tt[j, 1])
+            leaves[t][nid] = (
+                g / (h + params.lam) if h + params.lam > 0 else 0.0
+            )
+
+    def _col0():
+        return torch.tensor([t * C for (t, _) in node_ids], dtype=torch.int32)
+
+    for depth in range(params.max_depth + 1):
+        n_active = len(node_ids)
+        if n_active == 0:
+            break
+        if depth == params.max_depth:
+            _finalize(range(n_active), totals)
+            _capture_leaves(range(n_active), offsets, row_idx)
+            break
+
+        # ----- fused histograms for this level ---------------------------
+        if hists is None:
+            new_h = ops.hist_build_forest(
+                bins, gh, row_idx, offsets, _col0(), B, C, gh_max,
+            )
+            if comm is not None:
+                comm.all_reduce_(new_h)
+            hists = new_h
+        else:
+            built_idx = [j for j in range(n_active) if built_mask[j]]
+            child_hists = torch.zeros(
+                n_active, F, B, C, dtype=torch.float32, device=device
+            )
+            if built_idx:
+                off_list = offsets.tolist()
+                b_off = [0]
+                segs = []
+                for j in built_idx:
+                    s, e = off_list[j], off_list[j + 1]
+                    segs.append((s, e))
+                    b_off.append(b_off[-1] + (e - s))
+                build_rows = (
+                    torch.cat([row_idx[s:e] for s, e in segs])
+                    if len(segs) > 1
+                    else row_idx[segs[0][0]: segs[0][1]]
+                )
+                col0_b = torch.tensor(
+                    [node_ids[j][0] * C for j in built_idx], dtype=torch.int32
+                )
+                bh = ops.hist_build_forest(
+                    bins, gh, build_rows,
+                    torch.tensor(b_off, dtype=torch.int64), col0_b, B, C,
+                    gh_max,
+                )
+                if comm is not None:
+                    comm.all_reduce_(bh)
+                for k, j in enumerate(built_idx):
+                    child_hists[j] = bh[k]
+            for j in range(n_active):
+                if not built_mask[j]:
+                    sib = j - 1 if j % 2 == 1 else j + 1
+                    child_hists[j] = hists[parent_of[j]] - child_hists[sib]
+            hists = child_hists
+
+        # ----- split decision (identical on every rank) ------------------
+        gain, feat, b, left_stats = ops.split_search(
+            hists, params.lam, params.min_child_weight,
+            params.min_instances_per_node, params.min_info_gain, d_dims=1,
+        )
+        pr_async = None
+        if bins.is_cuda:
+            pr_async = ops.partition_rows_async(bins, row_idx, offsets, feat, b)
+
+        gain_cpu = gain.cpu()
+        feat_cpu = feat.cpu()
+        b_cpu = b.cpu()
+        left_stats = left_stats.cpu()
+
+        do_split = torch.isfinite(gain_cpu)
+        ns_idx = (~do_split).nonzero(as_tuple=True)[0]
+        if ns_idx.numel():
+            _finalize(ns_idx.tolist(), totals[ns_idx])
+            _capture_leaves(ns_idx.tolist(), offsets, row_idx)
+        if not bool(do_split.any()):
+            break
+
+        split_feat = torch.where(do_split, feat_cpu, torch.full_like(feat_cpu, -1))
+        child_ids: List[Optional[Tuple[int, int]]] = []
+        for i in range(n_active):
+            if bool(do_split[i]):
+                t, nid = node_ids[i]
+                f = int(feat_cpu[i])
+                tb = int(b_cpu[i])
+                feats[t][nid] = f
+                fi[t, f] += float(gain_cpu[i])
+                thrs[t][nid] = float(edges_cpu[f, tb])
+                cid = alloc_nodes(t, 2)
+                lefts[t][nid] = cid
+                child_ids.append((cid, cid + 1))
+            else:
+                child_ids.append(None)
+
+        if pr_async is not None:
+            new_rows, new_offs, _ = ops.partition_rows_finish(*pr_async)
+        else:
+            new_rows, new_offs, _ = ops.partition_rows(
+                bins, row_idx, offsets,
+                split_feat.to(torch.int32), b_cpu.to(torch.int32),
+            )
+        offs_list = new_offs.tolist()
+
+        next_nodes: List[Tuple[int, int]] = []
+        next_off: List[int] = [0]
+        next_tot_rows: List[torch.Tensor] = []
+        nb_mask: List[bool] = []
+        nb_parent: List[int] = []
+        keep_segs: List[Tuple[int, int]] = []
+        surviving = [i for i in range(n_active) if child_ids[i] is not None]
+        hist_keep = torch.tensor(surviving, dtype=torch.long, device=device)
+        for rank_i, i in enumerate(surviving):
+            t, _ = node_ids[i]
+            lcid, rcid = child_ids[i]
+            l_stats = left_stats[i]
+            r_stats = totals[i] - l_stats
+            l_cnt = float(l_stats[idx_c])
+            r_cnt = float(r_stats[idx_c])
+            ls, le, re = offs_list[2 * i], offs_list[2 * i + 1], offs_list[2 * i + 2]
+            keep_segs.append((ls, re))
+            for (cid, s, e, st, built) in (
+                (lcid, ls, le, l_stats, l_cnt <= r_cnt),
+                (rcid, le, re, r_stats, l_cnt > r_cnt),
+            ):
+                next_nodes.append((t, cid))
+                next_off.append(next_off[-1] + (e - s))
+                next_tot_rows.append(st)
+                nb_mask.append(built)
+                nb_parent.append(rank_i)
+
+        if len(keep_segs) == 1:
+            s, e = keep_segs[0]
+            row_idx = new_rows[s:e]
+        else:
+            row_idx = torch.cat([new_rows[s:e] for s, e in keep_segs])
+        offsets = torch.tensor(next_off, dtype=torch.int64)
+        node_ids = next_nodes
+        totals = torch.stack(next_tot_rows)
+        hists = hists.index_select(0, hist_keep)
+        parent_of = nb_parent
+        built_mask = nb_mask
+
+    # assemble per-tree arrays
+    trees: List[Dict[str, torch.Tensor]] = []
+    for t in range(T):
+        n_nodes = len(feats[t])
+        leaf_value = torch.zeros(n_nodes, 1, dtype=torch.float32)
+        for i, lv in enumerate(leaves[t]):
+            if lv is not None:
+                leaf_value[i, 0] = lv
+        fi_tot = float(fi[t].sum())
+        tree = {
+            "feature": torch.tensor(feats[t], dtype=torch.int32),
+            "threshold": torch.tensor(thrs[t], dtype=torch.float32),
+            "left_child": torch.tensor(lefts[t], dtype=torch.int32),
+            "leaf_value": leaf_value,
+            "feature_importance": (fi[t] / fi_tot if fi_tot > 0 else fi[t]).to(
+                torch.float32
+            ),
+        }
+        trees.append({k: _to_dev_async(v, device) for k, v in tree.items()})
+
+    if capture:
+        tp = torch.zeros(N, T, dtype=torch.float32, device=device)
+        for t in range(T):
+            if not pred_rows[t]:
+                continue
+            rows_cat = torch.cat(pred_rows[t]).long()
+            counts = _to_dev_async(
+                torch.tensor([r.numel() for r in pred_rows[t]]), device
+            )
+            vals = torch.repeat_interleave(
+                _to_dev_async(torch.tensor(pred_vals[t], dtype=torch.float32),
+                              device),
+                counts,
+            )
+            tp[rows_cat, t] = vals
+        train_pred_out.append(tp)
+    return trees
+
+
 def _to_dev_async(t: torch.Tensor, device) -> torch.Tensor:
     """Pinned + non_blocking H2D for small host arrays: a pageable
     .to(device) makes the host WAIT for every queued kernel before the

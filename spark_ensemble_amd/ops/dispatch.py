@@ -141,9 +141,45 @@ def hist_build(bins, gh, row_idx, node_offsets, num_bins, d_dims=-1, max_abs=Non
                 int(d_dims),
                 max_abs.to(torch.float32),
                 bool(identity_rows),
+                _EMPTY_I32,
+                0,
             )
             return out
     return reference.hist_build(bins, gh, row_idx, node_offsets, num_bins)
+
+
+_EMPTY_I32 = torch.empty(0, dtype=torch.int32)
+
+
+def hist_build_forest(bins, gh, row_idx, node_offsets, node_col0, num_bins,
+                      c_per_node, max_abs):
+    """Forest histogram build: gh is [N, T*C] with per-tree channel groups
+    [g_t, h_t(, cnt_t)]; node_col0[n] = owning_tree(n) * C selects the
+    group each node accumulates.  One launch covers every active node of
+    every tree in the level (the MI355X form of the reference's parallel
+    per-class futures, GBMClassifier.scala:377-411).  ``max_abs`` is
+    slot-wise ([C]): max over the corresponding column of every tree."""
+    if bins.is_cuda:
+        m = _require_hip("hist_build")
+        if m is not None:
+            n_nodes = node_offsets.numel() - 1
+            F = bins.shape[1]
+            out = torch.zeros(
+                n_nodes, F, num_bins, c_per_node,
+                dtype=torch.float32, device=bins.device,
+            )
+            m.hist_build(
+                out, bins, gh.contiguous(), row_idx.to(torch.int32),
+                node_offsets.to(torch.int64).cpu(), int(num_bins), 1,
+                max_abs.to(torch.float32),
+                False,
+                node_col0.to(torch.int32).cpu().contiguous(),
+                int(c_per_node),
+            )
+            return out
+    return reference.hist_build_forest(
+        bins, gh, row_idx, node_offsets, node_col0, num_bins, c_per_node
+    )
 
 
 def split_search(hist, lam=1e-6, min_child_weight=0.0, min_instances=1.0, min_info_gain=0.0, d_dims=-1):

@@ -141,6 +141,36 @@ def hist_build(
     return out
 
 
+def hist_build_forest(bins, gh, row_idx, node_offsets, node_col0, num_bins,
+                      c_per_node):
+    """Forest-build reference: gh [N, T*C] interleaves per-tree channel
+    groups; node nd accumulates columns [col0, col0+C) only."""
+    n_nodes = node_offsets.numel() - 1
+    N, F = bins.shape
+    C = int(c_per_node)
+    out = torch.zeros(n_nodes, F, num_bins, C, dtype=torch.float32,
+                      device=bins.device)
+    offs = node_offsets.tolist()
+    col0s = node_col0.tolist()
+    fb = torch.arange(F, device=bins.device, dtype=torch.long) * num_bins
+    for nd in range(n_nodes):
+        s, e = offs[nd], offs[nd + 1]
+        if e <= s:
+            continue
+        c0 = int(col0s[nd])
+        rows = row_idx[s:e].long()
+        b = bins.index_select(0, rows).long()
+        flat = (b + fb.unsqueeze(0)).reshape(-1)
+        vals = (
+            gh[:, c0:c0 + C].index_select(0, rows)
+            .unsqueeze(1)
+            .expand(-1, F, -1)
+            .reshape(-1, C)
+        )
+        out[nd].reshape(F * num_bins, C).index_add_(0, flat, vals)
+    return out
+
+
 # ---------------------------------------------------------------------------
 # Split search (vectorized over nodes x features x bins; works on both
 # devices as plain tensor algebra — small relative to hist_build)

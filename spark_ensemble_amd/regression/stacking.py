@@ -84,43 +84,60 @@ class StackingRegressor(Regressor, _StackingRegressorParams):
         wcol = _check_weight_support(learners, stacker, self.getWeightCol())
         use_w = w if wcol else torch.ones_like(w)
 
+        from ..parallel.streams import parallel_fits
+
+        # ONE frame whose derived-data cache (binned features) is shared
+        # by every fold/final fit below — withColumn propagates the cache,
+        # so the features tensor is quantile-binned once per maxBins for
+        # the WHOLE stacking fit instead of once per member fit
+        shared = TensorFrame(features=x, label=y, weight=use_w)
+
         if self.getOrDefault("inSample"):
-            models = [
-                self.fit_base_learner(
-                    lr,
-                    TensorFrame(features=x, label=y, weight=use_w),
-                    weight_col="weight",
-                )
-                for lr in learners
-            ]
+            models = parallel_fits(
+                [
+                    (lambda lr=lr: self.fit_base_learner(
+                        lr, shared, weight_col="weight"))
+                    for lr in learners
+                ],
+                self.getParallelism(), warm_first=True,
+            )
             meta = _meta_features_reg(models, x)
         else:
             num_folds = self.getOrDefault("numFolds")
+            # identical seeding on every rank: the fold ASSIGNMENT function
+            # is shared, the row draws differ because each rank holds
+            # different rows of the global dataset
             g = torch.Generator().manual_seed(self.getOrDefault("seed"))
-            # rank-offset so folds are iid across shards
             fold = torch.randint(
                 0, num_folds, (n,), generator=g
             ).to(x.device)
             meta = torch.zeros(n, len(learners), dtype=torch.float32, device=x.device)
-            for mi, lr in enumerate(learners):
-                for f in range(num_folds):
+
+            def fold_task(lr, f):
+                def task():
                     wmask = use_w * (fold != f).float()
-                    m = self.fit_base_learner(
-                        lr,
-                        TensorFrame(features=x, label=y, weight=wmask),
+                    return self.fit_base_learner(
+                        lr, shared.withColumn("weight", wmask),
                         weight_col="weight",
                     )
-                    sel = fold == f
-                    meta[sel, mi] = m.predict(x[sel])
-            # final base models refit on everything
-            models = [
-                self.fit_base_learner(
-                    lr,
-                    TensorFrame(features=x, label=y, weight=use_w),
-                    weight_col="weight",
-                )
+                return task
+
+            tasks = [fold_task(lr, f)
+                     for lr in learners for f in range(num_folds)]
+            # final base models refit on everything, same pool
+            tasks += [
+                (lambda lr=lr: self.fit_base_learner(
+                    lr, shared, weight_col="weight"))
                 for lr in learners
             ]
+            fitted = parallel_fits(tasks, self.getParallelism(),
+                                   warm_first=True)
+            for mi, lr in enumerate(learners):
+                for f in range(num_folds):
+                    m = fitted[mi * num_folds + f]
+                    sel = fold == f
+                    meta[sel, mi] = m.predict(x[sel])
+            models = fitted[len(learners) * num_folds:]
 
         stack = self.fit_base_learner(
             stacker,
