@@ -26,13 +26,34 @@ import torch
 import torch.distributed as dist
 
 
+class _NoopWork:
+    def wait(self):
+        return True
+
+
 class Comm:
-    """Thin communicator: no-op at world_size 1, collective otherwise."""
+    """Thin communicator: no-op at world_size 1, collective otherwise.
+
+    Tracing: set ``SEA_COMM_TRACE=<path>`` to append one line per
+    collective — ``op dtype numel`` — so tests can assert that every rank
+    issues the IDENTICAL collective sequence (the property that keeps
+    RCCL deadlock-free) and that payload sizes match the documented
+    per-level histogram math (docs/distributed.md)."""
 
     def __init__(self, rank: int = 0, world_size: int = 1, device=None):
         self.rank = rank
         self.world_size = world_size
         self.device = device
+        base = os.environ.get("SEA_COMM_TRACE")
+        self._trace_path = f"{base}.r{rank}" if base else None
+
+    def _trace(self, op: str, tensor: torch.Tensor):
+        if self._trace_path:
+            with open(self._trace_path, "a") as f:
+                f.write(
+                    f"{op} {str(tensor.dtype).replace('torch.', '')} "
+                    f"{tensor.numel()}\n"
+                )
 
     # -- factory ----------------------------------------------------------
     @property
@@ -40,17 +61,26 @@ class Comm:
         return self.world_size > 1
 
     # -- collectives ------------------------------------------------------
+    _OPS = {"sum": dist.ReduceOp.SUM, "max": dist.ReduceOp.MAX,
+            "min": dist.ReduceOp.MIN} if dist.is_available() else {}
+
     def all_reduce_(self, tensor: torch.Tensor, op: str = "sum") -> torch.Tensor:
         """In-place all-reduce. ``op`` in {sum, max, min}."""
         if not self.is_distributed:
             return tensor
-        ops = {
-            "sum": dist.ReduceOp.SUM,
-            "max": dist.ReduceOp.MAX,
-            "min": dist.ReduceOp.MIN,
-        }
-        dist.all_reduce(tensor, op=ops[op])
+        self._trace(f"all_reduce_{op}", tensor)
+        dist.all_reduce(tensor, op=self._OPS[op])
         return tensor
+
+    def all_reduce_async(self, tensor: torch.Tensor, op: str = "sum"):
+        """Enqueue an all-reduce and return a work handle; ``.wait()``
+        orders the CURRENT stream after the collective (host does not
+        block on NCCL/RCCL).  Lets callers pipeline chunked histogram
+        reduces against split-search compute (tree_grower)."""
+        if not self.is_distributed:
+            return _NoopWork()
+        self._trace(f"all_reduce_async_{op}", tensor)
+        return dist.all_reduce(tensor, op=self._OPS[op], async_op=True)
 
     def all_reduce_scalar(self, value: float, op: str = "sum") -> float:
         if not self.is_distributed:
@@ -70,6 +100,7 @@ class Comm:
     def broadcast_(self, tensor: torch.Tensor, src: int = 0) -> torch.Tensor:
         if not self.is_distributed:
             return tensor
+        self._trace("broadcast", tensor)
         dist.broadcast(tensor, src=src)
         return tensor
 
