@@ -444,27 +444,59 @@ def _binned_edges(pack, F):
     return edges
 
 
+def _run_fp2(m, x_in, pack, binned):
+    out = torch.zeros(x_in.shape[0], pack["D"], dtype=torch.float32,
+                      device=x_in.device)
+    if binned:
+        edges = _binned_edges(pack, x_in.shape[1])
+        xb = torch.empty(x_in.shape, dtype=torch.uint8, device=x_in.device)
+        m.bin_features(xb, x_in, edges)
+        m.forest_predict2(out, xb, pack["node64b"], pack["leaves"],
+                          pack["offsets32"], pack["w"], pack["groups_t"],
+                          pack["D"], pack["max_nodes"])
+    else:
+        m.forest_predict2(out, x_in, pack["node64"], pack["leaves"],
+                          pack["offsets32"], pack["w"], pack["groups_t"],
+                          pack["D"], pack["max_nodes"])
+    return out
+
+
 def _forest_predict_packed(m, x, pack):
     D = pack["D"]
-    out = torch.zeros(x.shape[0], D, dtype=torch.float32, device=x.device)
     if pack["v2"] and x.shape[1] < 32768 and hasattr(m, "forest_predict2"):
         xc = x.contiguous()
-        if (pack.get("binned_ok") and pack["bin_fmax"] <= x.shape[1]
-                and os.environ.get("SEA_SERVE_RAW") != "1"):
-            edges = _binned_edges(pack, x.shape[1])
-            xb = torch.empty(x.shape, dtype=torch.uint8, device=x.device)
-            m.bin_features(xb, xc, edges)
-            m.forest_predict2(out, xb, pack["node64b"], pack["leaves"],
-                              pack["offsets32"], pack["w"], pack["groups_t"],
-                              D, pack["max_nodes"])
-        else:
-            m.forest_predict2(out, xc, pack["node64"], pack["leaves"],
-                              pack["offsets32"], pack["w"], pack["groups_t"],
-                              D, pack["max_nodes"])
-    else:
-        m.forest_predict(out, x.contiguous(), pack["feats"], pack["thrs"],
-                         pack["lefts"], pack["leaves"], pack["offsets32"],
-                         pack["w"], D)
+        binned_avail = (
+            pack.get("binned_ok") and pack["bin_fmax"] <= x.shape[1]
+            and os.environ.get("SEA_SERVE_RAW") != "1"
+        )
+        if not binned_avail:
+            return _run_fp2(m, xc, pack, False)
+        # which walk mode wins depends on the forest's x-access pattern
+        # (measured r02: binned 1.8x faster for a 100-tree GBM, raw 1.7x
+        # faster for 50 subspace-bagged trees) — both are EXACT, so on
+        # the first large batch time both once and remember the winner
+        mode = pack.get("mode")
+        if mode is None:
+            if x.shape[0] < (1 << 20):
+                return _run_fp2(m, xc, pack, True)  # small batch: either way
+            import time
+
+            probe = xc[: 1 << 19]
+            times = {}
+            for b in (True, False):
+                _run_fp2(m, probe, pack, b)  # warm
+                torch.cuda.synchronize()
+                t0 = time.perf_counter()
+                _run_fp2(m, probe, pack, b)
+                torch.cuda.synchronize()
+                times[b] = time.perf_counter() - t0
+            mode = "binned" if times[True] <= times[False] else "raw"
+            pack["mode"] = mode
+        return _run_fp2(m, xc, pack, mode == "binned")
+    out = torch.zeros(x.shape[0], D, dtype=torch.float32, device=x.device)
+    m.forest_predict(out, x.contiguous(), pack["feats"], pack["thrs"],
+                     pack["lefts"], pack["leaves"], pack["offsets32"],
+                     pack["w"], D)
     return out
 
 
