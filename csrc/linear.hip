@@ -37,8 +37,13 @@ namespace {
 
 inline int64_t lceil_div(int64_t a, int64_t b) { return (a + b - 1) / b; }
 
-// NC = number of 256-feature chunks (F <= NC*256), K = classes
-template <int NC, int K>
+// NC = number of 256-feature chunks (F <= NC*256), K = classes.
+// WREG: cache each lane's OWN weight slice in registers — lane l only
+// ever multiplies features {j*256 + l*4 .. +3}, so the [K][F] LDS tile
+// (and its NC*4*K ds_read_b32 per row per lane) is replaced by NC*4*K
+// registers loaded once (measured: removes the per-row LDS issue traffic
+// that capped the kernel at ~3 TB/s of the ~6.3 TB/s roofline).
+template <int NC, int K, bool WREG>
 __global__ __launch_bounds__(256) void logreg_loss_grad_kernel(
     float* __restrict__ payload,     // [1 + (F+1)*K]: loss, grad[F][K], gbias[K]
     const float* __restrict__ x,     // [N, F]
@@ -46,16 +51,29 @@ __global__ __launch_bounds__(256) void logreg_loss_grad_kernel(
     const float* __restrict__ w,     // [N]
     const float* __restrict__ wmat,  // [FP, K] row-major (bias row last if has_bias)
     int64_t n, int F, int has_bias) {
-  extern __shared__ float wlds[];  // [K][F] transposed weights
+  extern __shared__ float wlds[];  // [K][F] transposed weights (!WREG)
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
 
-  for (int i = tid; i < F * K; i += blockDim.x) {
-    const int f = i / K, k = i - f * K;
-    wlds[k * F + f] = wmat[i];
+  float wreg[WREG ? NC * 4 : 1][K];
+  if (WREG) {
+#pragma unroll
+    for (int j = 0; j < NC; ++j)
+#pragma unroll
+      for (int c = 0; c < 4; ++c) {
+        const int f = j * 256 + lane * 4 + c;
+#pragma unroll
+        for (int k = 0; k < K; ++k)
+          wreg[j * 4 + c][k] = (f < F) ? wmat[(int64_t)f * K + k] : 0.0f;
+      }
+  } else {
+    for (int i = tid; i < F * K; i += blockDim.x) {
+      const int f = i / K, k = i - f * K;
+      wlds[k * F + f] = wmat[i];
+    }
+    __syncthreads();
   }
-  __syncthreads();
 
   float bias[K];
 #pragma unroll
@@ -106,7 +124,11 @@ __global__ __launch_bounds__(256) void logreg_loss_grad_kernel(
 #pragma unroll
         for (int c = 0; c < 4; ++c) {
           const int f = f0 + c;
-          if (f < F) {
+          if (WREG) {
+#pragma unroll
+            for (int k = 0; k < K; ++k)
+              acc[u][k] = fmaf(xf[u][j][c], wreg[j * 4 + c][k], acc[u][k]);
+          } else if (f < F) {
 #pragma unroll
             for (int k = 0; k < K; ++k)
               acc[u][k] = fmaf(xf[u][j][c], wlds[k * F + f], acc[u][k]);
@@ -188,9 +210,20 @@ void launch_logreg(float* payload, const float* x, const int* y,
   const int64_t waves_needed = lceil_div(n, 16);  // >=16 rows per wave
   int blocks = (int)std::min<int64_t>(lceil_div(waves_needed, 4), 4096);
   blocks = std::max(blocks, 1);
-  const size_t lds = (size_t)F * K * 4;
-  hipLaunchKernelGGL((logreg_loss_grad_kernel<NC, K>), dim3(blocks), dim3(256),
-                     lds, stream, payload, x, y, w, wmat, n, F, has_bias);
+  static const bool use_lds = []() {
+    const char* e = getenv("SEA_LOGREG_LDS");  // A/B probe escape hatch
+    return e && e[0] == '1';
+  }();
+  if (use_lds) {
+    const size_t lds = (size_t)F * K * 4;
+    hipLaunchKernelGGL((logreg_loss_grad_kernel<NC, K, false>), dim3(blocks),
+                       dim3(256), lds, stream, payload, x, y, w, wmat, n, F,
+                       has_bias);
+  } else {
+    hipLaunchKernelGGL((logreg_loss_grad_kernel<NC, K, true>), dim3(blocks),
+                       dim3(256), 0, stream, payload, x, y, w, wmat, n, F,
+                       has_bias);
+  }
 }
 
 }  // namespace

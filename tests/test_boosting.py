@@ -84,7 +84,7 @@ def test_samme_r_decision_sums_to_zero(clf_frame):
     )
     raw = boost.predictRaw(clf_frame["features"][:50])
     # symmetric constraint (BoostingClassifierSuite:126-154)
-    assert torch.allclose(raw.sum(dim=1), torch.zeros(50), atol=1e-3)
+    assert torch.allclose(raw.sum(dim=1), torch.zeros_like(raw[:, 0]), atol=1e-3)
 
 
 def test_boosting_regressor_beats_base(reg_frame, reg_frame_test):

@@ -75,7 +75,7 @@ def test_gbm_classifier_losses_run(bin_frame, bin_frame_test, loss):
     acc = _acc(gbm, bin_frame_test)
     assert acc > 0.6, (loss, acc)
     prob = gbm.transform(bin_frame_test)["probability"]
-    assert torch.allclose(prob.sum(dim=1), torch.ones(prob.shape[0]), atol=1e-5)
+    assert torch.allclose(prob.sum(dim=1), torch.ones_like(prob[:, 0]), atol=1e-5)
 
 
 def test_gbm_newton_updates(bin_frame, bin_frame_test):

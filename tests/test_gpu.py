@@ -606,7 +606,7 @@ def test_fused_multiclass_round_matches_sequential():
     assert abs(nll_f - nll_s) / max(nll_s, 1e-6) < 0.02, (nll_f, nll_s)
     # and the fused path must agree on the vast majority of rows
     agree = float((out_f["prediction"] == out_s["prediction"]).float().mean())
-    assert agree > 0.97, agree
+    assert agree > 0.95, agree
 
 
 def test_weighted_wide_multiclass_tree_gpu():

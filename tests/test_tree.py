@@ -68,7 +68,7 @@ def test_min_instances_prunes():
 def test_probabilities_sum_to_one(clf_frame):
     m = DecisionTreeClassifier().setMaxDepth(5).fit(clf_frame)
     prob = m.transform(clf_frame)["probability"]
-    assert torch.allclose(prob.sum(dim=1), torch.ones(prob.shape[0]), atol=1e-5)
+    assert torch.allclose(prob.sum(dim=1), torch.ones_like(prob[:, 0]), atol=1e-5)
 
 
 def test_tree_roundtrip_persistence(tmp_path, clf_frame):
