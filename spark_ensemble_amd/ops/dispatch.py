@@ -102,30 +102,38 @@ def hist_build(bins, gh, row_idx, node_offsets, num_bins, d_dims=-1, max_abs=Non
             if max_abs is None:
                 max_abs = gh.abs().amax(dim=0).cpu()
             if C > 8:
-                # wide one-hot targets (K-class gini trees, K > 7): chunk
-                # signed channels into kernel-sized groups, each carrying
-                # the non-negative tail (hess/count) for bookkeeping.
-                # Chunk size is 8 - len(tail) so every recursive call has at
-                # most 8 channels and takes the direct kernel path (a fixed
-                # chunk of 7 would re-enter this branch forever when the
-                # tail has 2 channels, e.g. weighted multiclass).
+                # wide one-hot targets (K-class gini trees, K > 7): the
+                # kernel caps at 8 channels, so re-layout gh ONCE into
+                # contiguous per-chunk groups [g_s..g_{s+w-1}, tail...]
+                # (tail = hess/count duplicated per chunk, <= 2 cols) and
+                # launch one kernel per chunk with a column offset — no
+                # per-chunk gh copies (r01 chunking index_select'ed the
+                # whole gh every chunk).  Chunk width is 8 - len(tail) so
+                # every launch is within the channel cap.
                 D = d_dims if d_dims > 0 else C - 1
                 tail = list(range(D, C))
                 step = 8 - len(tail)
                 assert 1 <= step <= 7, f"hist_build tail too wide: {len(tail)}"
-                parts = []
+                spans = []  # (col0_in_wide, w)
+                cols: list = []
                 for s in range(0, D, step):
                     w = min(step, D - s)
-                    cols = list(range(s, s + w)) + tail
-                    idx = torch.tensor(cols, device=gh.device)
-                    part = hist_build(
-                        bins, gh.index_select(1, idx).contiguous(), row_idx,
-                        node_offsets, num_bins, w,
-                        max_abs[torch.tensor(cols)], identity_rows,
+                    spans.append((len(cols), w))
+                    cols.extend(range(s, s + w))
+                    cols.extend(tail)
+                idx = torch.tensor(cols, device=gh.device)
+                gh_wide = gh.index_select(1, idx).contiguous()  # once
+                ma_wide = max_abs[torch.tensor(cols)]
+                parts = []
+                for ci, (c0, w) in enumerate(spans):
+                    cc = w + len(tail)
+                    part = hist_build_forest(
+                        bins, gh_wide, row_idx, node_offsets,
+                        torch.full((n_nodes,), c0, dtype=torch.int32),
+                        num_bins, cc, ma_wide[c0:c0 + cc], d_dims=w,
                     )
-                    assert part.shape[-1] <= 8  # recursion must have bottomed out
                     parts.append(part[..., :w])
-                    if s + step >= D:
+                    if ci == len(spans) - 1:
                         parts.append(part[..., w:])  # tail once
                 return torch.cat(parts, dim=-1)
             out = torch.zeros(
@@ -152,13 +160,15 @@ _EMPTY_I32 = torch.empty(0, dtype=torch.int32)
 
 
 def hist_build_forest(bins, gh, row_idx, node_offsets, node_col0, num_bins,
-                      c_per_node, max_abs):
-    """Forest histogram build: gh is [N, T*C] with per-tree channel groups
-    [g_t, h_t(, cnt_t)]; node_col0[n] = owning_tree(n) * C selects the
-    group each node accumulates.  One launch covers every active node of
-    every tree in the level (the MI355X form of the reference's parallel
+                      c_per_node, max_abs, d_dims=1):
+    """Forest histogram build: gh is [N, CH] with contiguous per-node
+    channel groups; node_col0[n] selects the base column node n
+    accumulates (tree fusion: owning_tree * C; wide-gini chunking: the
+    chunk's column offset).  One launch covers every active node of every
+    tree in the level (the MI355X form of the reference's parallel
     per-class futures, GBMClassifier.scala:377-411).  ``max_abs`` is
-    slot-wise ([C]): max over the corresponding column of every tree."""
+    slot-wise ([c_per_node]): max over the corresponding column of every
+    group."""
     if bins.is_cuda:
         m = _require_hip("hist_build")
         if m is not None:
@@ -170,7 +180,8 @@ def hist_build_forest(bins, gh, row_idx, node_offsets, node_col0, num_bins,
             )
             m.hist_build(
                 out, bins, gh.contiguous(), row_idx.to(torch.int32),
-                node_offsets.to(torch.int64).cpu(), int(num_bins), 1,
+                node_offsets.to(torch.int64).cpu(), int(num_bins),
+                int(d_dims),
                 max_abs.to(torch.float32),
                 False,
                 node_col0.to(torch.int32).cpu().contiguous(),
