@@ -210,9 +210,13 @@ void launch_logreg(float* payload, const float* x, const int* y,
   const int64_t waves_needed = lceil_div(n, 16);  // >=16 rows per wave
   int blocks = (int)std::min<int64_t>(lceil_div(waves_needed, 4), 4096);
   blocks = std::max(blocks, 1);
+  // measured (gpurun_out/r02_logreg.json): W-in-LDS 3.1 TB/s vs
+  // W-in-registers 2.86 TB/s at 5M x 1024 K=2 — the LDS reads hide
+  // behind the x-load latency and the extra registers cost occupancy,
+  // so LDS stays the default; SEA_LOGREG_WREG=1 flips for probing
   static const bool use_lds = []() {
-    const char* e = getenv("SEA_LOGREG_LDS");  // A/B probe escape hatch
-    return e && e[0] == '1';
+    const char* e = getenv("SEA_LOGREG_WREG");
+    return !(e && e[0] == '1');
   }();
   if (use_lds) {
     const size_t lds = (size_t)F * K * 4;

@@ -138,9 +138,21 @@ class BaggingClassificationModel(
         )
 
     def predictRaw(self, features: torch.Tensor) -> torch.Tensor:
+        from ..ensemble.utils import packed_forest_vote
+
         x = features.float()
         k = self._num_classes
         soft = self.getVotingStrategy() == "soft"
+        # fast path: every member a built-in classification tree -> ONE
+        # packed forest kernel with per-node leaf transforms (normalized
+        # probs for soft, one-hot argmax for hard) — exact
+        packed = packed_forest_vote(
+            x, self._models, self._subspaces, x.shape[1], soft,
+            cache=self.__dict__.setdefault(
+                "_pack_cache_soft" if soft else "_pack_cache_hard", {}),
+        )
+        if packed is not None:
+            return packed
         acc = torch.zeros(x.shape[0], k, dtype=torch.float32, device=x.device)
         for sub, m in zip(self._subspaces, self._models):
             xs = slice_features(x, sub)

@@ -103,6 +103,56 @@ def packed_forest_margin(x, models, weights, subspaces, num_features,
     return _ops.forest_predict(x, trees, w, cache=cache).squeeze(1)
 
 
+def packed_forest_vote(x, models, subspaces, num_features, soft,
+                       cache=None):
+    """Bagging-classifier voting as ONE packed forest_predict call.
+
+    Exactness by leaf transform (cached per model): soft voting sums
+    per-row probability vectors — equal to summing leaf vectors
+    NORMALIZED per node; hard voting sums one-hot argmax votes — equal to
+    summing per-node ONE-HOT(argmax(leaf)) vectors.  Returns [N, K] vote
+    sums or None when a member is not a built-in classification tree."""
+    from ..models.tree import DecisionTreeClassificationModel
+    from ..ops import dispatch as _ops
+
+    if not models:
+        return None
+    trees = []
+    attr = "_leaf_soft" if soft else "_leaf_hard"
+    for m, sub in zip(models, subspaces):
+        if not isinstance(m, DecisionTreeClassificationModel):
+            return None
+        t = m._tree
+        lv = getattr(m, attr, None)
+        if lv is None:
+            leaf = t["leaf_value"]
+            if soft:
+                lv = leaf / leaf.sum(dim=1, keepdim=True).clamp_min(1e-12)
+            else:
+                k = leaf.shape[1]
+                lv = torch.zeros_like(leaf)
+                lv.scatter_(1, leaf.argmax(dim=1, keepdim=True), 1.0)
+            setattr(m, attr, lv)
+        identity = sub is None or (
+            sub.numel() == num_features
+            and bool((sub.cpu() == torch.arange(num_features)).all())
+        )
+        feat = t["feature"]
+        if not identity:
+            remapped = getattr(m, "_tree_orig_feats", None)
+            if remapped is None:
+                fl = feat.long()
+                sub_dev = sub.to(fl.device)
+                remapped = torch.where(
+                    fl >= 0, sub_dev[fl.clamp_min(0)], fl
+                ).to(torch.int32)
+                m._tree_orig_feats = remapped
+            feat = remapped
+        trees.append(dict(t, feature=feat, leaf_value=lv))
+    w = torch.ones(len(trees), dtype=torch.float32)
+    return _ops.forest_predict(x, trees, w, cache=cache)
+
+
 def ensemble_feature_importances(models, weights, subspaces, num_features):
     """Weighted, subspace-mapped aggregate of member featureImportances
     (normalized to sum 1; zeros when no member exposes importances)."""

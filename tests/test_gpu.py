@@ -609,6 +609,40 @@ def test_fused_multiclass_round_matches_sequential():
     assert agree > 0.95, agree
 
 
+@pytest.mark.parametrize("voting", ["soft", "hard"])
+def test_bagging_vote_packed_matches_loop(voting):
+    """Classification bagging predictRaw: the packed forest path (leaf
+    transforms: normalized probs / one-hot argmax) must exactly match the
+    per-member loop."""
+    import spark_ensemble_amd as sea
+    from spark_ensemble_amd.utils.io import synthetic_classification
+
+    df = synthetic_classification(60000, 24, k=3, seed=29, device=DEV)
+    bag = (
+        sea.BaggingClassifier()
+        .setNumBaseLearners(6)
+        .setSubspaceRatio(0.6)
+        .setSubsampleRatio(0.8)
+        .setVotingStrategy(voting)
+        .setSeed(4)
+        .fit(df)
+    )
+    x = df["features"]
+    fast = bag.predictRaw(x)
+
+    # force the loop fallback by hiding the packed path
+    from spark_ensemble_amd.ensemble import utils as eutils
+
+    orig = eutils.packed_forest_vote
+    eutils.packed_forest_vote = lambda *a, **k: None
+    try:
+        slow = bag.predictRaw(x)
+    finally:
+        eutils.packed_forest_vote = orig
+    assert torch.allclose(fast, slow, rtol=1e-5, atol=1e-5), \
+        float((fast - slow).abs().max())
+
+
 def test_weighted_wide_multiclass_tree_gpu():
     """End-to-end: a weighted 8-class gini tree (BoostingClassifier's
     reweighted rounds hit exactly this shape)."""

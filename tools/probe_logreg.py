@@ -13,7 +13,7 @@ import torch
 
 def run_arm(lds: bool):
     env = dict(os.environ)
-    env["SEA_LOGREG_LDS"] = "1" if lds else "0"
+    env["SEA_LOGREG_WREG"] = "0" if lds else "1"
     r = subprocess.run([sys.executable, __file__, "arm"], env=env,
                        capture_output=True, text=True, cwd="/root/repo")
     assert r.returncode == 0, r.stderr[-1500:]
@@ -47,7 +47,7 @@ def arm():
     dt = (time.time() - t0) / reps
     bytes_x = n * F * 4
     print(json.dumps({
-        "mode": "lds" if os.environ.get("SEA_LOGREG_LDS") == "1" else "wreg",
+        "mode": "wreg" if os.environ.get("SEA_LOGREG_WREG") == "1" else "lds",
         "ms": round(dt * 1000, 2),
         "tb_per_s": round(bytes_x / dt / 1e12, 2),
         "loss": float(p[0]),
