@@ -23,6 +23,11 @@ import torch
 from ..parallel import Comm
 from .losses import GBMLoss
 
+# instrumentation: number of fused evaluations the last 1-D search used
+# (each is a kernel + all-reduce + host sync — the per-round cost scales
+# with it); read by the GBM fit loops for per-round logging
+LAST_EVALS = 0
+
 
 def _eval(loss, label, pred, direction, weight, coeff, want_hess=False):
     """Returns (weighted loss sum, per-dim gradient sums[, hessian sum]) as
@@ -82,6 +87,7 @@ def optimize_weight_1d(
     evaluation is a kernel launch + an RCCL all-reduce + a sync).  Brent
     remains for the non-smooth losses (absolute, huber, quantile) and as
     the fallback when Newton fails to bracket."""
+    global LAST_EVALS
     if getattr(loss, "smooth", False) and loss.has_hessian:
         a = _newton_1d(loss, label, pred, direction, weight, comm,
                        max_iter, tol, lo, hi)
@@ -112,6 +118,7 @@ def optimize_weight_1d(
         method="bounded",
         options={"maxiter": max_iter, "xatol": max(tol, 1e-8)},
     )
+    LAST_EVALS = len(cache)
     return float(res.x)
 
 
@@ -167,11 +174,14 @@ def _newton_1d(loss, label, pred, direction, weight, comm, max_iter, tol,
         pc = payload.cpu()  # one sync for all three scalars
         return float(pc[0]), float(pc[1]), float(pc[2])
 
+    global LAST_EVALS
     a = 1.0  # natural stage weight
     blo, bhi = lo, hi
     best_a, best_f = None, float("inf")
+    LAST_EVALS = 0
     for _ in range(max(8, min(max_iter, 20))):
         f, g, h = eval_at(a)
+        LAST_EVALS += 1
         if not (np.isfinite(f) and np.isfinite(g) and np.isfinite(h)):
             return None
         if f < best_f:

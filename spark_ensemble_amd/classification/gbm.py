@@ -295,6 +295,8 @@ class GBMClassifier(ProbabilisticClassifier, _GBMClassifierParams):
             wt = torch.tensor(iweights, dtype=torch.float32, device=device)
             predictions = predictions + directions * wt.unsqueeze(0)
 
+            from ..boosting import line_search as _ls
+
             if val is not None:
                 xvs = slice_features(xv, idx)
                 vdir = torch.stack([m.predict(xvs) for m in imodels], dim=1)
@@ -306,9 +308,10 @@ class GBMClassifier(ProbabilisticClassifier, _GBMClassifierParams):
                     best_err = err
                     v = 0
                 instr.log_round(i, weight=float(iweights[0]), val_loss=err,
-                                patience=v)
+                                patience=v, ls_evals=_ls.LAST_EVALS)
             else:
-                instr.log_round(i, weight=float(iweights[0]))
+                instr.log_round(i, weight=float(iweights[0]),
+                                ls_evals=_ls.LAST_EVALS)
             interval = self.getCheckpointInterval()
             if ckpt_dir and interval > 0 and (i + 1) % interval == 0:
                 ckpt.save_round_state(

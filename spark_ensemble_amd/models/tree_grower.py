@@ -498,6 +498,7 @@ def grow_forest(
     hess_is_count: Optional[bool] = None,
     train_pred_out: Optional[list] = None,
     gh_max_in: Optional[torch.Tensor] = None,
+    root_rows: Optional[List[torch.Tensor]] = None,  # per-tree row sets
 ) -> List[Dict[str, torch.Tensor]]:
     """Grow T single-output trees LEVEL-SYNCHRONOUSLY in fused launches.
 
@@ -531,6 +532,7 @@ def grow_forest(
             out.extend(grow_forest(
                 bins, edges, grads[:, sl].contiguous(), h_sl, params, comm,
                 hess_is_count, sub_pred, gh_max_in,
+                root_rows[sl] if root_rows is not None else None,
             ))
             if preds is not None:
                 preds.append(sub_pred[0])
@@ -557,8 +559,16 @@ def grow_forest(
     else:
         gh_max = None
 
-    assert T * N < 2**31, "fused forest row arena exceeds int32"
-    row_idx = torch.arange(N, dtype=torch.int32, device=device).repeat(T)
+    if root_rows is None:
+        assert T * N < 2**31, "fused forest row arena exceeds int32"
+        row_idx = torch.arange(N, dtype=torch.int32, device=device).repeat(T)
+        root_lens = [N] * T
+    else:
+        assert len(root_rows) == T
+        root_rows = [r.to(device=device, dtype=torch.int32) for r in root_rows]
+        row_idx = torch.cat(root_rows) if T > 1 else root_rows[0]
+        root_lens = [int(r.numel()) for r in root_rows]
+        assert sum(root_lens) < 2**31
 
     # per-tree flat node arrays
     feats = [[] for _ in range(T)]
@@ -576,14 +586,25 @@ def grow_forest(
         return start
 
     # root totals: one fused reduction (+ one all-reduce) for all trees
-    g_sum = grads.sum(dim=0)  # [T]
-    if h_shared:
-        h_sum = hess.sum().reshape(1).expand(T)
+    if root_rows is None:
+        g_sum = grads.sum(dim=0)  # [T]
+        if h_shared:
+            h_sum = hess.sum().reshape(1).expand(T)
+        else:
+            h_sum = hess.sum(dim=0)
+        cnt_col = torch.full((T,), float(N), device=device)
     else:
-        h_sum = hess.sum(dim=0)
+        gs, hs = [], []
+        for t in range(T):
+            r = root_rows[t].long()
+            gs.append(grads[r, t].sum())
+            hs.append((hess if h_shared else hess[:, t])[r].sum())
+        g_sum = torch.stack(gs)
+        h_sum = torch.stack(hs)
+        cnt_col = torch.tensor([float(v) for v in root_lens], device=device)
     cols = [g_sum, h_sum]
     if C == 3:
-        cols.append(torch.full((T,), float(N), device=device))
+        cols.append(cnt_col)
     root_tot = torch.stack(cols, dim=1)  # [T, C] device
     if comm is not None:
         comm.all_reduce_(root_tot)
@@ -591,7 +612,10 @@ def grow_forest(
 
     # active level state (across all trees)
     node_ids: List[Tuple[int, int]] = [(t, alloc_nodes(t, 1)) for t in range(T)]
-    offsets = torch.arange(0, (T + 1) * N, N, dtype=torch.int64)
+    off_acc = [0]
+    for v in root_lens:
+        off_acc.append(off_acc[-1] + v)
+    offsets = torch.tensor(off_acc, dtype=torch.int64)
     hists: Optional[torch.Tensor] = None
     parent_of: List[int] = []
     built_mask: List[bool] = []

@@ -301,6 +301,129 @@ class GBMRegressor(Regressor, _GBMRegressorParams):
             model.set(p, self.getOrDefault(p))
         return model
 
+    # ---- fold-vectorized fitting (OOF stacking fast path) ----------------
+    def _can_fit_folds(self) -> bool:
+        """The fold-fused fit covers the default configuration: squared
+        loss, gradient updates, constant init, full bags, identity
+        subspace, built-in tree learner, no validation/checkpointing.
+        Anything else falls back to independent per-fold fits."""
+        from ..models.tree import DecisionTreeRegressor
+
+        learner = self.getOrNone("baseLearner") or self._default_base_learner()
+        return (
+            self.getLoss() == "squared"
+            and self.getOrDefault("initStrategy") == "constant"
+            and self.getOrDefault("updates") == "gradient"
+            and self.getSubspaceRatio() >= 1.0
+            and self.getSubsampleRatio() >= 1.0
+            and not self.getReplacement()
+            and self.getOrNone("validationIndicatorCol") is None
+            and not self.getCheckpointDir()
+            and type(learner) is DecisionTreeRegressor
+            and learner.getOrDefault("minWeightFractionPerNode") == 0.0
+        )
+
+    def _fit_folds(self, dataset: TensorFrame, fold: torch.Tensor,
+                   num_folds: int) -> List["GBMRegressionModel"]:
+        """num_folds leave-one-fold-out GBM fits grown JOINTLY: every
+        boosting round builds ONE fused forest (grow_forest with per-tree
+        root row sets = each fold's training rows) and the squared-loss
+        stage weights come in closed form (a* = sum(w d r)/sum(w d^2) on
+        the fold's rows — the exact minimizer Brent/Newton converge to),
+        so a round costs one forest build + two reductions regardless of
+        the fold count.  The MI355X answer to OOF stacking's many small
+        sequential fits (reference StackingRegressor.scala:141-153 runs
+        whole fits in driver futures)."""
+        from ..models.dummy import DummyRegressionModel
+        from ..models.tree import DecisionTreeRegressionModel
+        from ..models.tree_grower import grow_forest
+
+        comm = get_comm()
+        learner = self.getOrNone("baseLearner") or self._default_base_learner()
+        T = int(num_folds)
+        k_stages = self.getNumBaseLearners()
+        lr_rate = self.getOrDefault("learningRate")
+        optimized = self.getOrDefault("optimizedWeights")
+
+        x, y, w = self._extract_xyw(dataset)
+        n, num_features = x.shape
+        device = x.device
+        binned = BinnedDataset(x, dataset)
+        edges, bins = binned.get(int(learner.getOrDefault("maxBins")))
+
+        fold = fold.to(device)
+        masks = [(fold != f) for f in range(T)]
+        root_rows = [m.nonzero(as_tuple=True)[0].to(torch.int32)
+                     for m in masks]
+        mask_f = torch.stack([m.float() for m in masks], dim=1)  # [N, T]
+        w2 = w.unsqueeze(1) * mask_f
+
+        # constant init: the sequential path's DummyRegressor reproduces
+        # the reference's UNWEIGHTED SQL mean (DummyRegressor.scala:113-129
+        # ignores instance weights — a reference quirk), so every fold
+        # model starts from the same global unweighted label mean
+        red = torch.stack([y.sum(), torch.tensor(float(n), device=device)])
+        comm.all_reduce_(red)
+        init_c = (red[0] / red[1].clamp_min(1e-12)).expand(T)  # [T] device
+
+        margins = init_c.unsqueeze(0).expand(n, T).contiguous()
+        hic = bool((w == 1).all())
+        gp = learner._grow_params(1.0)
+
+        fold_trees: List[List] = [[] for _ in range(T)]
+        alphas_per_round: List[torch.Tensor] = []
+        for i in range(k_stages):
+            resid = y.unsqueeze(1) - margins
+            grads = (resid * w.unsqueeze(1)).contiguous()
+            stats = torch.stack([grads.abs().max(), w.max()])
+            if comm.is_distributed:
+                comm.all_reduce_(stats, "max")
+            sc = stats.cpu()
+            gh_max = torch.tensor(
+                [float(sc[0]), float(sc[1])] + ([] if hic else [1.0])
+            )
+            tp: list = []
+            trees = grow_forest(bins, edges, grads, w, gp, comm,
+                                hess_is_count=hic, train_pred_out=tp,
+                                gh_max_in=gh_max, root_rows=root_rows)
+            d = tp[0]  # [N, T]; zero on held-out rows
+            if optimized:
+                red = torch.cat([
+                    (w2 * d * resid).sum(0), (w2 * d * d).sum(0)
+                ])
+                comm.all_reduce_(red)
+                alphas = (red[:T] / red[T:].clamp_min(1e-30)).clamp(0.0, 100.0)
+            else:
+                alphas = torch.ones(T, device=device)
+            alphas = alphas * lr_rate
+            alphas_per_round.append(alphas)
+            margins = margins + d * alphas.unsqueeze(0)
+            for t, tree in enumerate(trees):
+                m = DecisionTreeRegressionModel()
+                m._set_tree(tree, num_features)
+                m._copy_cols_from(learner)
+                fold_trees[t].append(m)
+
+        stage_w = torch.stack(alphas_per_round).cpu()  # [k, T] one sync
+        init_cpu = init_c.cpu()
+        ident = torch.arange(num_features)
+        out: List[GBMRegressionModel] = []
+        for t in range(T):
+            init = DummyRegressionModel()
+            init._constant = float(init_cpu[t])
+            init._num_features = num_features
+            init.set("strategy", "constant")
+            model = GBMRegressionModel()
+            model._init = init
+            model._models = fold_trees[t]
+            model._weights = [float(v) for v in stage_w[:, t]]
+            model._subspaces = [ident] * k_stages
+            model._num_features = num_features
+            for p in ("featuresCol", "labelCol", "predictionCol"):
+                model.set(p, self.getOrDefault(p))
+            out.append(model)
+        return out
+
     def _save_impl(self, path: str):
         persistence.save_metadata(self, path)
         self._save_learner(path)

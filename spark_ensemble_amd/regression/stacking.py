@@ -113,6 +113,22 @@ class StackingRegressor(Regressor, _StackingRegressorParams):
             ).to(x.device)
             meta = torch.zeros(n, len(learners), dtype=torch.float32, device=x.device)
 
+            # learners with a fold-vectorized fit grow all their fold
+            # models JOINTLY (one fused forest per boosting round —
+            # GBMRegressor._fit_folds); the rest run per-fold fits on the
+            # stream pool
+            fused_ok = bool((use_w > 0).all())
+            fused = {
+                mi for mi, lr in enumerate(learners)
+                if fused_ok and hasattr(lr, "_can_fit_folds")
+                and lr._can_fit_folds()
+            }
+            fold_models: dict = {}
+            for mi in fused:
+                fold_models[mi] = learners[mi]._fit_folds(
+                    shared, fold, num_folds
+                )
+
             def fold_task(lr, f):
                 def task():
                     wmask = use_w * (fold != f).float()
@@ -122,8 +138,9 @@ class StackingRegressor(Regressor, _StackingRegressorParams):
                     )
                 return task
 
-            tasks = [fold_task(lr, f)
-                     for lr in learners for f in range(num_folds)]
+            plan = [(mi, f) for mi, lr in enumerate(learners)
+                    if mi not in fused for f in range(num_folds)]
+            tasks = [fold_task(learners[mi], f) for mi, f in plan]
             # final base models refit on everything, same pool
             tasks += [
                 (lambda lr=lr: self.fit_base_learner(
@@ -132,12 +149,14 @@ class StackingRegressor(Regressor, _StackingRegressorParams):
             ]
             fitted = parallel_fits(tasks, self.getParallelism(),
                                    warm_first=True)
-            for mi, lr in enumerate(learners):
+            for j, (mi, f) in enumerate(plan):
+                fold_models.setdefault(mi, [None] * num_folds)[f] = fitted[j]
+            for mi in range(len(learners)):
                 for f in range(num_folds):
-                    m = fitted[mi * num_folds + f]
+                    m = fold_models[mi][f]
                     sel = fold == f
                     meta[sel, mi] = m.predict(x[sel])
-            models = fitted[len(learners) * num_folds:]
+            models = fitted[len(plan):]
 
         stack = self.fit_base_learner(
             stacker,

@@ -99,6 +99,78 @@ def test_forest_batching_over_max_fused():
     assert torch.allclose(pred_b[0], pred_f[0])
 
 
+def test_forest_root_rows_matches_masked_trees():
+    """Per-tree root row sets (OOF folds): each fused tree must equal a
+    grow_tree fit restricted to the same rows via row_mask."""
+    n, f, b, T = 18000, 10, 32, 3
+    x, edges, bins, g = _data(n, f, b, seed=7)
+    params = GrowParams(max_depth=4, max_bins=b)
+    grads = torch.randn(n, T, generator=g)
+    hess = torch.ones(n)
+    fold = torch.randint(0, T, (n,), generator=g)
+    rows = [(fold != t).nonzero(as_tuple=True)[0].to(torch.int32)
+            for t in range(T)]
+
+    pred_f: list = []
+    forest = grow_forest(bins, edges, grads, hess, params,
+                         train_pred_out=pred_f, root_rows=rows)
+    for t in range(T):
+        single = grow_tree(bins, edges, grads[:, t:t + 1].contiguous(),
+                           hess, params, row_mask=(fold != t))
+        _assert_tree_equal(forest[t], single, t)
+        # captured training predictions are zero on held-out rows
+        held = fold == t
+        assert torch.equal(pred_f[0][held, t],
+                           torch.zeros(int(held.sum())))
+
+
+def test_gbm_fit_folds_matches_sequential_oof():
+    """GBMRegressor._fit_folds (fused fold-vectorized fit) must
+    reproduce num_folds independent weight-masked GBM fits."""
+    import spark_ensemble_amd as sea
+    from spark_ensemble_amd.frame import TensorFrame
+    from spark_ensemble_amd.utils.io import synthetic_regression
+
+    df = synthetic_regression(9000, 12, seed=23)
+    x, y = df["features"], df["label"]
+    T = 3
+    g = torch.Generator().manual_seed(5)
+    fold = torch.randint(0, T, (x.shape[0],), generator=g)
+
+    est = sea.GBMRegressor().setNumBaseLearners(4).setSeed(1)
+    assert est._can_fit_folds()
+    shared = TensorFrame(features=x, label=y, weight=torch.ones_like(y))
+    fused = est._fit_folds(shared, fold, T)
+
+    for t in range(T):
+        wmask = (fold != t).float()
+        seq = (
+            sea.GBMRegressor().setNumBaseLearners(4).setSeed(1)
+            .set("weightCol", "weight")
+            .fit(TensorFrame(features=x, label=y, weight=wmask))
+        )
+        # same init/stage-weights up to float (closed-form alpha == the
+        # Newton minimizer of the same quadratic)
+        assert abs(fused[t]._init._constant - seq._init._constant) < 1e-6
+        for wa, wb in zip(fused[t]._weights, seq._weights):
+            assert abs(wa - wb) < 1e-5, (wa, wb)
+        # early trees identical; late rounds may flip near-tie splits as
+        # ~1e-7 stage-weight noise compounds through the margins
+        for r in range(2):
+            fa, sb = fused[t]._models[r]._tree, seq._models[r]._tree
+            assert torch.equal(fa["feature"], sb["feature"]), (t, r)
+            assert torch.equal(fa["threshold"], sb["threshold"]), (t, r)
+        a = fused[t].predict(x)
+        b = seq.predict(x)
+        # models must be statistically interchangeable
+        held = fold == t
+        mse_a = float(((a - y)[held] ** 2).mean())
+        mse_b = float(((b - y)[held] ** 2).mean())
+        assert abs(mse_a - mse_b) <= 0.02 * max(mse_b, 1e-9), (mse_a, mse_b)
+        close = float((a - b).abs().lt(0.05 * y.std()).float().mean())
+        assert close > 0.95, close
+
+
 def test_hist_build_forest_reference():
     """The per-node column-offset histogram itself."""
     g = torch.Generator().manual_seed(9)

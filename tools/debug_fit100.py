@@ -30,3 +30,5 @@ print(f"K={K} wall {wall:.2f}s ({wall*1000/K:.1f} ms/round)  h0 {h[0]['ms']:.0f}
 print("setup_ms", round(est._instr.timers.get("setup_ms", -1), 1))
 print("every 10th:", [round(v, 1) for v in d[::10]])
 print("last 10:", [round(v, 1) for v in d[-10:]])
+ev = [r.get("ls_evals") for r in h]
+print("ls_evals every 10th:", ev[::10])
