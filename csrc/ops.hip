@@ -990,13 +990,19 @@ __device__ inline float scalar_loss_grad(int loss_id, float param, float y,
       return e;
     }
     case L_BERNOULLI: {
+      // overflow-safe forms: exp(z) -> inf past |z|~88 made the naive
+      // hess 4e/(1+e)^2 = inf/inf = NaN once boosting margins grow,
+      // kicking the stage-weight search from 3-6 Newton evals to ~25
+      // Brent evals per round (measured +4 ms/round after round ~40).
+      // sigmoid(-z) = t/(1+t) with t = exp(-|z|) in (0,1] is finite
+      // everywhere; grad = -2y*sigmoid(-z), hess = 4*sig*(1-sig).
       float z = 2.0f * y * p;
-      float e = __expf(z);
-      *grad = -2.0f * y / (1.0f + e);
-      float ip = 1.0f + e;
-      *hess = 4.0f * e / (ip * ip);
       float az = fabsf(z);
-      return fmaxf(-z, 0.0f) + log1pf(__expf(-az));
+      float t = __expf(-az);
+      float sig_neg = (z >= 0.0f) ? t / (1.0f + t) : 1.0f / (1.0f + t);
+      *grad = -2.0f * y * sig_neg;
+      *hess = 4.0f * sig_neg * (1.0f - sig_neg);
+      return fmaxf(-z, 0.0f) + log1pf(t);
     }
   }
   *grad = 0.0f;

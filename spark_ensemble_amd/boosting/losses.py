@@ -283,8 +283,12 @@ class BernoulliLoss(_ClassificationLoss):
         return -2 * label / (1 + torch.exp(2 * label * pred))
 
     def hessian(self, label, pred):
-        e = torch.exp(2 * pred * label)
-        return 4 * e * label * label / (1 + e) ** 2
+        # overflow-safe: exp(z) -> inf makes 4e/(1+e)^2 NaN for |z|>~88
+        # (large late-boosting margins); 4*sig(z)*sig(-z) is finite
+        # everywhere (matches the HIP kernel's form, csrc/ops.hip)
+        z = 2 * pred * label
+        s = torch.sigmoid(z)
+        return 4 * label * label * s * (1 - s)
 
     def raw2probability(self, raw):
         # reference sign quirk (GBMLoss.scala:311-316)
