@@ -270,13 +270,21 @@ class DecisionTreeRegressionModel(_TreeModelMixin, RegressionModel, _TreeParams)
         return self._predict_values(features.float()).squeeze(1)
 
 
-def fit_tree_forest(learner, edges, bins, labels, weights, comm=None):
+def fit_tree_forest(learner, edges, bins, labels, weights, comm=None,
+                    subspaces=None, root_rows=None):
     """T independent ``DecisionTreeRegressor`` fits as ONE fused forest
     grow (tree_grower.grow_forest): the MI355X form of the reference's
-    per-class / per-learner fit futures.  ``labels`` [N, T]; ``weights``
-    [N] shared or [N, T] per-tree, all strictly positive (callers
-    fall back to sequential fits when zero weights exist — the fused
-    grower has no row mask).  Returns (models, train_pred [N, T]).
+    per-class / per-learner fit futures.  ``labels`` [N, T] (or [N, 1]
+    broadcast); ``weights`` [N] shared or [N, T] per-tree, strictly
+    positive on each tree's rows (``root_rows`` restricts a tree to a row
+    subset — pass the rows with positive bag weight; without it callers
+    must guarantee all-positive weights).  ``subspaces``: per-tree sorted
+    feature-index tensors — trees are grown in the FULL feature space
+    with banned features masked out of the split search (identical
+    splits to a sliced fit) and their split ids remapped to the
+    subspace-local space afterwards, so the returned models are
+    indistinguishable from sliced fits.  Returns (models,
+    train_pred [N, T]) — train_pred is zero outside a tree's rows.
 
     Caller contract: ``learner`` is a plain DecisionTreeRegressor with
     minWeightFractionPerNode == 0 (per-tree total weights would otherwise
@@ -285,8 +293,12 @@ def fit_tree_forest(learner, edges, bins, labels, weights, comm=None):
     from .tree_grower import grow_forest
 
     comm = comm or _gc()
+    F = bins.shape[1]
     w2 = weights if weights.dim() == 2 else weights.unsqueeze(1)
     grads = (labels * w2).contiguous()
+    T = max(grads.shape[1], w2.shape[1])
+    if grads.shape[1] != T:
+        grads = grads.expand(-1, T).contiguous()
     # one fused stats sync: slot-wise quantization maxima + unit check
     stats = torch.stack([
         grads.abs().max(),
@@ -300,15 +312,33 @@ def fit_tree_forest(learner, edges, bins, labels, weights, comm=None):
     gh_max = torch.tensor(
         [float(stats_c[0]), float(stats_c[1])] + ([] if hic else [1.0])
     )
+    fmasks = None
+    if subspaces is not None and any(
+        s.numel() != F for s in subspaces
+    ):
+        fmasks = torch.zeros(T, F, dtype=torch.float32, device=bins.device)
+        for t, sub in enumerate(subspaces):
+            fmasks[t, sub.to(bins.device)] = 1.0
     gp = learner._grow_params(1.0)  # minWeightFraction==0 ⇒ total unused
     tp_out: list = []
     trees = grow_forest(bins, edges, grads, weights, gp, comm,
                         hess_is_count=hic, train_pred_out=tp_out,
-                        gh_max_in=gh_max)
+                        gh_max_in=gh_max, root_rows=root_rows,
+                        feature_masks=fmasks)
     models = []
     for t, tree in enumerate(trees):
+        nf = F
+        if fmasks is not None:
+            sub = subspaces[t]
+            nf = int(sub.numel())
+            feat = tree["feature"].long()
+            sub_dev = sub.to(device=feat.device, dtype=torch.long)
+            local = torch.searchsorted(sub_dev, feat.clamp_min(0))
+            tree = dict(tree, feature=torch.where(
+                feat >= 0, local, feat
+            ).to(torch.int32))
         m = DecisionTreeRegressionModel()
-        m._set_tree(tree, bins.shape[1])
+        m._set_tree(tree, nf)
         m._train_pred = tp_out[0][:, t]
         m._copy_cols_from(learner)
         models.append(m)

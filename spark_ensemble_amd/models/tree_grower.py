@@ -400,9 +400,16 @@ def _finish_level_split(
     device,
     comm: Optional[Comm],
     split_args: dict,
+    split_mask: Optional[torch.Tensor] = None,  # [n_active, F] 1/0
 ):
     """All-reduce the built histograms, assemble the level's full hist
     tensor (scatter + sibling subtraction) and run split_search.
+
+    ``split_mask`` implements per-node FEATURE SUBSPACES without slicing
+    the binned matrix: the split-search input is multiplied by the mask,
+    so a banned feature's histogram is all-zero and every candidate split
+    on it fails the min_instances>=1 child check — it can never win.  The
+    UNMASKED histograms are kept for sibling subtraction.
 
     Distributed: the reduce is FEATURE-CHUNKED and pipelined — chunk c's
     all-reduce overlaps chunk c-1's assembly + split-search compute
@@ -455,7 +462,10 @@ def _finish_level_split(
                 assemble_chunk(hists, built, 0, F)
             else:
                 assemble_chunk(hists, built, 0, F)  # subtraction only
-        g, ft, b, ls = ops_mod.split_search(hists, **split_args)
+        sh = hists
+        if split_mask is not None:
+            sh = hists * split_mask.unsqueeze(-1).unsqueeze(-1)
+        g, ft, b, ls = ops_mod.split_search(sh, **split_args)
         return hists, g, ft, b, ls
 
     n_chunks = max(2, min(4, F // 16))
@@ -472,7 +482,10 @@ def _finish_level_split(
     for f0, f1, ch, h in inflight:
         h.wait()
         assemble_chunk(hists, ch, f0, f1)
-        part = hists[:, f0:f1].contiguous()
+        part = hists[:, f0:f1]
+        if split_mask is not None:
+            part = part * split_mask[:, f0:f1].unsqueeze(-1).unsqueeze(-1)
+        part = part.contiguous()
         g, ft, b, ls = ops_mod.split_search(part, **split_args)
         ft = torch.where(ft >= 0, ft + f0, ft)
         results.append((g, ft, b, ls))
@@ -499,6 +512,7 @@ def grow_forest(
     train_pred_out: Optional[list] = None,
     gh_max_in: Optional[torch.Tensor] = None,
     root_rows: Optional[List[torch.Tensor]] = None,  # per-tree row sets
+    feature_masks: Optional[torch.Tensor] = None,  # [T, F] 1/0 subspaces
 ) -> List[Dict[str, torch.Tensor]]:
     """Grow T single-output trees LEVEL-SYNCHRONOUSLY in fused launches.
 
@@ -533,6 +547,7 @@ def grow_forest(
                 bins, edges, grads[:, sl].contiguous(), h_sl, params, comm,
                 hess_is_count, sub_pred, gh_max_in,
                 root_rows[sl] if root_rows is not None else None,
+                feature_masks[sl] if feature_masks is not None else None,
             ))
             if preds is not None:
                 preds.append(sub_pred[0])
@@ -667,6 +682,11 @@ def grow_forest(
             min_info_gain=params.min_info_gain,
             d_dims=1,
         )
+        lvl_mask = None
+        if feature_masks is not None:
+            tid = torch.tensor([t for (t, _) in node_ids],
+                               dtype=torch.long, device=device)
+            lvl_mask = feature_masks.index_select(0, tid)
         # ----- fused histograms for this level (+ pipelined reduce) ------
         if hists is None:
             new_h = ops.hist_build_forest(
@@ -674,7 +694,7 @@ def grow_forest(
             )
             hists, gain, feat, b, left_stats = _finish_level_split(
                 new_h, None, None, None, None, n_active, (F, B, C), device,
-                comm, split_args,
+                comm, split_args, split_mask=lvl_mask,
             )
         else:
             built_idx = [j for j in range(n_active) if built_mask[j]]
@@ -702,7 +722,7 @@ def grow_forest(
                 )
             hists, gain, feat, b, left_stats = _finish_level_split(
                 bh, built_idx, hists, parent_of, built_mask, n_active,
-                (F, B, C), device, comm, split_args,
+                (F, B, C), device, comm, split_args, split_mask=lvl_mask,
             )
         pr_async = None
         if bins.is_cuda:

@@ -69,28 +69,59 @@ class BaggingRegressor(Regressor, _BaggingRegressorParams):
         subspaces = [
             subspace(self.getSubspaceRatio(), num_features, seed + i) for i in range(k)
         ]
-        if learner is not None and learner.hasParam("maxBins"):
-            binned.get(int(learner.getOrDefault("maxBins")))  # pre-warm once
 
-        def fit_one(i):
-            def task():
-                bag_w = self.sample_weights(
-                    self.getReplacement(),
-                    self.getSubsampleRatio(),
-                    n,
-                    seed + i,
-                    x.device,
-                    w,
-                    comm.rank,
+        from ..models.tree import DecisionTreeRegressor, fit_tree_forest
+
+        if (
+            type(learner) is DecisionTreeRegressor
+            and learner.getOrDefault("minWeightFractionPerNode") == 0.0
+        ):
+            # fused path: all k bagged trees grow level-synchronously in
+            # shared launches (one histogram build + one all-reduce per
+            # level for the whole ensemble); subspaces become split-search
+            # feature masks, bags become per-tree root row sets — exact
+            # parity with k sliced weight-masked fits
+            # (tree_grower.grow_forest; reference futures analog
+            # BaggingRegressor.scala:145-166)
+            f_edges, f_bins = binned.get(int(learner.getOrDefault("maxBins")))
+            bag_w = torch.stack([
+                self.sample_weights(
+                    self.getReplacement(), self.getSubsampleRatio(), n,
+                    seed + i, x.device, w, comm.rank,
                 )
-                fr = binned.fit_frame(learner, y, bag_w, subspaces[i])
-                return self.fit_base_learner(learner, fr, weight_col="weight")
-            return task
+                for i in range(k)
+            ], dim=1)
+            root_rows = [
+                (bag_w[:, i] > 0).nonzero(as_tuple=True)[0].to(torch.int32)
+                for i in range(k)
+            ]
+            models, _ = fit_tree_forest(
+                learner, f_edges, f_bins, y.unsqueeze(1), bag_w, comm,
+                subspaces=subspaces, root_rows=root_rows,
+            )
+        else:
+            if learner is not None and learner.hasParam("maxBins"):
+                binned.get(int(learner.getOrDefault("maxBins")))  # pre-warm
 
-        from ..parallel.streams import parallel_fits
+            def fit_one(i):
+                def task():
+                    bw = self.sample_weights(
+                        self.getReplacement(),
+                        self.getSubsampleRatio(),
+                        n,
+                        seed + i,
+                        x.device,
+                        w,
+                        comm.rank,
+                    )
+                    fr = binned.fit_frame(learner, y, bw, subspaces[i])
+                    return self.fit_base_learner(learner, fr, weight_col="weight")
+                return task
 
-        models = parallel_fits([fit_one(i) for i in range(k)],
-                               self.getParallelism())
+            from ..parallel.streams import parallel_fits
+
+            models = parallel_fits([fit_one(i) for i in range(k)],
+                                   self.getParallelism())
 
         model = BaggingRegressionModel()
         model._models = models
