@@ -10,7 +10,7 @@ import time
 
 import torch
 
-sys.path.insert(0, ".")
+sys.path.insert(0, __import__("os").path.dirname(__import__("os").path.dirname(__import__("os").path.abspath(__file__))))
 import spark_ensemble_amd as sea
 from spark_ensemble_amd.models import DecisionTreeRegressor
 from spark_ensemble_amd.utils.io import synthetic_classification, synthetic_regression

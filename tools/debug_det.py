@@ -2,7 +2,7 @@
 import hashlib
 import sys
 import torch
-sys.path.insert(0, ".")
+sys.path.insert(0, __import__("os").path.dirname(__import__("os").path.dirname(__import__("os").path.abspath(__file__))))
 import spark_ensemble_amd as sea
 from spark_ensemble_amd.boosting.losses import get_classification_loss
 from spark_ensemble_amd.boosting.line_search import optimize_weight_1d

@@ -4,7 +4,7 @@ import time
 
 import torch
 
-sys.path.insert(0, ".")
+sys.path.insert(0, __import__("os").path.dirname(__import__("os").path.dirname(__import__("os").path.abspath(__file__))))
 import spark_ensemble_amd as sea
 from spark_ensemble_amd.models import DecisionTreeRegressor
 from spark_ensemble_amd.parallel import Comm, set_comm

@@ -6,7 +6,7 @@ import time
 import torch
 
 os.environ["SEA_GROW_PROF"] = "1"
-sys.path.insert(0, ".")
+sys.path.insert(0, __import__("os").path.dirname(__import__("os").path.dirname(__import__("os").path.abspath(__file__))))
 import spark_ensemble_amd as sea
 from spark_ensemble_amd.parallel import Comm, set_comm
 from spark_ensemble_amd.utils.io import synthetic_regression

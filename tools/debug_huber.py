@@ -1,5 +1,5 @@
 import sys, torch
-sys.path.insert(0, ".")
+sys.path.insert(0, __import__("os").path.dirname(__import__("os").path.dirname(__import__("os").path.abspath(__file__))))
 import spark_ensemble_amd as sea
 from spark_ensemble_amd.utils.io import synthetic_regression
 from spark_ensemble_amd.boosting.losses import HuberLoss

@@ -1,6 +1,6 @@
 import sys, time
 import torch
-sys.path.insert(0, ".")
+sys.path.insert(0, __import__("os").path.dirname(__import__("os").path.dirname(__import__("os").path.abspath(__file__))))
 import spark_ensemble_amd as sea
 from spark_ensemble_amd.ensemble.binning import BinnedDataset
 from spark_ensemble_amd.ensemble.utils import subspace

@@ -1,7 +1,7 @@
 """Debug harness: split_argmax HIP kernel vs torch reference, verbose."""
 import sys
 import torch
-sys.path.insert(0, ".")
+sys.path.insert(0, __import__("os").path.dirname(__import__("os").path.dirname(__import__("os").path.abspath(__file__))))
 from spark_ensemble_amd.ops import dispatch as hip, reference as ref
 
 def run(n, f, b, c, d, mig, mcw, seed=21):

@@ -1,7 +1,7 @@
 """Per-depth step timing + phase attribution for one GBM round."""
 import sys, time
 import torch
-sys.path.insert(0, ".")
+sys.path.insert(0, __import__("os").path.dirname(__import__("os").path.dirname(__import__("os").path.abspath(__file__))))
 import spark_ensemble_amd as sea
 from spark_ensemble_amd.boosting.losses import get_classification_loss
 from spark_ensemble_amd.boosting.line_search import optimize_weight_1d

@@ -21,7 +21,7 @@ def run_arm(lds: bool):
 
 
 def arm():
-    sys.path.insert(0, ".")
+    sys.path.insert(0, __import__("os").path.dirname(__import__("os").path.dirname(__import__("os").path.abspath(__file__))))
     from spark_ensemble_amd.ops import dispatch
 
     m = dispatch._load_hip()
