@@ -885,7 +885,7 @@ void forest_predict2(torch::Tensor out, torch::Tensor x, torch::Tensor nodes,
   int F = (int)x.size(1);
   const int G = (int)groups.size(0);
   const size_t lds = (size_t)max_group_nodes * 8;
-  TORCH_CHECK(lds <= 131072, "forest_predict2: group too big for LDS");
+  TORCH_CHECK(lds <= 163840, "forest_predict2: group too big for LDS");
   auto stream = at::hip::getCurrentHIPStream();
   const int threads = 1024;
   int rblocks = (int)std::min<int64_t>(ceil_div(n, threads), 8192);

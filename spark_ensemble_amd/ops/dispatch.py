@@ -312,7 +312,7 @@ def forest_predict(x, trees, weights=None, max_depth=64, cache=None):
     return reference.forest_predict(x, trees, weights, max_depth)
 
 
-_FP2_GROUP_NODES = 16384  # 128 KiB of packed 8-B nodes per LDS group
+_FP2_GROUP_NODES = 20352  # ~159 KiB of packed 8-B nodes per LDS group (160 KiB LDS/CU)
 
 
 def _pack_forest(trees, weights, dev):

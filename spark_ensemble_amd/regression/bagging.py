@@ -131,6 +131,74 @@ class BaggingRegressor(Regressor, _BaggingRegressorParams):
             model.set(p, self.getOrDefault(p))
         return model
 
+    # ---- fold-vectorized fitting (OOF stacking fast path) ----------------
+    def _can_fit_folds(self) -> bool:
+        from ..models.tree import DecisionTreeRegressor
+
+        learner = self.getOrNone("baseLearner") or self._default_base_learner()
+        return (
+            type(learner) is DecisionTreeRegressor
+            and learner.getOrDefault("minWeightFractionPerNode") == 0.0
+        )
+
+    def _fit_folds(self, dataset: TensorFrame, fold: torch.Tensor,
+                   num_folds: int) -> List["BaggingRegressionModel"]:
+        """num_folds leave-one-fold-out bagging fits as ONE fused forest:
+        every (fold, member) tree gets its own bag row set (bag weight > 0
+        AND in-fold) and its member's subspace mask — num_folds * k trees
+        grow level-synchronously.  Exact parity with per-fold sequential
+        fits: members reuse the same seed + i bags/subspaces per fold,
+        just as repeated `fit` calls would."""
+        from ..models.tree import fit_tree_forest
+
+        comm = get_comm()
+        learner = self.getOrNone("baseLearner") or self._default_base_learner()
+        seed = self.getOrDefault("seed")
+        k = self.getNumBaseLearners()
+        x, y, w = self._extract_xyw(dataset)
+        n, num_features = x.shape
+        binned = BinnedDataset(x, dataset)
+        edges, bins = binned.get(int(learner.getOrDefault("maxBins")))
+        fold = fold.to(x.device)
+
+        subspaces = [
+            subspace(self.getSubspaceRatio(), num_features, seed + i)
+            for i in range(k)
+        ]
+        bags = [
+            self.sample_weights(
+                self.getReplacement(), self.getSubsampleRatio(), n,
+                seed + i, x.device, w, comm.rank,
+            )
+            for i in range(k)
+        ]
+        cols = []
+        subs_T = []
+        for f in range(num_folds):
+            mask = (fold != f).float()
+            for i in range(k):
+                cols.append(bags[i] * mask)
+                subs_T.append(subspaces[i])
+        w_T = torch.stack(cols, dim=1)
+        root_rows = [
+            (w_T[:, t] > 0).nonzero(as_tuple=True)[0].to(torch.int32)
+            for t in range(w_T.shape[1])
+        ]
+        models, _ = fit_tree_forest(
+            learner, edges, bins, y.unsqueeze(1), w_T, comm,
+            subspaces=subs_T, root_rows=root_rows,
+        )
+        out = []
+        for f in range(num_folds):
+            model = BaggingRegressionModel()
+            model._models = models[f * k:(f + 1) * k]
+            model._subspaces = subspaces
+            model._num_features = num_features
+            for p in ("featuresCol", "labelCol", "predictionCol"):
+                model.set(p, self.getOrDefault(p))
+            out.append(model)
+        return out
+
     def _save_impl(self, path: str):
         persistence.save_metadata(self, path)
         self._save_learner(path)

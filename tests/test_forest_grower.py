@@ -216,6 +216,45 @@ def test_fused_bagging_matches_sequential():
         assert m._num_features == int(sub.numel())
 
 
+def test_bagging_fit_folds_matches_sequential():
+    """BaggingRegressor._fit_folds (all fold x member trees in one
+    forest) must reproduce per-fold weight-masked fits."""
+    import spark_ensemble_amd as sea
+    from spark_ensemble_amd.frame import TensorFrame
+    from spark_ensemble_amd.utils.io import synthetic_regression
+
+    df = synthetic_regression(10000, 14, seed=37)
+    x, y = df["features"], df["label"]
+    T = 3
+    g = torch.Generator().manual_seed(2)
+    fold = torch.randint(0, T, (x.shape[0],), generator=g)
+
+    def mk():
+        return (
+            sea.BaggingRegressor()
+            .setNumBaseLearners(4)
+            .setSubspaceRatio(0.6)
+            .setSubsampleRatio(0.8)
+            .setReplacement(True)
+            .setSeed(9)
+        )
+
+    est = mk()
+    assert est._can_fit_folds()
+    shared = TensorFrame(features=x, label=y, weight=torch.ones_like(y))
+    fused = est._fit_folds(shared, fold, T)
+
+    for f in range(T):
+        wmask = (fold != f).float()
+        seq = mk().set("weightCol", "weight").fit(
+            TensorFrame(features=x, label=y, weight=wmask)
+        )
+        a = fused[f].predict(x)
+        b = seq.predict(x)
+        assert torch.allclose(a, b, rtol=1e-4, atol=1e-5), \
+            (f, float((a - b).abs().max()))
+
+
 def test_hist_build_forest_reference():
     """The per-node column-offset histogram itself."""
     g = torch.Generator().manual_seed(9)

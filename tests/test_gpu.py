@@ -643,6 +643,26 @@ def test_bagging_vote_packed_matches_loop(voting):
         float((fast - slow).abs().max())
 
 
+def test_letter_shape_26class_gbm_gpu():
+    """Letter-shape 26-class logloss GBM: the fused K-tree round grows 26
+    trees per boosting round level-synchronously on the HIP path and the
+    model must clearly beat the class prior."""
+    import spark_ensemble_amd as sea
+    from spark_ensemble_amd.utils.io import synthetic_classification
+
+    df = synthetic_classification(60000, 16, k=26, seed=6, device=DEV)
+    m = (
+        sea.GBMClassifier()
+        .setLoss("logloss")
+        .setNumBaseLearners(2)
+        .setLearningRate(0.5)
+        .setSeed(3)
+        .fit(df)
+    )
+    acc = float((m.transform(df)["prediction"] == df["label"]).float().mean())
+    assert acc > 0.3, acc  # prior ~ 1/26
+
+
 def test_weighted_wide_multiclass_tree_gpu():
     """End-to-end: a weighted 8-class gini tree (BoostingClassifier's
     reweighted rounds hit exactly this shape)."""
