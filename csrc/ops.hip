@@ -802,12 +802,17 @@ void forest_predict(torch::Tensor out, torch::Tensor x, torch::Tensor feat,
 // ---------------------------------------------------------------------------
 
 // BINNED == false: x is [N, F] f32, payload high-32 = f32 threshold bits.
-// BINNED == true:  x is [N, F] u8 rank-transformed rows (each feature
-//   value replaced by its lower-bound rank within the forest's OWN sorted
+// BINNED == true:  x is u8 rank-transformed rows (each feature value
+//   replaced by its lower-bound rank within the forest's OWN sorted
 //   per-feature threshold set — an EXACT transform: x <= thr  <=>
 //   rank(x) <= rank(thr)), payload high-32 = the threshold's rank.
 //   Rows shrink 4x (u8 vs f32), so the divergent per-lane gathers stop
 //   thrashing L1 (256 B rows: 4 cache lines instead of 16).
+//   xt != 0: the u8 matrix is TRANSPOSED [F, N] — consecutive lanes hold
+//   consecutive rows, and all lanes of a wave sit on the SAME node for
+//   the first tree levels (they entered the tree together), so their
+//   gathers land on consecutive bytes of one feature column: hop h costs
+//   <= min(2^h, 64) cache lines instead of 64.
 template <bool D1, bool BINNED>
 __global__ void forest_predict2_kernel(
     float* __restrict__ out,                       // [N, D] (+=, pre-zeroed)
@@ -817,7 +822,7 @@ __global__ void forest_predict2_kernel(
     const int* __restrict__ tree_off,              // [T] global node base
     const float* __restrict__ w,                   // [T]
     const int* __restrict__ groups,                // [G, 4]
-    int64_t n, int F, int D) {
+    int64_t n, int F, int D, int xt) {
   extern __shared__ unsigned long long tlds[];  // group nodes
   const int* grp = groups + blockIdx.y * 4;
   const int first_tree = grp[0];
@@ -832,7 +837,9 @@ __global__ void forest_predict2_kernel(
   const int64_t stride = gridDim.x * (int64_t)blockDim.x;
   for (; r < n; r += stride) {
     const float* xr = BINNED ? nullptr : (const float*)xv_ + r * F;
-    const uint8_t* br = BINNED ? (const uint8_t*)xv_ + r * F : nullptr;
+    const uint8_t* br =
+        BINNED ? (const uint8_t*)xv_ + (xt ? r : r * F) : nullptr;
+    // transposed stride: br[f * n] addresses feature f of this row
     float acc[8];
     if (!D1)
 #pragma unroll
@@ -847,7 +854,8 @@ __global__ void forest_predict2_kernel(
         const int left = (short)((nd >> 16) & 0xFFFFu);
         bool go_left;
         if (BINNED) {
-          go_left = br[f] <= (unsigned)(nd >> 32);
+          const uint8_t bv = xt ? br[(int64_t)f * n] : br[f];
+          go_left = bv <= (unsigned)(nd >> 32);
         } else {
           go_left = xr[f] <= __uint_as_float((unsigned)(nd >> 32));
         }
@@ -876,13 +884,14 @@ __global__ void forest_predict2_kernel(
 void forest_predict2(torch::Tensor out, torch::Tensor x, torch::Tensor nodes,
                      torch::Tensor leaf, torch::Tensor tree_off,
                      torch::Tensor w, torch::Tensor groups, int64_t D,
-                     int64_t max_group_nodes) {
+                     int64_t max_group_nodes, int64_t x_transposed) {
   CHECK_GPU(out); CHECK_GPU(x); CHECK_GPU(nodes); CHECK_GPU(groups);
   CHECK_CONTIG(out); CHECK_CONTIG(x); CHECK_CONTIG(nodes);
   TORCH_CHECK(D <= 8, "forest_predict2: D <= 8");
   const bool binned = x.scalar_type() == torch::kUInt8;
-  int64_t n = x.size(0);
-  int F = (int)x.size(1);
+  TORCH_CHECK(!x_transposed || binned, "x_transposed implies u8 input");
+  int64_t n = x.size(x_transposed ? 1 : 0);
+  int F = (int)x.size(x_transposed ? 0 : 1);
   const int G = (int)groups.size(0);
   const size_t lds = (size_t)max_group_nodes * 8;
   TORCH_CHECK(lds <= 163840, "forest_predict2: group too big for LDS");
@@ -904,7 +913,8 @@ void forest_predict2(torch::Tensor out, torch::Tensor x, torch::Tensor nodes,
                        (const unsigned long long*)nodes.data_ptr<int64_t>(),  \
                        leaf.numel() ? leaf.data_ptr<float>() : nullptr,       \
                        tree_off.data_ptr<int>(), w.data_ptr<float>(),         \
-                       groups.data_ptr<int>(), n, F, (int)D);                 \
+                       groups.data_ptr<int>(), n, F, (int)D,                  \
+                       (int)x_transposed);                                    \
   } while (0)
   if (D == 1 && binned) FP2_LAUNCH(true, true);
   else if (D == 1) FP2_LAUNCH(true, false);

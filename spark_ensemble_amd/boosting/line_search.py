@@ -35,7 +35,7 @@ def _eval(loss, label, pred, direction, weight, coeff, want_hess=False):
     fused kernel pass (csrc line_search_eval) instead of ~6 eager passes.
     ``want_hess`` (scalar dim only) additionally accumulates
     sum_i w_i d_i^2 loss''(p_i + a d_i) for the Newton weight search."""
-    if pred.is_cuda:
+    if pred.is_cuda and pred.shape[1] <= 8:
         from ..ops import dispatch
 
         m = dispatch._require_hip("line_search_eval")

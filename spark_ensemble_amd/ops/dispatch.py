@@ -455,55 +455,70 @@ def _binned_edges(pack, F):
     return edges
 
 
-def _run_fp2(m, x_in, pack, binned):
+def _run_fp2(m, x_in, pack, mode):
+    """mode: 'raw' (f32 rows), 'binned' (u8 rank rows), 'binned_t'
+    (transposed u8 — wave-coalesced top-level gathers)."""
     out = torch.zeros(x_in.shape[0], pack["D"], dtype=torch.float32,
                       device=x_in.device)
-    if binned:
-        edges = _binned_edges(pack, x_in.shape[1])
-        xb = torch.empty(x_in.shape, dtype=torch.uint8, device=x_in.device)
-        m.bin_features(xb, x_in, edges)
-        m.forest_predict2(out, xb, pack["node64b"], pack["leaves"],
-                          pack["offsets32"], pack["w"], pack["groups_t"],
-                          pack["D"], pack["max_nodes"])
-    else:
+    if mode == "raw":
         m.forest_predict2(out, x_in, pack["node64"], pack["leaves"],
                           pack["offsets32"], pack["w"], pack["groups_t"],
-                          pack["D"], pack["max_nodes"])
+                          pack["D"], pack["max_nodes"], 0)
+        return out
+    edges = _binned_edges(pack, x_in.shape[1])
+    xb = torch.empty(x_in.shape, dtype=torch.uint8, device=x_in.device)
+    m.bin_features(xb, x_in, edges)
+    if mode == "binned_t":
+        xb = xb.t().contiguous()
+        m.forest_predict2(out, xb, pack["node64b"], pack["leaves"],
+                          pack["offsets32"], pack["w"], pack["groups_t"],
+                          pack["D"], pack["max_nodes"], 1)
+    else:
+        m.forest_predict2(out, xb, pack["node64b"], pack["leaves"],
+                          pack["offsets32"], pack["w"], pack["groups_t"],
+                          pack["D"], pack["max_nodes"], 0)
     return out
+
+
+_SERVE_MODES = ("binned_t", "binned", "raw")
 
 
 def _forest_predict_packed(m, x, pack):
     D = pack["D"]
     if pack["v2"] and x.shape[1] < 32768 and hasattr(m, "forest_predict2"):
         xc = x.contiguous()
+        env_mode = os.environ.get("SEA_SERVE_MODE")
+        if os.environ.get("SEA_SERVE_RAW") == "1":
+            env_mode = "raw"
         binned_avail = (
             pack.get("binned_ok") and pack["bin_fmax"] <= x.shape[1]
-            and os.environ.get("SEA_SERVE_RAW") != "1"
         )
         if not binned_avail:
-            return _run_fp2(m, xc, pack, False)
+            return _run_fp2(m, xc, pack, "raw")
+        if env_mode in _SERVE_MODES:
+            return _run_fp2(m, xc, pack, env_mode)
         # which walk mode wins depends on the forest's x-access pattern
         # (measured r02: binned 1.8x faster for a 100-tree GBM, raw 1.7x
-        # faster for 50 subspace-bagged trees) — both are EXACT, so on
-        # the first large batch time both once and remember the winner
+        # faster for 50 subspace-bagged trees) — all modes are EXACT, so
+        # on the first large batch time each once and remember the winner
         mode = pack.get("mode")
         if mode is None:
             if x.shape[0] < (1 << 20):
-                return _run_fp2(m, xc, pack, True)  # small batch: either way
+                return _run_fp2(m, xc, pack, "binned")  # small: either way
             import time
 
             probe = xc[: 1 << 19]
             times = {}
-            for b in (True, False):
-                _run_fp2(m, probe, pack, b)  # warm
+            for cand in _SERVE_MODES:
+                _run_fp2(m, probe, pack, cand)  # warm
                 torch.cuda.synchronize()
                 t0 = time.perf_counter()
-                _run_fp2(m, probe, pack, b)
+                _run_fp2(m, probe, pack, cand)
                 torch.cuda.synchronize()
-                times[b] = time.perf_counter() - t0
-            mode = "binned" if times[True] <= times[False] else "raw"
+                times[cand] = time.perf_counter() - t0
+            mode = min(times, key=times.get)
             pack["mode"] = mode
-        return _run_fp2(m, xc, pack, mode == "binned")
+        return _run_fp2(m, xc, pack, mode)
     out = torch.zeros(x.shape[0], D, dtype=torch.float32, device=x.device)
     m.forest_predict(out, x.contiguous(), pack["feats"], pack["thrs"],
                      pack["lefts"], pack["leaves"], pack["offsets32"],

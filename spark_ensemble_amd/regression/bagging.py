@@ -142,7 +142,7 @@ class BaggingRegressor(Regressor, _BaggingRegressorParams):
         )
 
     def _fit_folds(self, dataset: TensorFrame, fold: torch.Tensor,
-                   num_folds: int) -> List["BaggingRegressionModel"]:
+                   num_folds: int, include_full: bool = False):
         """num_folds leave-one-fold-out bagging fits as ONE fused forest:
         every (fold, member) tree gets its own bag row set (bag weight > 0
         AND in-fold) and its member's subspace mask — num_folds * k trees
@@ -179,6 +179,13 @@ class BaggingRegressor(Regressor, _BaggingRegressorParams):
             for i in range(k):
                 cols.append(bags[i] * mask)
                 subs_T.append(subspaces[i])
+        n_groups = num_folds
+        if include_full:
+            # the final full-data refit rides in the same fused forest
+            for i in range(k):
+                cols.append(bags[i])
+                subs_T.append(subspaces[i])
+            n_groups += 1
         w_T = torch.stack(cols, dim=1)
         root_rows = [
             (w_T[:, t] > 0).nonzero(as_tuple=True)[0].to(torch.int32)
@@ -189,7 +196,7 @@ class BaggingRegressor(Regressor, _BaggingRegressorParams):
             subspaces=subs_T, root_rows=root_rows,
         )
         out = []
-        for f in range(num_folds):
+        for f in range(n_groups):
             model = BaggingRegressionModel()
             model._models = models[f * k:(f + 1) * k]
             model._subspaces = subspaces
@@ -197,6 +204,8 @@ class BaggingRegressor(Regressor, _BaggingRegressorParams):
             for p in ("featuresCol", "labelCol", "predictionCol"):
                 model.set(p, self.getOrDefault(p))
             out.append(model)
+        if include_full:
+            return out[:-1], out[-1]
         return out
 
     def _save_impl(self, path: str):

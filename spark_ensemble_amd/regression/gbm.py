@@ -324,7 +324,7 @@ class GBMRegressor(Regressor, _GBMRegressorParams):
         )
 
     def _fit_folds(self, dataset: TensorFrame, fold: torch.Tensor,
-                   num_folds: int) -> List["GBMRegressionModel"]:
+                   num_folds: int, include_full: bool = False):
         """num_folds leave-one-fold-out GBM fits grown JOINTLY: every
         boosting round builds ONE fused forest (grow_forest with per-tree
         root row sets = each fold's training rows) and the squared-loss
@@ -353,6 +353,11 @@ class GBMRegressor(Regressor, _GBMRegressorParams):
 
         fold = fold.to(device)
         masks = [(fold != f) for f in range(T)]
+        if include_full:
+            # one extra "fold" seeing every row = the final full-data
+            # refit, grown in the same fused forest
+            masks.append(torch.ones_like(fold, dtype=torch.bool))
+            T = T + 1
         root_rows = [m.nonzero(as_tuple=True)[0].to(torch.int32)
                      for m in masks]
         mask_f = torch.stack([m.float() for m in masks], dim=1)  # [N, T]
@@ -422,6 +427,8 @@ class GBMRegressor(Regressor, _GBMRegressorParams):
             for p in ("featuresCol", "labelCol", "predictionCol"):
                 model.set(p, self.getOrDefault(p))
             out.append(model)
+        if include_full:
+            return out[:-1], out[-1]
         return out
 
     def _save_impl(self, path: str):

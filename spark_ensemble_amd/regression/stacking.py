@@ -124,9 +124,10 @@ class StackingRegressor(Regressor, _StackingRegressorParams):
                 and lr._can_fit_folds()
             }
             fold_models: dict = {}
+            final_models: dict = {}
             for mi in fused:
-                fold_models[mi] = learners[mi]._fit_folds(
-                    shared, fold, num_folds
+                fold_models[mi], final_models[mi] = learners[mi]._fit_folds(
+                    shared, fold, num_folds, include_full=True
                 )
 
             def fold_task(lr, f):
@@ -141,22 +142,26 @@ class StackingRegressor(Regressor, _StackingRegressorParams):
             plan = [(mi, f) for mi, lr in enumerate(learners)
                     if mi not in fused for f in range(num_folds)]
             tasks = [fold_task(learners[mi], f) for mi, f in plan]
-            # final base models refit on everything, same pool
+            # final refits for NON-fused learners (fused ones grew theirs
+            # inside the fold forest), same pool
+            non_fused = [mi for mi in range(len(learners)) if mi not in fused]
             tasks += [
-                (lambda lr=lr: self.fit_base_learner(
+                (lambda lr=learners[mi]: self.fit_base_learner(
                     lr, shared, weight_col="weight"))
-                for lr in learners
+                for mi in non_fused
             ]
             fitted = parallel_fits(tasks, self.getParallelism(),
                                    warm_first=True)
             for j, (mi, f) in enumerate(plan):
                 fold_models.setdefault(mi, [None] * num_folds)[f] = fitted[j]
+            for j, mi in enumerate(non_fused):
+                final_models[mi] = fitted[len(plan) + j]
             for mi in range(len(learners)):
                 for f in range(num_folds):
                     m = fold_models[mi][f]
                     sel = fold == f
                     meta[sel, mi] = m.predict(x[sel])
-            models = fitted[len(plan):]
+            models = [final_models[mi] for mi in range(len(learners))]
 
         stack = self.fit_base_learner(
             stacker,
