@@ -269,13 +269,18 @@ class CrossValidator(Estimator):
     winner and refits it on the full data."""
 
     def __init__(self, uid=None, estimator=None, estimatorParamMaps=None,
-                 evaluator=None, numFolds=3, seed=0):
+                 evaluator=None, numFolds=3, seed=0, parallelism=1):
         super().__init__(uid)
         self.estimator = estimator
         self.estimatorParamMaps = estimatorParamMaps or [{}]
         self.evaluator = evaluator
         self.numFolds = numFolds
         self.seed = seed
+        self.parallelism = parallelism
+
+    def setParallelism(self, v):
+        self.parallelism = int(v)
+        return self
 
     def setEstimator(self, v):
         self.estimator = v
@@ -304,15 +309,26 @@ class CrossValidator(Estimator):
     def _fit(self, dataset: TensorFrame) -> "CrossValidatorModel":
         assert self.estimator is not None and self.evaluator is not None
         folds = self._fold_ids(dataset.count(), dataset.device)
-        avg = []
-        for pm in self.estimatorParamMaps:
-            scores = []
-            for f in range(self.numFolds):
+
+        def cell(pm, f):
+            def task():
                 tr = dataset.filter(folds != f)
                 va = dataset.filter(folds == f)
                 model = self.estimator.fit(tr, pm or None)
-                scores.append(self.evaluator.evaluate(model.transform(va)))
-            avg.append(sum(scores) / len(scores))
+                return self.evaluator.evaluate(model.transform(va))
+            return task
+
+        # (param-map, fold) fits run concurrently on per-thread HIP
+        # streams (the Spark CrossValidator parallelism analog)
+        from .parallel.streams import parallel_fits
+
+        tasks = [cell(pm, f) for pm in self.estimatorParamMaps
+                 for f in range(self.numFolds)]
+        flat = parallel_fits(tasks, self.parallelism, warm_first=True)
+        avg = []
+        for i in range(len(self.estimatorParamMaps)):
+            chunk = flat[i * self.numFolds:(i + 1) * self.numFolds]
+            avg.append(sum(chunk) / len(chunk))
         better = max if self.evaluator.isLargerBetter() else min
         best_idx = avg.index(better(avg))
         best_model = self.estimator.fit(
