@@ -833,88 +833,53 @@ __global__ void forest_predict2_kernel(
     tlds[i] = nodes[node_base + i];
   __syncthreads();
 
-  // two rows per lane, consecutive-wave-window pairs: each hop issues
-  // TWO independent LDS-node + x-gather chains (doubled memory-level
-  // parallelism) while wave rows stay consecutive for the transposed
-  // coalescing
-  const int64_t span = (int64_t)blockDim.x;
-  const int64_t stride = gridDim.x * span * 2;
-  for (int64_t r0 = blockIdx.x * span * 2 + threadIdx.x; r0 < n;
-       r0 += stride) {
-    const int64_t r1 = r0 + span;
-    const bool has1 = r1 < n;
-    const float* xr0 = BINNED ? nullptr : (const float*)xv_ + r0 * F;
-    const float* xr1 =
-        (BINNED || !has1) ? nullptr : (const float*)xv_ + r1 * F;
-    const uint8_t* b0 =
-        BINNED ? (const uint8_t*)xv_ + (xt ? r0 : r0 * F) : nullptr;
-    const uint8_t* b1 = (BINNED && has1)
-        ? (const uint8_t*)xv_ + (xt ? r1 : r1 * F) : nullptr;
-    float acc[2][8];
+  // one row per lane: a 2-row-per-lane ILP variant was measured 1.5-4x
+  // SLOWER (merged divergent walk loops serialize both chains through
+  // every iteration); wave-level parallelism already covers the latency
+  int64_t r = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  const int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; r < n; r += stride) {
+    const float* xr = BINNED ? nullptr : (const float*)xv_ + r * F;
+    const uint8_t* br =
+        BINNED ? (const uint8_t*)xv_ + (xt ? r : r * F) : nullptr;
+    // transposed stride: br[f * n] addresses feature f of this row
+    float acc[8];
     if (!D1)
 #pragma unroll
-      for (int u = 0; u < 2; ++u)
-#pragma unroll
-        for (int d = 0; d < 8; ++d) acc[u][d] = 0.0f;
-    float acc1_0 = 0.0f, acc1_1 = 0.0f;
+      for (int d = 0; d < 8; ++d) acc[d] = 0.0f;
+    float acc1 = 0.0f;
     for (int t = 0; t < n_trees; ++t) {
       const int toff = tree_off[first_tree + t] - node_base;
-      int n0 = toff, n1 = toff;
-      unsigned long long nd0 = tlds[toff];
-      unsigned long long nd1 = nd0;
-      int f0 = (short)(nd0 & 0xFFFFu);
-      int f1 = has1 ? f0 : -1;
-      while (f0 >= 0 || f1 >= 0) {
-        if (f0 >= 0) {
-          bool gl;
-          if (BINNED) {
-            const uint8_t bv = xt ? b0[(int64_t)f0 * n] : b0[f0];
-            gl = bv <= (unsigned)(nd0 >> 32);
-          } else {
-            gl = xr0[f0] <= __uint_as_float((unsigned)(nd0 >> 32));
-          }
-          n0 = toff + (short)((nd0 >> 16) & 0xFFFFu) + (gl ? 0 : 1);
-          nd0 = tlds[n0];
-          f0 = (short)(nd0 & 0xFFFFu);
+      int node = toff;
+      unsigned long long nd = tlds[node];
+      int f = (short)(nd & 0xFFFFu);
+      while (f >= 0) {
+        const int left = (short)((nd >> 16) & 0xFFFFu);
+        bool go_left;
+        if (BINNED) {
+          const uint8_t bv = xt ? br[(int64_t)f * n] : br[f];
+          go_left = bv <= (unsigned)(nd >> 32);
+        } else {
+          go_left = xr[f] <= __uint_as_float((unsigned)(nd >> 32));
         }
-        if (f1 >= 0) {
-          bool gl;
-          if (BINNED) {
-            const uint8_t bv = xt ? b1[(int64_t)f1 * n] : b1[f1];
-            gl = bv <= (unsigned)(nd1 >> 32);
-          } else {
-            gl = xr1[f1] <= __uint_as_float((unsigned)(nd1 >> 32));
-          }
-          n1 = toff + (short)((nd1 >> 16) & 0xFFFFu) + (gl ? 0 : 1);
-          nd1 = tlds[n1];
-          f1 = (short)(nd1 & 0xFFFFu);
-        }
+        node = toff + left + (go_left ? 0 : 1);
+        nd = tlds[node];
+        f = (short)(nd & 0xFFFFu);
       }
       if (D1) {
-        acc1_0 += __uint_as_float((unsigned)(nd0 >> 32));
-        if (has1) acc1_1 += __uint_as_float((unsigned)(nd1 >> 32));
+        acc1 += __uint_as_float((unsigned)(nd >> 32));  // weighted leaf
       } else {
         const float wt = w[first_tree + t];
-        const float* lv0 = leaf + (int64_t)(node_base + n0) * D;
-        for (int d = 0; d < D; ++d) acc[0][d] += wt * lv0[d];
-        if (has1) {
-          const float* lv1 = leaf + (int64_t)(node_base + n1) * D;
-          for (int d = 0; d < D; ++d) acc[1][d] += wt * lv1[d];
-        }
+        const float* lv = leaf + (int64_t)(node_base + node) * D;
+        for (int d = 0; d < D; ++d) acc[d] += wt * lv[d];
       }
     }
     if (D1) {
-      atomicAdd(out + r0, acc1_0);
-      if (has1) atomicAdd(out + r1, acc1_1);
+      atomicAdd(out + r, acc1);
     } else {
-      float* o0 = out + r0 * D;
+      float* o = out + r * D;
       for (int d = 0; d < D; ++d)
-        if (acc[0][d] != 0.0f) atomicAdd(o0 + d, acc[0][d]);
-      if (has1) {
-        float* o1 = out + r1 * D;
-        for (int d = 0; d < D; ++d)
-          if (acc[1][d] != 0.0f) atomicAdd(o1 + d, acc[1][d]);
-      }
+        if (acc[d] != 0.0f) atomicAdd(o + d, acc[d]);
     }
   }
 }
@@ -935,10 +900,10 @@ void forest_predict2(torch::Tensor out, torch::Tensor x, torch::Tensor nodes,
   TORCH_CHECK(lds <= 163840, "forest_predict2: group too big for LDS");
   auto stream = at::hip::getCurrentHIPStream();
   const int threads = 1024;
-  int rblocks = (int)std::min<int64_t>(ceil_div(n, threads * 2), 8192);
+  int rblocks = (int)std::min<int64_t>(ceil_div(n, threads), 8192);
   // fill the chip even for few groups
   rblocks = std::max(rblocks, (int)std::min<int64_t>(
-                                  ceil_div(512, (int64_t)G), ceil_div(n, 128)));
+                                  ceil_div(512, (int64_t)G), ceil_div(n, 64)));
 #define FP2_LAUNCH(DD1, BB)                                                   \
   do {                                                                        \
     if (lds > 65536)                                                          \
