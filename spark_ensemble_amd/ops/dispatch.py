@@ -469,7 +469,10 @@ def _run_fp2(m, x_in, pack, mode):
     xb = torch.empty(x_in.shape, dtype=torch.uint8, device=x_in.device)
     m.bin_features(xb, x_in, edges)
     if mode == "binned_t":
-        xb = xb.t().contiguous()
+        xbt = torch.empty(xb.shape[1], xb.shape[0], dtype=torch.uint8,
+                          device=xb.device)
+        m.transpose_u8(xbt, xb)
+        xb = xbt
         m.forest_predict2(out, xb, pack["node64b"], pack["leaves"],
                           pack["offsets32"], pack["w"], pack["groups_t"],
                           pack["D"], pack["max_nodes"], 1)

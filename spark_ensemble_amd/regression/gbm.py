@@ -274,9 +274,12 @@ class GBMRegressor(Regressor, _GBMRegressorParams):
                 elif err < best_err:
                     best_err = err
                     v = 0
-                instr.log_round(i, weight=weight, val_loss=err, patience=v)
+                from ..boosting import line_search as _ls
+                instr.log_round(i, weight=weight, val_loss=err, patience=v,
+                                ls_evals=_ls.LAST_EVALS)
             else:
-                instr.log_round(i, weight=weight)
+                from ..boosting import line_search as _ls
+                instr.log_round(i, weight=weight, ls_evals=_ls.LAST_EVALS)
             interval = self.getCheckpointInterval()
             if ckpt_dir and interval > 0 and (i + 1) % interval == 0:
                 ckpt.save_round_state(

@@ -643,6 +643,17 @@ def test_bagging_vote_packed_matches_loop(voting):
         float((fast - slow).abs().max())
 
 
+@pytest.mark.parametrize("n,f", [(4096, 256), (1000, 37), (64, 64), (70, 5)])
+def test_transpose_u8_matches_torch(hip, n, f):
+    import spark_ensemble_amd._hip_ops as m
+
+    g = torch.Generator().manual_seed(13)
+    x = torch.randint(0, 256, (n, f), generator=g, dtype=torch.uint8).to(DEV)
+    out = torch.empty(f, n, dtype=torch.uint8, device=DEV)
+    m.transpose_u8(out, x)
+    assert torch.equal(out, x.t().contiguous())
+
+
 def test_letter_shape_26class_gbm_gpu():
     """Letter-shape 26-class logloss GBM: the fused K-tree round grows 26
     trees per boosting round level-synchronously on the HIP path and the
