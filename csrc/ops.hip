@@ -188,8 +188,10 @@ void bin_features(torch::Tensor out, torch::Tensor x, torch::Tensor edges) {
   auto stream = at::hip::getCurrentHIPStream();
   const int threads = 256;
   const int fgroups = (int)ceil_div(F, 16);
+  // 4 blocks/CU: a 1-block/CU grid (old 4096/fgroups cap) left the
+  // float4 row loads latency-bound at ~1.2 TB/s (rocprof r02)
   int rblocks = (int)std::min<int64_t>(
-      ceil_div(n, threads), std::max<int64_t>(1, 4096 / fgroups));
+      ceil_div(n, threads), std::max<int64_t>(1, 16384 / fgroups));
   const size_t lds = (size_t)16 * nedges * 4;
   hipLaunchKernelGGL(bin_features_kernel, dim3(rblocks, fgroups),
                      dim3(threads), lds, stream, out.data_ptr<uint8_t>(),
