@@ -190,16 +190,22 @@ class GBMClassifier(ProbabilisticClassifier, _GBMClassifierParams):
 
         instr.timers["setup_ms"] = instr.elapsed_ms()
 
-        # fused-round eligibility that depends only on params (checked
-        # once); the per-round weight positivity check needs a sync and
-        # is done lazily inside _can_fuse_round
+        # fused-round eligibility.  FIT-LOCAL state (not attributes on
+        # self): concurrent fits of one shared estimator instance (fold
+        # tasks on the stream pool) must not see each other's cached
+        # weight-positivity answer — a stale True would run the fused
+        # grower on zero-weight rows.  The static part depends only on
+        # params; the positivity check needs one sync, done lazily on
+        # the first round.
         from ..models.tree import DecisionTreeRegressor
 
-        self._fuse_static_ok = (
-            type(learner) is DecisionTreeRegressor
-            and learner.getOrDefault("minWeightFractionPerNode") == 0.0
-        )
-        self._fuse_w_ok: Optional[bool] = None
+        fuse_state = {
+            "static_ok": (
+                type(learner) is DecisionTreeRegressor
+                and learner.getOrDefault("minWeightFractionPerNode") == 0.0
+            ),
+            "w_ok": None,
+        }
 
         while i < k_stages and v < self.getOrDefault("numRounds"):
             idx = subspaces[i]
@@ -236,7 +242,8 @@ class GBMClassifier(ProbabilisticClassifier, _GBMClassifierParams):
             # round — tree_grower.grow_forest).  Generic base learners
             # keep the sequential per-class path.
             directions = None
-            if dim > 1 and self._can_fuse_round(learner, res_weight):
+            if dim > 1 and self._can_fuse_round(learner, res_weight,
+                                                fuse_state):
                 from ..models.tree import fit_tree_forest
 
                 f_edges, f_bins = binned.sliced_binned(
@@ -341,18 +348,20 @@ class GBMClassifier(ProbabilisticClassifier, _GBMClassifierParams):
             model.set(p, self.getOrDefault(p))
         return model
 
-    def _can_fuse_round(self, learner, res_weight) -> bool:
+    def _can_fuse_round(self, learner, res_weight, fuse_state) -> bool:
         """Fused K-tree rounds need: built-in tree learner, no per-tree
         weight thresholds, and strictly positive weights on every row
         (the fused grower has no zero-weight row mask).  Sub-sampling
-        introduces zeros, so only the full-bag configs fuse."""
-        if not getattr(self, "_fuse_static_ok", False):
+        introduces zeros, so only the full-bag configs fuse.
+        ``fuse_state`` is fit-local (see _fit) so concurrent fits of a
+        shared estimator cannot cross-contaminate the cached answer."""
+        if not fuse_state.get("static_ok", False):
             return False
         if self.getSubsampleRatio() < 1.0 or self.getReplacement():
             return False
-        if self._fuse_w_ok is None:
-            self._fuse_w_ok = bool((res_weight > 0).all())
-        return self._fuse_w_ok
+        if fuse_state["w_ok"] is None:
+            fuse_state["w_ok"] = bool((res_weight > 0).all())
+        return fuse_state["w_ok"]
 
     def _save_impl(self, path: str):
         persistence.save_metadata(self, path)

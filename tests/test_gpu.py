@@ -577,12 +577,11 @@ def test_fused_multiclass_round_matches_sequential():
         .setSeed(9).fit(df)
     est_seq = sea.GBMClassifier().setLoss("logloss").setNumBaseLearners(3) \
         .setSeed(9)
-    est_seq._fuse_static_ok = False  # force the sequential path
 
-    # _fuse_static_ok is recomputed inside _fit; patch the method instead
+    # force the sequential path by patching the eligibility method
     import spark_ensemble_amd.classification.gbm as gbm_mod
     orig = gbm_mod.GBMClassifier._can_fuse_round
-    gbm_mod.GBMClassifier._can_fuse_round = lambda self, l, w: False
+    gbm_mod.GBMClassifier._can_fuse_round = lambda self, l, w, st: False
     try:
         m_seq = est_seq.fit(df)
     finally:
