@@ -131,8 +131,14 @@ def optimize_weight_nd(
     comm: Optional[Comm] = None,
     max_iter: int = 100,
     tol: float = 1e-6,
+    x0: Optional[np.ndarray] = None,
 ) -> np.ndarray:
-    """L-BFGS-B with per-dim bounds [0, inf), started at ones(dim)."""
+    """L-BFGS-B with per-dim bounds [0, inf), started at ``x0`` (default
+    ones(dim), the reference's start — GBMClassifier.scala:427).  Callers
+    may warm-start from the previous round's solution: the minimizer of
+    the convex per-round problem is start-independent, and stage weights
+    drift slowly across rounds, so the search converges in fewer
+    full-data evaluations (each is a kernel + all-reduce + host sync)."""
     from scipy.optimize import minimize
 
     dim = pred.shape[1]
@@ -150,9 +156,10 @@ def optimize_weight_nd(
             g = np.nan_to_num(g, nan=0.0, posinf=1e30, neginf=-1e30)
         return v, g
 
+    start = np.ones(dim) if x0 is None else np.clip(np.asarray(x0, float), 0.0, None)
     res = minimize(
         f,
-        np.ones(dim),
+        start,
         jac=True,
         method="L-BFGS-B",
         bounds=[(0.0, None)] * dim,

@@ -188,6 +188,7 @@ class GBMClassifier(ProbabilisticClassifier, _GBMClassifierParams):
             i = r0
             instr.log_named_value("resumed_from_round", r0)
 
+        prev_sol = None  # warm start for the n-D stage-weight search
         instr.timers["setup_ms"] = instr.elapsed_ms()
 
         # fused-round eligibility.  FIT-LOCAL state (not attributes on
@@ -291,7 +292,9 @@ class GBMClassifier(ProbabilisticClassifier, _GBMClassifierParams):
                     sol = optimize_weight_nd(
                         loss, ylab, predictions, directions, bag_w, comm,
                         self.getOrDefault("maxIter"), self.getOrDefault("tol"),
+                        x0=prev_sol,
                     ).tolist()
+                    prev_sol = sol
             else:
                 sol = [1.0] * dim
             iweights = [s * lr_rate for s in sol]
