@@ -220,8 +220,8 @@ def grow_tree(
                 identity_rows=(row_mask is None and bins.is_cuda),
             )
             hists, gain, feat, b, left_stats = _finish_level_split(
-                new_h, None, None, None, None, n_active, (F, B, C), device,
-                comm, split_args,
+                new_h, None, None, None, None, None, n_active, (F, B, C),
+                device, comm, split_args,
             )
         else:
             # build only the flagged (smaller) children, then subtract
@@ -244,9 +244,17 @@ def grow_tree(
                     bins, gh, build_rows, torch.tensor(b_off, dtype=torch.int64),
                     B, D, gh_max,
                 )
+            nb = [j for j in range(n_active) if not built_mask[j]]
             hists, gain, feat, b, left_stats = _finish_level_split(
-                bh, built_idx, hists, parent_of, built_mask, n_active,
-                (F, B, C), device, comm, split_args,
+                bh,
+                torch.as_tensor(built_idx, dtype=torch.long, device=device),
+                hists,
+                torch.as_tensor(nb, dtype=torch.long, device=device),
+                torch.as_tensor([j ^ 1 for j in nb], dtype=torch.long,
+                                device=device),
+                torch.as_tensor([parent_of[j] for j in nb], dtype=torch.long,
+                                device=device),
+                n_active, (F, B, C), device, comm, split_args,
             )
 
         _tick("hist+split", _t0)
@@ -391,10 +399,11 @@ MAX_FUSED_TREES = 32  # bounds the [T*N] row-index arena per fused batch
 
 def _finish_level_split(
     built: Optional[torch.Tensor],  # [n_built, F, B, C] LOCAL sums (or None)
-    built_idx,                      # active indices that were built (root: all)
+    built_t: Optional[torch.Tensor],  # device long: built active indices
     parent_hists: Optional[torch.Tensor],  # [n_parent, F, B, C] GLOBAL
-    parent_of,                      # active idx -> parent row (non-root)
-    built_mask,                     # per active idx (non-root)
+    nb_t: Optional[torch.Tensor],   # device long: non-built active indices
+    sib_t: Optional[torch.Tensor],  # device long: their (built) siblings
+    par_t: Optional[torch.Tensor],  # device long: their parent rows
     n_active: int,
     shape,                          # (F, B, C)
     device,
@@ -429,19 +438,9 @@ def _finish_level_split(
         if root:
             hists[:, f0:f1] = ch
             return
-        if built_idx is not None and len(built_idx):
-            bi = torch.as_tensor(built_idx, dtype=torch.long, device=device)
-            hists[bi, f0:f1] = ch
-        nb = [j for j in range(n_active) if not built_mask[j]]
-        if nb:
-            nb_t = torch.as_tensor(nb, dtype=torch.long, device=device)
-            sib_t = torch.as_tensor(
-                [j - 1 if j % 2 == 1 else j + 1 for j in nb],
-                dtype=torch.long, device=device,
-            )
-            par_t = torch.as_tensor(
-                [parent_of[j] for j in nb], dtype=torch.long, device=device
-            )
+        if built_t is not None and built_t.numel():
+            hists[built_t, f0:f1] = ch
+        if nb_t is not None and nb_t.numel():
             hists[nb_t, f0:f1] = (
                 parent_hists[par_t, f0:f1] - hists[sib_t, f0:f1]
             )
@@ -585,20 +584,17 @@ def grow_forest(
         root_lens = [int(r.numel()) for r in root_rows]
         assert sum(root_lens) < 2**31
 
-    # per-tree flat node arrays
-    feats = [[] for _ in range(T)]
-    thrs = [[] for _ in range(T)]
-    lefts = [[] for _ in range(T)]
-    leaves = [[] for _ in range(T)]
+    # per-tree node arrays, PREALLOCATED: every per-level decision below
+    # is tensorized (fancy-indexed writes over (tree, node) pairs) — the
+    # r02 per-node Python loops cost ~80 ms/round at letter shape
+    # (26 trees x deep levels) against ~4 ms of kernel work
+    max_nodes = 2 ** (params.max_depth + 1) - 1
+    feats_t = torch.full((T, max_nodes), -1, dtype=torch.int32)
+    thrs_t = torch.zeros(T, max_nodes, dtype=torch.float32)
+    lefts_t = torch.full((T, max_nodes), -1, dtype=torch.int32)
+    leaves_t = torch.zeros(T, max_nodes, dtype=torch.float32)
+    cur_len = torch.ones(T, dtype=torch.long)
     fi = torch.zeros(T, F, dtype=torch.float64)
-
-    def alloc_nodes(t: int, k: int) -> int:
-        start = len(feats[t])
-        feats[t].extend([-1] * k)
-        thrs[t].extend([0.0] * k)
-        lefts[t].extend([-1] * k)
-        leaves[t].extend([None] * k)
-        return start
 
     # root totals: one fused reduction (+ one all-reduce) for all trees
     if root_rows is None:
@@ -625,54 +621,68 @@ def grow_forest(
         comm.all_reduce_(root_tot)
     totals = root_tot.cpu()
 
-    # active level state (across all trees)
-    node_ids: List[Tuple[int, int]] = [(t, alloc_nodes(t, 1)) for t in range(T)]
+    # active level state (host tensors; sorted by tree by construction)
+    node_tree = torch.arange(T, dtype=torch.long)
+    node_nid = torch.zeros(T, dtype=torch.long)
     off_acc = [0]
     for v in root_lens:
         off_acc.append(off_acc[-1] + v)
     offsets = torch.tensor(off_acc, dtype=torch.int64)
     hists: Optional[torch.Tensor] = None
-    parent_of: List[int] = []
-    built_mask: List[bool] = []
+    parent_of_t: Optional[torch.Tensor] = None  # cpu long, per active idx
+    built_mask_t: Optional[torch.Tensor] = None  # cpu bool
 
     edges_cpu = edges.cpu()
     idx_c = C - 1
 
-    pred_rows: List[List[torch.Tensor]] = [[] for _ in range(T)]
-    pred_vals: List[List[float]] = [[] for _ in range(T)]
     capture = train_pred_out is not None
+    cap_rows: List[torch.Tensor] = []
+    cap_trees: List[torch.Tensor] = []
+    cap_vals: List[torch.Tensor] = []
 
-    def _capture_leaves(act_idx, offs, ridx):
-        if not capture:
+    def _ragged_positions(starts, lens):
+        """Device index positions for the concatenation of
+        [starts_i, starts_i + lens_i) ranges (cpu long inputs)."""
+        total = int(lens.sum())
+        if total == 0:
+            return torch.empty(0, dtype=torch.long, device=device)
+        base = torch.repeat_interleave(starts, lens)
+        firsts = torch.cumsum(lens, 0) - lens
+        within = torch.arange(total, dtype=torch.long) - \
+            torch.repeat_interleave(firsts, lens)
+        return (base + within).to(device)
+
+    def _finalize_and_capture(idx_t, offs, ridx):
+        if idx_t.numel() == 0:
             return
-        ol = offs.tolist() if isinstance(offs, torch.Tensor) else offs
-        for i in act_idx:
-            s0, e0 = int(ol[i]), int(ol[i + 1])
-            if e0 > s0:
-                t, nid = node_ids[i]
-                pred_rows[t].append(ridx[s0:e0])
-                pred_vals[t].append(float(leaves[t][nid]))
-
-    def _finalize(act_idx, tots):
-        tt = tots.cpu() if isinstance(tots, torch.Tensor) else tots
-        for j, i in enumerate(act_idx):
-            t, nid = node_ids[i]
-            g = float(tt[j, 0])
-            h = float(tt[j, 1])
-            leaves[t][nid] = (
-                g / (h + params.lam) if h + params.lam > 0 else 0.0
-            )
-
-    def _col0():
-        return torch.tensor([t * C for (t, _) in node_ids], dtype=torch.int32)
+        t_ids = node_tree[idx_t]
+        nids = node_nid[idx_t]
+        g = totals[idx_t, 0]
+        h = totals[idx_t, 1]
+        denom = h + params.lam
+        vals = torch.where(denom > 0, g / denom, torch.zeros_like(g))
+        leaves_t[t_ids, nids] = vals
+        if capture:
+            starts = offs[idx_t]
+            lens = offs[idx_t + 1] - starts
+            nz = (lens > 0).nonzero(as_tuple=True)[0]
+            if nz.numel():
+                pos = _ragged_positions(starts[nz], lens[nz])
+                lens_d = lens[nz].to(device)
+                cap_rows.append(ridx[pos].long())
+                cap_trees.append(
+                    torch.repeat_interleave(t_ids[nz].to(device), lens_d)
+                )
+                cap_vals.append(
+                    torch.repeat_interleave(vals[nz].to(device), lens_d)
+                )
 
     for depth in range(params.max_depth + 1):
-        n_active = len(node_ids)
+        n_active = int(node_tree.numel())
         if n_active == 0:
             break
         if depth == params.max_depth:
-            _finalize(range(n_active), totals)
-            _capture_leaves(range(n_active), offsets, row_idx)
+            _finalize_and_capture(torch.arange(n_active), offsets, row_idx)
             break
 
         split_args = dict(
@@ -684,45 +694,36 @@ def grow_forest(
         )
         lvl_mask = None
         if feature_masks is not None:
-            tid = torch.tensor([t for (t, _) in node_ids],
-                               dtype=torch.long, device=device)
-            lvl_mask = feature_masks.index_select(0, tid)
+            lvl_mask = feature_masks.index_select(0, node_tree.to(device))
         # ----- fused histograms for this level (+ pipelined reduce) ------
         if hists is None:
             new_h = ops.hist_build_forest(
-                bins, gh, row_idx, offsets, _col0(), B, C, gh_max,
+                bins, gh, row_idx, offsets, (node_tree * C).to(torch.int32),
+                B, C, gh_max,
             )
             hists, gain, feat, b, left_stats = _finish_level_split(
-                new_h, None, None, None, None, n_active, (F, B, C), device,
-                comm, split_args, split_mask=lvl_mask,
+                new_h, None, None, None, None, None, n_active, (F, B, C),
+                device, comm, split_args, split_mask=lvl_mask,
             )
         else:
-            built_idx = [j for j in range(n_active) if built_mask[j]]
+            built_cpu = built_mask_t.nonzero(as_tuple=True)[0]
+            nb_cpu = (~built_mask_t).nonzero(as_tuple=True)[0]
             bh = None
-            if built_idx:
-                off_list = offsets.tolist()
-                b_off = [0]
-                segs = []
-                for j in built_idx:
-                    s, e = off_list[j], off_list[j + 1]
-                    segs.append((s, e))
-                    b_off.append(b_off[-1] + (e - s))
-                build_rows = (
-                    torch.cat([row_idx[s:e] for s, e in segs])
-                    if len(segs) > 1
-                    else row_idx[segs[0][0]: segs[0][1]]
-                )
-                col0_b = torch.tensor(
-                    [node_ids[j][0] * C for j in built_idx], dtype=torch.int32
-                )
+            if built_cpu.numel():
+                starts = offsets[built_cpu]
+                lens = offsets[built_cpu + 1] - starts
+                b_off = torch.zeros(built_cpu.numel() + 1, dtype=torch.int64)
+                b_off[1:] = torch.cumsum(lens, 0)
+                build_rows = row_idx[_ragged_positions(starts, lens)]
+                col0_b = (node_tree[built_cpu] * C).to(torch.int32)
                 bh = ops.hist_build_forest(
-                    bins, gh, build_rows,
-                    torch.tensor(b_off, dtype=torch.int64), col0_b, B, C,
-                    gh_max,
+                    bins, gh, build_rows, b_off, col0_b, B, C, gh_max,
                 )
             hists, gain, feat, b, left_stats = _finish_level_split(
-                bh, built_idx, hists, parent_of, built_mask, n_active,
-                (F, B, C), device, comm, split_args, split_mask=lvl_mask,
+                bh, built_cpu.to(device), hists, nb_cpu.to(device),
+                (nb_cpu ^ 1).to(device),
+                parent_of_t[nb_cpu].to(device), n_active, (F, B, C), device,
+                comm, split_args, split_mask=lvl_mask,
             )
         pr_async = None
         if bins.is_cuda:
@@ -735,27 +736,27 @@ def grow_forest(
 
         do_split = torch.isfinite(gain_cpu)
         ns_idx = (~do_split).nonzero(as_tuple=True)[0]
-        if ns_idx.numel():
-            _finalize(ns_idx.tolist(), totals[ns_idx])
-            _capture_leaves(ns_idx.tolist(), offsets, row_idx)
+        _finalize_and_capture(ns_idx, offsets, row_idx)
         if not bool(do_split.any()):
             break
 
         split_feat = torch.where(do_split, feat_cpu, torch.full_like(feat_cpu, -1))
-        child_ids: List[Optional[Tuple[int, int]]] = []
-        for i in range(n_active):
-            if bool(do_split[i]):
-                t, nid = node_ids[i]
-                f = int(feat_cpu[i])
-                tb = int(b_cpu[i])
-                feats[t][nid] = f
-                fi[t, f] += float(gain_cpu[i])
-                thrs[t][nid] = float(edges_cpu[f, tb])
-                cid = alloc_nodes(t, 2)
-                lefts[t][nid] = cid
-                child_ids.append((cid, cid + 1))
-            else:
-                child_ids.append(None)
+        s_idx = do_split.nonzero(as_tuple=True)[0]  # cpu long, sorted
+        t_ids = node_tree[s_idx]
+        nids = node_nid[s_idx]
+        f_l = feat_cpu[s_idx].long()
+        b_l = b_cpu[s_idx].long()
+        feats_t[t_ids, nids] = f_l.to(torch.int32)
+        thrs_t[t_ids, nids] = edges_cpu[f_l, b_l]
+        fi.index_put_((t_ids, f_l), gain_cpu[s_idx].double(), accumulate=True)
+        # child allocation: split nodes are grouped by tree, so the rank
+        # within the tree's segment gives each pair's base id
+        counts = torch.bincount(t_ids, minlength=T)
+        firsts = torch.cumsum(counts, 0) - counts
+        rank = torch.arange(s_idx.numel(), dtype=torch.long) - firsts[t_ids]
+        cid = cur_len[t_ids] + 2 * rank
+        lefts_t[t_ids, nids] = cid.to(torch.int32)
+        cur_len += 2 * counts
 
         if pr_async is not None:
             new_rows, new_offs, _ = ops.partition_rows_finish(*pr_async)
@@ -764,82 +765,54 @@ def grow_forest(
                 bins, row_idx, offsets,
                 split_feat.to(torch.int32), b_cpu.to(torch.int32),
             )
-        offs_list = new_offs.tolist()
 
-        next_nodes: List[Tuple[int, int]] = []
-        next_off: List[int] = [0]
-        next_tot_rows: List[torch.Tensor] = []
-        nb_mask: List[bool] = []
-        nb_parent: List[int] = []
-        keep_segs: List[Tuple[int, int]] = []
-        surviving = [i for i in range(n_active) if child_ids[i] is not None]
-        hist_keep = torch.tensor(surviving, dtype=torch.long, device=device)
-        for rank_i, i in enumerate(surviving):
-            t, _ = node_ids[i]
-            lcid, rcid = child_ids[i]
-            l_stats = left_stats[i]
-            r_stats = totals[i] - l_stats
-            l_cnt = float(l_stats[idx_c])
-            r_cnt = float(r_stats[idx_c])
-            ls, le, re = offs_list[2 * i], offs_list[2 * i + 1], offs_list[2 * i + 2]
-            keep_segs.append((ls, re))
-            for (cid, s, e, st, built) in (
-                (lcid, ls, le, l_stats, l_cnt <= r_cnt),
-                (rcid, le, re, r_stats, l_cnt > r_cnt),
-            ):
-                next_nodes.append((t, cid))
-                next_off.append(next_off[-1] + (e - s))
-                next_tot_rows.append(st)
-                nb_mask.append(built)
-                nb_parent.append(rank_i)
-
-        if len(keep_segs) == 1:
-            s, e = keep_segs[0]
-            row_idx = new_rows[s:e]
+        # ----- next level bookkeeping (all tensor ops) -------------------
+        l_st = left_stats[s_idx]                       # [n_split, C]
+        r_st = totals[s_idx] - l_st
+        totals = torch.stack([l_st, r_st], dim=1).reshape(-1, C)
+        l_cnt = l_st[:, idx_c]
+        r_cnt = r_st[:, idx_c]
+        built_mask_t = torch.stack([l_cnt <= r_cnt, l_cnt > r_cnt],
+                                   dim=1).reshape(-1)
+        parent_of_t = torch.arange(s_idx.numel(),
+                                   dtype=torch.long).repeat_interleave(2)
+        node_tree = t_ids.repeat_interleave(2)
+        node_nid = torch.stack([cid, cid + 1], dim=1).reshape(-1)
+        ls = new_offs[2 * s_idx]
+        le = new_offs[2 * s_idx + 1]
+        re = new_offs[2 * s_idx + 2]
+        lens_pairs = torch.stack([le - ls, re - le], dim=1).reshape(-1)
+        offsets = torch.zeros(lens_pairs.numel() + 1, dtype=torch.int64)
+        offsets[1:] = torch.cumsum(lens_pairs, 0)
+        if int(s_idx.numel()) == n_active:
+            row_idx = new_rows
         else:
-            row_idx = torch.cat([new_rows[s:e] for s, e in keep_segs])
-        offsets = torch.tensor(next_off, dtype=torch.int64)
-        node_ids = next_nodes
-        totals = torch.stack(next_tot_rows)
-        hists = hists.index_select(0, hist_keep)
-        parent_of = nb_parent
-        built_mask = nb_mask
+            row_idx = new_rows[_ragged_positions(ls, re - ls)]
+        hists = hists.index_select(0, s_idx.to(device))
 
     # assemble per-tree arrays
     trees: List[Dict[str, torch.Tensor]] = []
+    fi_tot = fi.sum(dim=1)
     for t in range(T):
-        n_nodes = len(feats[t])
-        leaf_value = torch.zeros(n_nodes, 1, dtype=torch.float32)
-        for i, lv in enumerate(leaves[t]):
-            if lv is not None:
-                leaf_value[i, 0] = lv
-        fi_tot = float(fi[t].sum())
+        n_nodes = int(cur_len[t])
         tree = {
-            "feature": torch.tensor(feats[t], dtype=torch.int32),
-            "threshold": torch.tensor(thrs[t], dtype=torch.float32),
-            "left_child": torch.tensor(lefts[t], dtype=torch.int32),
-            "leaf_value": leaf_value,
-            "feature_importance": (fi[t] / fi_tot if fi_tot > 0 else fi[t]).to(
-                torch.float32
-            ),
+            "feature": feats_t[t, :n_nodes].clone(),
+            "threshold": thrs_t[t, :n_nodes].clone(),
+            "left_child": lefts_t[t, :n_nodes].clone(),
+            "leaf_value": leaves_t[t, :n_nodes].clone().unsqueeze(1),
+            "feature_importance": (
+                fi[t] / fi_tot[t] if float(fi_tot[t]) > 0 else fi[t]
+            ).to(torch.float32),
         }
         trees.append({k: _to_dev_async(v, device) for k, v in tree.items()})
 
     if capture:
         tp = torch.zeros(N, T, dtype=torch.float32, device=device)
-        for t in range(T):
-            if not pred_rows[t]:
-                continue
-            rows_cat = torch.cat(pred_rows[t]).long()
-            counts = _to_dev_async(
-                torch.tensor([r.numel() for r in pred_rows[t]]), device
+        if cap_rows:
+            tp.index_put_(
+                (torch.cat(cap_rows), torch.cat(cap_trees)),
+                torch.cat(cap_vals),
             )
-            vals = torch.repeat_interleave(
-                _to_dev_async(torch.tensor(pred_vals[t], dtype=torch.float32),
-                              device),
-                counts,
-            )
-            tp[rows_cat, t] = vals
         train_pred_out.append(tp)
     return trees
 
