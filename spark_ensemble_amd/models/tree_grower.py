@@ -642,15 +642,20 @@ def grow_forest(
 
     def _ragged_positions(starts, lens):
         """Device index positions for the concatenation of
-        [starts_i, starts_i + lens_i) ranges (cpu long inputs)."""
+        [starts_i, starts_i + lens_i) ranges (cpu long inputs).  Only the
+        SMALL per-node starts/lens tensors cross to the device; the
+        [total]-sized expansion happens there (a cpu-side expansion
+        pageable-copied per level measured 16-60x slower end to end)."""
         total = int(lens.sum())
         if total == 0:
             return torch.empty(0, dtype=torch.long, device=device)
-        base = torch.repeat_interleave(starts, lens)
-        firsts = torch.cumsum(lens, 0) - lens
-        within = torch.arange(total, dtype=torch.long) - \
-            torch.repeat_interleave(firsts, lens)
-        return (base + within).to(device)
+        starts_d = _to_dev_async(starts, device)
+        lens_d = _to_dev_async(lens, device)
+        base = torch.repeat_interleave(starts_d, lens_d)
+        firsts = torch.cumsum(lens_d, 0) - lens_d
+        within = torch.arange(total, dtype=torch.long, device=device) - \
+            torch.repeat_interleave(firsts, lens_d)
+        return base + within
 
     def _finalize_and_capture(idx_t, offs, ridx):
         if idx_t.numel() == 0:
@@ -694,7 +699,9 @@ def grow_forest(
         )
         lvl_mask = None
         if feature_masks is not None:
-            lvl_mask = feature_masks.index_select(0, node_tree.to(device))
+            lvl_mask = feature_masks.index_select(
+                0, _to_dev_async(node_tree, device)
+            )
         # ----- fused histograms for this level (+ pipelined reduce) ------
         if hists is None:
             new_h = ops.hist_build_forest(
@@ -720,10 +727,11 @@ def grow_forest(
                     bins, gh, build_rows, b_off, col0_b, B, C, gh_max,
                 )
             hists, gain, feat, b, left_stats = _finish_level_split(
-                bh, built_cpu.to(device), hists, nb_cpu.to(device),
-                (nb_cpu ^ 1).to(device),
-                parent_of_t[nb_cpu].to(device), n_active, (F, B, C), device,
-                comm, split_args, split_mask=lvl_mask,
+                bh, _to_dev_async(built_cpu, device), hists,
+                _to_dev_async(nb_cpu, device),
+                _to_dev_async(nb_cpu ^ 1, device),
+                _to_dev_async(parent_of_t[nb_cpu], device), n_active,
+                (F, B, C), device, comm, split_args, split_mask=lvl_mask,
             )
         pr_async = None
         if bins.is_cuda:
@@ -788,7 +796,7 @@ def grow_forest(
             row_idx = new_rows
         else:
             row_idx = new_rows[_ragged_positions(ls, re - ls)]
-        hists = hists.index_select(0, s_idx.to(device))
+        hists = hists.index_select(0, _to_dev_async(s_idx, device))
 
     # assemble per-tree arrays
     trees: List[Dict[str, torch.Tensor]] = []
