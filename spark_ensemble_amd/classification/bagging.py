@@ -75,6 +75,49 @@ class BaggingClassifier(ProbabilisticClassifier, _BaggingClassifierParams):
         subspaces = [
             subspace(self.getSubspaceRatio(), num_features, seed + i) for i in range(k)
         ]
+
+        from ..models.tree import DecisionTreeClassifier, fit_class_tree_forest
+
+        if (
+            type(learner) is DecisionTreeClassifier
+            and learner.getOrDefault("minWeightFractionPerNode") == 0.0
+            and num_classes + 2 <= 8
+        ):
+            # fused path: all k gini trees (K one-hot channels each) grow
+            # level-synchronously; subspaces as split masks, bags as root
+            # row sets — same machinery as the regression fusion
+            f_edges, f_bins = binned.get(int(learner.getOrDefault("maxBins")))
+            bag_w = torch.stack([
+                self.sample_weights(
+                    self.getReplacement(), self.getSubsampleRatio(), n,
+                    seed + i, x.device, w, comm.rank,
+                )
+                for i in range(k)
+            ], dim=1)
+            root_rows = [
+                (bag_w[:, i] > 0).nonzero(as_tuple=True)[0].to(torch.int32)
+                for i in range(k)
+            ]
+            onehot = torch.zeros(n, num_classes, dtype=torch.float32,
+                                 device=x.device)
+            onehot.scatter_(1, y.long().unsqueeze(1), 1.0)
+            models = fit_class_tree_forest(
+                learner, f_edges, f_bins, onehot, bag_w, num_classes, comm,
+                subspaces=subspaces, root_rows=root_rows,
+            )
+            model = BaggingClassificationModel()
+            model._models = models
+            model._subspaces = subspaces
+            model._num_classes = num_classes
+            model._num_features = num_features
+            model.set("votingStrategy", self.getVotingStrategy())
+            for p in (
+                "featuresCol", "labelCol", "predictionCol",
+                "rawPredictionCol", "probabilityCol",
+            ):
+                model.set(p, self.getOrDefault(p))
+            return model
+
         if learner is not None and learner.hasParam("maxBins"):
             binned.get(int(learner.getOrDefault("maxBins")))  # pre-warm once
 

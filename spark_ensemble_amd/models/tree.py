@@ -345,6 +345,66 @@ def fit_tree_forest(learner, edges, bins, labels, weights, comm=None,
     return models, tp_out[0]
 
 
+def fit_class_tree_forest(learner, edges, bins, onehot, weights, k_classes,
+                          comm=None, subspaces=None, root_rows=None):
+    """T independent ``DecisionTreeClassifier`` fits (gini trees, D = K
+    one-hot channels) as fused forest grows.  ``onehot`` [N, K];
+    ``weights`` [N, T] per-tree bag weights (positive on each tree's
+    ``root_rows``).  K + 2 channels per tree must fit the kernel's
+    8-channel cap (K <= 6 with non-unit weights).  Members are chunked so
+    the [N, chunk, K] gradient tensor stays modest."""
+    from ..parallel import get_comm as _gc
+    from .tree_grower import grow_forest
+
+    comm = comm or _gc()
+    F = bins.shape[1]
+    N, K = onehot.shape
+    T = weights.shape[1]
+    gp = learner._grow_params(1.0)
+    stats = torch.stack([weights.abs().max()])
+    if comm.is_distributed:
+        comm.all_reduce_(stats, "max")
+    hmax = float(stats.cpu()[0])
+    gh_max = torch.tensor([max(hmax, 1.0)] * K + [hmax, 1.0])
+
+    models = []
+    chunk = max(2, 16 // max(K // 3, 1))
+    for s in range(0, T, chunk):
+        sl = slice(s, min(s + chunk, T))
+        w_sl = weights[:, sl]
+        grads = (onehot.unsqueeze(1) * w_sl.unsqueeze(2)).contiguous()
+        fmasks = None
+        subs_sl = subspaces[sl] if subspaces is not None else None
+        if subs_sl is not None and any(x.numel() != F for x in subs_sl):
+            fmasks = torch.zeros(len(subs_sl), F, dtype=torch.float32,
+                                 device=bins.device)
+            for t, sub in enumerate(subs_sl):
+                fmasks[t, sub.to(bins.device)] = 1.0
+        trees = grow_forest(
+            bins, edges, grads, w_sl.contiguous(), gp, comm,
+            hess_is_count=False, gh_max_in=gh_max,
+            root_rows=root_rows[sl] if root_rows is not None else None,
+            feature_masks=fmasks,
+        )
+        for t, tree in enumerate(trees):
+            nf = F
+            if fmasks is not None:
+                sub = subs_sl[t]
+                nf = int(sub.numel())
+                feat = tree["feature"].long()
+                sub_dev = sub.to(device=feat.device, dtype=torch.long)
+                local = torch.searchsorted(sub_dev, feat.clamp_min(0))
+                tree = dict(tree, feature=torch.where(
+                    feat >= 0, local, feat
+                ).to(torch.int32))
+            m = DecisionTreeClassificationModel()
+            m._set_tree(tree, nf)
+            m._num_classes = k_classes
+            m._copy_cols_from(learner)
+            models.append(m)
+    return models
+
+
 class DecisionTreeClassifier(ProbabilisticClassifier, _TreeParams):
     def _fit(self, dataset: TensorFrame) -> "DecisionTreeClassificationModel":
         x, y, w = self._extract_xyw(dataset)

@@ -533,13 +533,18 @@ def grow_forest(
     device = bins.device
     N, F = bins.shape
     T = grads.shape[1]
+    # multi-output trees (gini classification: D = num classes) carry D
+    # gradient channels per tree; grads is then [N, T, D]
+    D = grads.shape[2] if grads.dim() == 3 else 1
     B = params.max_bins
 
-    if T > MAX_FUSED_TREES:
+    # bound the fused batch so the interleaved gh matrix stays modest
+    batch_cap = MAX_FUSED_TREES if D == 1 else max(2, (MAX_FUSED_TREES * 3) // (D + 2))
+    if T > batch_cap:
         out: List[Dict[str, torch.Tensor]] = []
         preds = [] if train_pred_out is not None else None
-        for s in range(0, T, MAX_FUSED_TREES):
-            sl = slice(s, min(s + MAX_FUSED_TREES, T))
+        for s in range(0, T, batch_cap):
+            sl = slice(s, min(s + batch_cap, T))
             h_sl = hess if hess.dim() == 1 else hess[:, sl].contiguous()
             sub_pred = [] if preds is not None else None
             out.extend(grow_forest(
@@ -557,17 +562,21 @@ def grow_forest(
     h_shared = hess.dim() == 1
     if hess_is_count is None:
         hess_is_count = bool((hess == 1).all())
-    C = 2 if hess_is_count else 3
-    parts = [grads, hess.unsqueeze(1).expand(N, T) if h_shared else hess]
-    if C == 3:
-        parts.append(torch.ones(N, T, dtype=torch.float32, device=device))
-    gh = torch.stack(parts, dim=2).reshape(N, T * C).contiguous()
+    NN = 1 if hess_is_count else 2
+    C = D + NN  # channels per tree
+    assert C <= 8, f"fused forest: D + hess/count channels = {C} > 8"
+    g3 = grads if grads.dim() == 3 else grads.unsqueeze(2)
+    parts = [g3, (hess.unsqueeze(1).expand(N, T) if h_shared else hess)
+             .unsqueeze(2)]
+    if NN == 2:
+        parts.append(torch.ones(N, T, 1, dtype=torch.float32, device=device))
+    gh = torch.cat(parts, dim=2).reshape(N, T * C).contiguous()
 
     if gh_max_in is not None:
         gh_max = gh_max_in  # slot-wise [C] host tensor
     elif bins.is_cuda:
-        gm = [float(grads.abs().max()), float(hess.max())]
-        if C == 3:
+        gm = [float(g3.abs().max())] * D + [float(hess.max())]
+        if NN == 2:
             gm.append(1.0)
         gh_max = torch.tensor(gm)
     else:
@@ -592,13 +601,13 @@ def grow_forest(
     feats_t = torch.full((T, max_nodes), -1, dtype=torch.int32)
     thrs_t = torch.zeros(T, max_nodes, dtype=torch.float32)
     lefts_t = torch.full((T, max_nodes), -1, dtype=torch.int32)
-    leaves_t = torch.zeros(T, max_nodes, dtype=torch.float32)
+    leaves_t = torch.zeros(T, max_nodes, D, dtype=torch.float32)
     cur_len = torch.ones(T, dtype=torch.long)
     fi = torch.zeros(T, F, dtype=torch.float64)
 
     # root totals: one fused reduction (+ one all-reduce) for all trees
     if root_rows is None:
-        g_sum = grads.sum(dim=0)  # [T]
+        g_sum = g3.sum(dim=0)  # [T, D]
         if h_shared:
             h_sum = hess.sum().reshape(1).expand(T)
         else:
@@ -608,15 +617,15 @@ def grow_forest(
         gs, hs = [], []
         for t in range(T):
             r = root_rows[t].long()
-            gs.append(grads[r, t].sum())
+            gs.append(g3[r, t].sum(dim=0))
             hs.append((hess if h_shared else hess[:, t])[r].sum())
-        g_sum = torch.stack(gs)
+        g_sum = torch.stack(gs)  # [T, D]
         h_sum = torch.stack(hs)
         cnt_col = torch.tensor([float(v) for v in root_lens], device=device)
-    cols = [g_sum, h_sum]
-    if C == 3:
-        cols.append(cnt_col)
-    root_tot = torch.stack(cols, dim=1)  # [T, C] device
+    cols = [g_sum, h_sum.unsqueeze(1)]
+    if NN == 2:
+        cols.append(cnt_col.unsqueeze(1))
+    root_tot = torch.cat(cols, dim=1)  # [T, C] device
     if comm is not None:
         comm.all_reduce_(root_tot)
     totals = root_tot.cpu()
@@ -635,7 +644,7 @@ def grow_forest(
     edges_cpu = edges.cpu()
     idx_c = C - 1
 
-    capture = train_pred_out is not None
+    capture = train_pred_out is not None and D == 1
     cap_rows: List[torch.Tensor] = []
     cap_trees: List[torch.Tensor] = []
     cap_vals: List[torch.Tensor] = []
@@ -662,11 +671,12 @@ def grow_forest(
             return
         t_ids = node_tree[idx_t]
         nids = node_nid[idx_t]
-        g = totals[idx_t, 0]
-        h = totals[idx_t, 1]
-        denom = h + params.lam
+        g = totals[idx_t, :D]               # [n, D]
+        h = totals[idx_t, D]                # [n]
+        denom = (h + params.lam).unsqueeze(1)
         vals = torch.where(denom > 0, g / denom, torch.zeros_like(g))
         leaves_t[t_ids, nids] = vals
+        vals = vals[:, 0]  # capture is D == 1 only
         if capture:
             starts = offs[idx_t]
             lens = offs[idx_t + 1] - starts
@@ -695,7 +705,7 @@ def grow_forest(
             min_child_weight=params.min_child_weight,
             min_instances=params.min_instances_per_node,
             min_info_gain=params.min_info_gain,
-            d_dims=1,
+            d_dims=D,
         )
         lvl_mask = None
         if feature_masks is not None:
@@ -807,7 +817,7 @@ def grow_forest(
             "feature": feats_t[t, :n_nodes].clone(),
             "threshold": thrs_t[t, :n_nodes].clone(),
             "left_child": lefts_t[t, :n_nodes].clone(),
-            "leaf_value": leaves_t[t, :n_nodes].clone().unsqueeze(1),
+            "leaf_value": leaves_t[t, :n_nodes].clone().reshape(n_nodes, D),
             "feature_importance": (
                 fi[t] / fi_tot[t] if float(fi_tot[t]) > 0 else fi[t]
             ).to(torch.float32),

@@ -216,6 +216,41 @@ def test_fused_bagging_matches_sequential():
         assert m._num_features == int(sub.numel())
 
 
+def test_fused_classification_bagging_matches_sequential():
+    """BaggingClassifier's fused gini-forest path (multi-output trees,
+    D = K channels) must reproduce sequential member fits."""
+    import spark_ensemble_amd as sea
+    from spark_ensemble_amd.models import tree as tree_mod
+    from spark_ensemble_amd.utils.io import synthetic_classification
+
+    df = synthetic_classification(12000, 14, k=3, seed=19)
+
+    def mk():
+        return (
+            sea.BaggingClassifier()
+            .setNumBaseLearners(5)
+            .setSubspaceRatio(0.6)
+            .setSubsampleRatio(0.7)
+            .setReplacement(True)
+            .setVotingStrategy("soft")
+            .setSeed(3)
+        )
+
+    m_fused = mk().fit(df)
+
+    class _DT(tree_mod.DecisionTreeClassifier):
+        pass  # type check fails -> sequential path
+
+    m_seq = mk().setBaseLearner(_DT()).fit(df)
+    a = m_fused.transform(df)["probability"]
+    b = m_seq.transform(df)["probability"]
+    assert torch.allclose(a, b, rtol=1e-4, atol=1e-5), \
+        float((a - b).abs().max())
+    for m, sub in zip(m_fused._models, m_fused._subspaces):
+        assert int(m._tree["feature"].max()) < int(sub.numel())
+        assert m._num_classes == 3
+
+
 def test_bagging_fit_folds_matches_sequential():
     """BaggingRegressor._fit_folds (all fold x member trees in one
     forest) must reproduce per-fold weight-masked fits."""
