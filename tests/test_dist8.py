@@ -35,15 +35,25 @@ def _env(tmpdir, trace_name=None, port="29881"):
 
 
 def _torchrun(args, env, nproc, port):
-    cmd = [
-        sys.executable, "-m", "torch.distributed.run",
-        "--nnodes=1", f"--nproc-per-node={nproc}",
-        "--master-addr", "127.0.0.1", "--master-port", port,
-    ] + args
-    r = subprocess.run(cmd, env=env, capture_output=True, text=True,
-                       cwd=REPO, timeout=900)
-    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
-    return r
+    import random
+
+    last = None
+    for attempt in range(2):
+        # random port per attempt: back-to-back rendezvous on a fixed
+        # port can collide with a TIME_WAIT listener from another test
+        prt = str(random.randint(29500, 29989)) if attempt else port
+        cmd = [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", f"--nproc-per-node={nproc}",
+            "--master-addr", "127.0.0.1", "--master-port", prt,
+        ] + args
+        r = subprocess.run(cmd, env=env, capture_output=True, text=True,
+                           cwd=REPO, timeout=900)
+        if r.returncode == 0:
+            return r
+        last = r
+    assert last.returncode == 0, last.stdout[-2000:] + last.stderr[-2000:]
+    return last
 
 
 def test_bench_fake_8_world_collective_order(tmp_path):
