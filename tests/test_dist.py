@@ -94,13 +94,22 @@ def _run_world(tmpdir, nproc=2):
     env["SEA_OUT"] = outfile
     env["MASTER_ADDR"] = "127.0.0.1"
     env.pop("HIP_VISIBLE_DEVICES", None)
-    cmd = [
-        sys.executable, "-m", "torch.distributed.run",
-        "--nnodes=1", f"--nproc-per-node={nproc}",
-        "--master-addr", "127.0.0.1", "--master-port", "29871",
-        "--no-python" if False else script,
-    ]
-    r = subprocess.run(cmd, env=env, capture_output=True, text=True, timeout=600)
+    import random
+
+    last = None
+    for attempt in range(2):
+        prt = "29871" if attempt == 0 else str(random.randint(29500, 29989))
+        cmd = [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", f"--nproc-per-node={nproc}",
+            "--master-addr", "127.0.0.1", "--master-port", prt,
+            script,
+        ]
+        r = subprocess.run(cmd, env=env, capture_output=True, text=True,
+                           timeout=600)
+        if r.returncode == 0:
+            break
+        last = r
     assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
     with open(outfile) as f:
         return json.load(f)
