@@ -716,7 +716,7 @@ def grow_forest(
         if hists is None:
             new_h = ops.hist_build_forest(
                 bins, gh, row_idx, offsets, (node_tree * C).to(torch.int32),
-                B, C, gh_max,
+                B, C, gh_max, d_dims=D,
             )
             hists, gain, feat, b, left_stats = _finish_level_split(
                 new_h, None, None, None, None, None, n_active, (F, B, C),
@@ -735,6 +735,7 @@ def grow_forest(
                 col0_b = (node_tree[built_cpu] * C).to(torch.int32)
                 bh = ops.hist_build_forest(
                     bins, gh, build_rows, b_off, col0_b, B, C, gh_max,
+                    d_dims=D,
                 )
             hists, gain, feat, b, left_stats = _finish_level_split(
                 bh, _to_dev_async(built_cpu, device), hists,
