@@ -43,7 +43,7 @@ inline int64_t lceil_div(int64_t a, int64_t b) { return (a + b - 1) / b; }
 // (and its NC*4*K ds_read_b32 per row per lane) is replaced by NC*4*K
 // registers loaded once (measured: removes the per-row LDS issue traffic
 // that capped the kernel at ~3 TB/s of the ~6.3 TB/s roofline).
-template <int NC, int K, bool WREG>
+template <int NC, int K, bool WREG, int RR = 2>
 __global__ __launch_bounds__(256) void logreg_loss_grad_kernel(
     float* __restrict__ payload,     // [1 + (F+1)*K]: loss, grad[F][K], gbias[K]
     const float* __restrict__ x,     // [N, F]
@@ -93,12 +93,13 @@ __global__ __launch_bounds__(256) void logreg_loss_grad_kernel(
   const int64_t wave_id = (int64_t)blockIdx.x * (blockDim.x >> 6) + wave;
   const int64_t n_waves = (int64_t)gridDim.x * (blockDim.x >> 6);
 
-  // two rows per iteration: the 6-step shfl_xor margin reductions of the
-  // two rows interleave (independent chains), halving exposed shuffle
-  // latency per row
-  constexpr int RR = 2;
+  // RR rows per iteration: the shfl_xor margin-reduction chains of the
+  // rows interleave (independent), and — the round-2 finding — the
+  // kernel is INFLIGHT-BYTES-limited (~128 B/wave at RR=2 ≈ the measured
+  // 3 TB/s), so RR=4 for the small-K shapes doubles the outstanding
+  // loads per wave at an acceptable register cost
   for (int64_t r0 = wave_id * RR; r0 < n; r0 += n_waves * RR) {
-    const int nr = (r0 + 1 < n) ? RR : 1;
+    const int nr = (int)std::min<int64_t>(RR, n - r0);
     float xf[RR][NC][4];
     float acc[RR][K];
 #pragma unroll
@@ -218,20 +219,37 @@ void launch_logreg(float* payload, const float* x, const int* y,
   // measured (gpurun_out/r02_logreg.json): W-in-LDS 3.1 TB/s vs
   // W-in-registers 2.86 TB/s at 5M x 1024 K=2 — the LDS reads hide
   // behind the x-load latency and the extra registers cost occupancy,
-  // so LDS stays the default; SEA_LOGREG_WREG=1 flips for probing
+  // so LDS stays the default; SEA_LOGREG_WREG=1 flips for probing,
+  // SEA_LOGREG_RR2=1 forces the narrow 2-row pipeline
   static const bool use_lds = []() {
     const char* e = getenv("SEA_LOGREG_WREG");
     return !(e && e[0] == '1');
   }();
+  static const bool force_rr2 = []() {
+    const char* e = getenv("SEA_LOGREG_RR2");
+    return e && e[0] == '1';
+  }();
+  const size_t lds = use_lds ? (size_t)F * K * 4 : 0;
+  constexpr bool rr4_ok = (K <= 2 && NC <= 4);
+  if (rr4_ok && !force_rr2) {
+    if (use_lds)
+      hipLaunchKernelGGL((logreg_loss_grad_kernel<NC, K, false, 4>),
+                         dim3(blocks), dim3(256), lds, stream, payload, x, y,
+                         w, wmat, n, F, has_bias);
+    else
+      hipLaunchKernelGGL((logreg_loss_grad_kernel<NC, K, true, 4>),
+                         dim3(blocks), dim3(256), 0, stream, payload, x, y,
+                         w, wmat, n, F, has_bias);
+    return;
+  }
   if (use_lds) {
-    const size_t lds = (size_t)F * K * 4;
-    hipLaunchKernelGGL((logreg_loss_grad_kernel<NC, K, false>), dim3(blocks),
-                       dim3(256), lds, stream, payload, x, y, w, wmat, n, F,
-                       has_bias);
+    hipLaunchKernelGGL((logreg_loss_grad_kernel<NC, K, false, 2>),
+                       dim3(blocks), dim3(256), lds, stream, payload, x, y,
+                       w, wmat, n, F, has_bias);
   } else {
-    hipLaunchKernelGGL((logreg_loss_grad_kernel<NC, K, true>), dim3(blocks),
-                       dim3(256), 0, stream, payload, x, y, w, wmat, n, F,
-                       has_bias);
+    hipLaunchKernelGGL((logreg_loss_grad_kernel<NC, K, true, 2>),
+                       dim3(blocks), dim3(256), 0, stream, payload, x, y, w,
+                       wmat, n, F, has_bias);
   }
 }
 
