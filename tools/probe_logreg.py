@@ -11,9 +11,10 @@ import time
 import torch
 
 
-def run_arm(lds: bool):
+def run_arm(name, envvars):
     env = dict(os.environ)
-    env["SEA_LOGREG_WREG"] = "0" if lds else "1"
+    env.update(envvars)
+    env["SEA_ARM"] = name
     r = subprocess.run([sys.executable, __file__, "arm"], env=env,
                        capture_output=True, text=True, cwd="/root/repo")
     assert r.returncode == 0, r.stderr[-1500:]
@@ -47,7 +48,7 @@ def arm():
     dt = (time.time() - t0) / reps
     bytes_x = n * F * 4
     print(json.dumps({
-        "mode": "wreg" if os.environ.get("SEA_LOGREG_WREG") == "1" else "lds",
+        "mode": os.environ.get("SEA_ARM", "?"),
         "ms": round(dt * 1000, 2),
         "tb_per_s": round(bytes_x / dt / 1e12, 2),
         "loss": float(p[0]),
@@ -59,7 +60,15 @@ if __name__ == "__main__":
     if len(sys.argv) > 1 and sys.argv[1] == "arm":
         arm()
     else:
-        a = run_arm(False)
-        b = run_arm(True)
-        same = abs(a["loss"] - b["loss"]) / max(abs(b["loss"]), 1e-9) < 1e-5
-        print(json.dumps({"wreg": a, "lds": b, "loss_match": bool(same)}))
+        arms = {
+            "rr4_lds": {},
+            "rr2_lds": {"SEA_LOGREG_RR2": "1"},
+            "rr4_wreg": {"SEA_LOGREG_WREG": "1"},
+        }
+        out = {k: run_arm(k, v) for k, v in arms.items()}
+        ref = out["rr2_lds"]["loss"]
+        out["loss_match"] = all(
+            abs(v["loss"] - ref) / max(abs(ref), 1e-9) < 1e-5
+            for k, v in out.items() if isinstance(v, dict)
+        )
+        print(json.dumps(out))
