@@ -225,13 +225,16 @@ void launch_logreg(float* payload, const float* x, const int* y,
     const char* e = getenv("SEA_LOGREG_WREG");
     return !(e && e[0] == '1');
   }();
-  static const bool force_rr2 = []() {
-    const char* e = getenv("SEA_LOGREG_RR2");
+  // RR=4 measured SLOWER (2.21 vs 2.98 TB/s): the xf registers cost a
+  // wave per SIMD, shrinking total inflight bytes per CU below the RR=2
+  // level — opt-in only, kept for probing on future parts
+  static const bool force_rr4 = []() {
+    const char* e = getenv("SEA_LOGREG_RR4");
     return e && e[0] == '1';
   }();
   const size_t lds = use_lds ? (size_t)F * K * 4 : 0;
   constexpr bool rr4_ok = (K <= 2 && NC <= 4);
-  if (rr4_ok && !force_rr2) {
+  if (rr4_ok && force_rr4) {
     if (use_lds)
       hipLaunchKernelGGL((logreg_loss_grad_kernel<NC, K, false, 4>),
                          dim3(blocks), dim3(256), lds, stream, payload, x, y,

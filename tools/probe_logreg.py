@@ -61,9 +61,9 @@ if __name__ == "__main__":
         arm()
     else:
         arms = {
-            "rr4_lds": {},
-            "rr2_lds": {"SEA_LOGREG_RR2": "1"},
-            "rr4_wreg": {"SEA_LOGREG_WREG": "1"},
+            "rr2_lds": {},
+            "rr4_lds": {"SEA_LOGREG_RR4": "1"},
+            "rr2_wreg": {"SEA_LOGREG_WREG": "1"},
         }
         out = {k: run_arm(k, v) for k, v in arms.items()}
         ref = out["rr2_lds"]["loss"]
