@@ -351,6 +351,168 @@ class GBMClassifier(ProbabilisticClassifier, _GBMClassifierParams):
             model.set(p, self.getOrDefault(p))
         return model
 
+    # ---- fold-vectorized fitting (OOF stacking fast path) ----------------
+    def _can_fit_folds(self) -> bool:
+        """Covers the default classification configuration: gradient
+        updates, prior/uniform init, full bags, identity subspace,
+        built-in tree learner, no validation/checkpointing."""
+        from ..models.tree import DecisionTreeRegressor
+
+        learner = self.getOrNone("baseLearner") or self._default_base_learner()
+        return (
+            self.getOrDefault("updates") == "gradient"
+            and self.getSubspaceRatio() >= 1.0
+            and self.getSubsampleRatio() >= 1.0
+            and not self.getReplacement()
+            and self.getOrNone("validationIndicatorCol") is None
+            and not self.getCheckpointDir()
+            and type(learner) is DecisionTreeRegressor
+            and learner.getOrDefault("minWeightFractionPerNode") == 0.0
+        )
+
+    def _fit_folds(self, dataset: TensorFrame, fold: torch.Tensor,
+                   num_folds: int, include_full: bool = False):
+        """num_folds leave-one-fold-out GBM classifier fits grown JOINTLY:
+        every boosting round fuses ALL (fold, class) trees into one
+        forest (per-tree root row sets = the fold's rows); the per-fold
+        stage-weight searches run on fold-masked weights.  The init is
+        the GLOBAL prior — matching the sequential path, whose
+        DummyClassifier ignores instance weights exactly like the
+        reference's SQL aggregates (DummyClassifier.scala:102-109)."""
+        from ..boosting import line_search as _ls  # noqa: F401
+        from ..models.tree import DecisionTreeRegressionModel
+        from ..models.tree_grower import grow_forest
+
+        comm = get_comm()
+        learner = self.getOrNone("baseLearner") or self._default_base_learner()
+        k_stages = self.getNumBaseLearners()
+        lr_rate = self.getOrDefault("learningRate")
+        optimized = self.getOrDefault("optimizedWeights")
+
+        x, y, w = self._extract_xyw(dataset)
+        n, num_features = x.shape
+        device = x.device
+        num_classes = int(
+            comm.all_reduce_scalar(self._get_num_classes(dataset), "max")
+        )
+        binned = BinnedDataset(x, dataset)
+        edges, bins = binned.get(int(learner.getOrDefault("maxBins")))
+
+        loss = get_classification_loss(self.getLoss(), num_classes)
+        dim = loss.dim
+
+        # shared global init (see docstring)
+        init_strategy = self.getOrDefault("initStrategy")
+        if init_strategy == "prior" and dim == 1 and num_classes == 2:
+            prior = DummyClassifier().setStrategy("prior").fit(
+                TensorFrame(features=x, label=y, weight=w)
+            )
+            p1 = float(prior._prob[1])
+            logodds = torch.tensor([np.log(p1 / max(1.0 - p1, 1e-300))])
+            init = DummyClassificationModel.from_raw(
+                logodds, num_classes, num_features
+            )
+        else:
+            init = DummyClassifier().setStrategy(init_strategy).fit(
+                TensorFrame(features=x, label=y, weight=w)
+            )
+
+        fold = fold.to(device)
+        masks = [(fold != f) for f in range(num_folds)]
+        if include_full:
+            masks.append(torch.ones_like(fold, dtype=torch.bool))
+        Fo = len(masks)
+        mask_f = torch.stack([m.float() for m in masks], dim=1)  # [N, Fo]
+        wmask = w.unsqueeze(1) * mask_f
+        rows_per_fold = [m.nonzero(as_tuple=True)[0].to(torch.int32)
+                         for m in masks]
+        root_rows = [rows_per_fold[f] for f in range(Fo) for _ in range(dim)]
+
+        ylab = loss.encode_label(y)  # [N, dim]
+        init_raw = init.predictRaw(x)[:, :dim].contiguous()
+        margins = init_raw.unsqueeze(1).expand(n, Fo, dim).contiguous()
+        lab_e = ylab.unsqueeze(1).expand(n, Fo, dim).reshape(-1, dim) \
+            .contiguous()
+
+        gp = learner._grow_params(1.0)
+        stats = torch.stack([wmask.max()])
+        if comm.is_distributed:
+            comm.all_reduce_(stats, "max")
+        hmax = float(stats.cpu()[0])
+
+        fold_trees: List[List[List]] = [[] for _ in range(Fo)]
+        fold_weights: List[List[List[float]]] = [[] for _ in range(Fo)]
+        prev_sol: List = [None] * Fo
+        for i in range(k_stages):
+            g = loss.grad_hess_fused(
+                lab_e, margins.reshape(-1, dim), want_hess=False
+            )[0].reshape(n, Fo, dim)
+            grads_T = ((-g) * wmask.unsqueeze(2)).reshape(n, Fo * dim) \
+                .contiguous()
+            gmax = float(grads_T.abs().max())
+            if comm.is_distributed:
+                gmax = comm.all_reduce_scalar(gmax, "max")
+            gh_max = torch.tensor([max(gmax, 1e-30), max(hmax, 1e-30), 1.0])
+            hess_T = wmask.repeat_interleave(dim, dim=1).contiguous()
+            tp: list = []
+            trees = grow_forest(bins, edges, grads_T, hess_T, gp, comm,
+                                hess_is_count=False, train_pred_out=tp,
+                                gh_max_in=gh_max, root_rows=root_rows)
+            d = tp[0].reshape(n, Fo, dim)
+            wts_round = torch.ones(Fo, dim, device=device)
+            for f in range(Fo):
+                if optimized:
+                    mf = margins[:, f].contiguous()
+                    df = d[:, f].contiguous()
+                    if dim == 1:
+                        sol = [optimize_weight_1d(
+                            loss, ylab, mf, df, wmask[:, f].contiguous(),
+                            comm, self.getOrDefault("maxIter"),
+                            self.getOrDefault("tol"),
+                        )]
+                    else:
+                        sol = optimize_weight_nd(
+                            loss, ylab, mf, df, wmask[:, f].contiguous(),
+                            comm, self.getOrDefault("maxIter"),
+                            self.getOrDefault("tol"), x0=prev_sol[f],
+                        ).tolist()
+                        prev_sol[f] = sol
+                else:
+                    sol = [1.0] * dim
+                iweights = [s * lr_rate for s in sol]
+                fold_weights[f].append(iweights)
+                wts_round[f] = torch.tensor(iweights, device=device)
+                ms = []
+                for j in range(dim):
+                    m = DecisionTreeRegressionModel()
+                    m._set_tree(trees[f * dim + j], num_features)
+                    m._copy_cols_from(learner)
+                    ms.append(m)
+                fold_trees[f].append(ms)
+            margins = margins + d * wts_round.unsqueeze(0)
+
+        ident = torch.arange(num_features)
+        out = []
+        for f in range(Fo):
+            model = GBMClassificationModel()
+            model._init = init
+            model._models = fold_trees[f]
+            model._weights = fold_weights[f]
+            model._subspaces = [ident] * k_stages
+            model._num_classes = num_classes
+            model._dim = dim
+            model._num_features = num_features
+            model.set("loss", self.getLoss())
+            for p in (
+                "featuresCol", "labelCol", "predictionCol",
+                "rawPredictionCol", "probabilityCol",
+            ):
+                model.set(p, self.getOrDefault(p))
+            out.append(model)
+        if include_full:
+            return out[:-1], out[-1]
+        return out
+
     def _can_fuse_round(self, learner, res_weight, fuse_state) -> bool:
         """Fused K-tree rounds need: built-in tree learner, no per-tree
         weight thresholds, and strictly positive weights on every row

@@ -104,6 +104,21 @@ class StackingClassifier(Predictor, _StackingClassifierParams):
             g = torch.Generator().manual_seed(self.getOrDefault("seed"))
             fold = torch.randint(0, num_folds, (n,), generator=g).to(x.device)
 
+            # fold-vectorized learners (default-config GBM classifiers)
+            # grow all their fold models jointly — see regression/stacking
+            fused_ok = bool((use_w > 0).all())
+            fused = {
+                mi for mi, lr in enumerate(learners)
+                if fused_ok and hasattr(lr, "_can_fit_folds")
+                and lr._can_fit_folds()
+            }
+            fold_models: dict = {}
+            final_models: dict = {}
+            for mi in fused:
+                fold_models[mi], final_models[mi] = learners[mi]._fit_folds(
+                    shared, fold, num_folds, include_full=True
+                )
+
             def fold_task(lr, f):
                 def task():
                     wmask = use_w * (fold != f).float()
@@ -113,20 +128,26 @@ class StackingClassifier(Predictor, _StackingClassifierParams):
                     )
                 return task
 
-            tasks = [fold_task(lr, f)
-                     for lr in learners for f in range(num_folds)]
+            plan = [(mi, f) for mi, lr in enumerate(learners)
+                    if mi not in fused for f in range(num_folds)]
+            tasks = [fold_task(learners[mi], f) for mi, f in plan]
+            non_fused = [mi for mi in range(len(learners)) if mi not in fused]
             tasks += [
-                (lambda lr=lr: self.fit_base_learner(
+                (lambda lr=learners[mi]: self.fit_base_learner(
                     lr, shared, weight_col="weight"))
-                for lr in learners
+                for mi in non_fused
             ]
             fitted = parallel_fits(tasks, self.getParallelism(),
                                    warm_first=True)
+            for j, (mi, f) in enumerate(plan):
+                fold_models.setdefault(mi, [None] * num_folds)[f] = fitted[j]
+            for j, mi in enumerate(non_fused):
+                final_models[mi] = fitted[len(plan) + j]
             meta_cols: List[torch.Tensor] = []
             for mi, lr in enumerate(learners):
                 col = None
                 for f in range(num_folds):
-                    m = fitted[mi * num_folds + f]
+                    m = fold_models[mi][f]
                     sel = fold == f
                     part = _model_meta(m, x[sel], method)
                     if col is None:
@@ -136,7 +157,7 @@ class StackingClassifier(Predictor, _StackingClassifierParams):
                     col[sel] = part
                 meta_cols.append(col)
             meta = torch.cat(meta_cols, dim=1)
-            models = fitted[len(learners) * num_folds:]
+            models = [final_models[mi] for mi in range(len(learners))]
 
         stack = self.fit_base_learner(
             stacker,

@@ -251,6 +251,58 @@ def test_fused_classification_bagging_matches_sequential():
         assert m._num_classes == 3
 
 
+def test_gbm_classifier_fit_folds_matches_sequential():
+    """GBMClassifier._fit_folds: fused (fold x class) forests + per-fold
+    stage searches must reproduce sequential weight-masked fits."""
+    import spark_ensemble_amd as sea
+    from spark_ensemble_amd.frame import TensorFrame
+    from spark_ensemble_amd.utils.io import synthetic_classification
+
+    df = synthetic_classification(9000, 12, k=3, seed=29)
+    x, y = df["features"], df["label"]
+    T = 3
+    g = torch.Generator().manual_seed(8)
+    fold = torch.randint(0, T, (x.shape[0],), generator=g)
+
+    est = sea.GBMClassifier().setNumBaseLearners(3).setSeed(4)
+    assert est._can_fit_folds()
+    shared = TensorFrame(features=x, label=y, weight=torch.ones_like(y))
+    fused, final = est._fit_folds(shared, fold, T, include_full=True)
+    assert len(fused) == T
+
+    for t in range(T):
+        wmask = (fold != t).float()
+        seq = (
+            sea.GBMClassifier().setNumBaseLearners(3).setSeed(4)
+            .set("weightCol", "weight")
+            .fit(TensorFrame(features=x, label=y, weight=wmask))
+        )
+        for wa_row, wb_row in zip(fused[t]._weights, seq._weights):
+            for wa, wb in zip(wa_row, wb_row):
+                assert abs(wa - wb) < 1e-4, (wa, wb)
+        a = fused[t].transform(df)["probability"]
+        b = seq.transform(df)["probability"]
+        close = float((a - b).abs().lt(0.02).float().mean())
+        assert close > 0.97, close
+        held = fold == t
+        acc_a = float((fused[t].transform(df)["prediction"][held]
+                       == y[held]).float().mean())
+        acc_b = float((seq.transform(df)["prediction"][held]
+                       == y[held]).float().mean())
+        assert abs(acc_a - acc_b) < 0.02, (acc_a, acc_b)
+
+    # the rode-along full refit equals a plain fit
+    seq_full = (
+        sea.GBMClassifier().setNumBaseLearners(3).setSeed(4)
+        .set("weightCol", "weight")
+        .fit(TensorFrame(features=x, label=y, weight=torch.ones_like(y)))
+    )
+    a = final.transform(df)["probability"]
+    b = seq_full.transform(df)["probability"]
+    assert torch.allclose(a, b, rtol=1e-3, atol=1e-3), \
+        float((a - b).abs().max())
+
+
 def test_bagging_fit_folds_matches_sequential():
     """BaggingRegressor._fit_folds (all fold x member trees in one
     forest) must reproduce per-fold weight-masked fits."""
