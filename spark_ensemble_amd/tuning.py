@@ -310,6 +310,20 @@ class CrossValidator(Estimator):
         assert self.estimator is not None and self.evaluator is not None
         folds = self._fold_ids(dataset.count(), dataset.device)
 
+        def scores_fused(pm):
+            """Fold-vectorized estimators (default-config GBM / Bagging)
+            grow every fold model jointly — CV's leave-one-fold-out IS
+            the _fit_folds shape."""
+            est = self.estimator.copy(pm) if pm else self.estimator
+            if not (hasattr(est, "_can_fit_folds") and est._can_fit_folds()):
+                return None
+            models = est._fit_folds(dataset, folds, self.numFolds)
+            out = []
+            for f in range(self.numFolds):
+                va = dataset.filter(folds == f)
+                out.append(self.evaluator.evaluate(models[f].transform(va)))
+            return out
+
         def cell(pm, f):
             def task():
                 tr = dataset.filter(folds != f)
@@ -319,15 +333,28 @@ class CrossValidator(Estimator):
             return task
 
         # (param-map, fold) fits run concurrently on per-thread HIP
-        # streams (the Spark CrossValidator parallelism analog)
+        # streams (the Spark CrossValidator parallelism analog); fused
+        # param maps skip the pool entirely
         from .parallel.streams import parallel_fits
 
-        tasks = [cell(pm, f) for pm in self.estimatorParamMaps
-                 for f in range(self.numFolds)]
+        fused_scores = {}
+        plan = []
+        for i, pm in enumerate(self.estimatorParamMaps):
+            sc = scores_fused(pm)
+            if sc is not None:
+                fused_scores[i] = sc
+            else:
+                plan.append(i)
+        tasks = [cell(self.estimatorParamMaps[i], f)
+                 for i in plan for f in range(self.numFolds)]
         flat = parallel_fits(tasks, self.parallelism, warm_first=True)
         avg = []
         for i in range(len(self.estimatorParamMaps)):
-            chunk = flat[i * self.numFolds:(i + 1) * self.numFolds]
+            if i in fused_scores:
+                chunk = fused_scores[i]
+            else:
+                j = plan.index(i)
+                chunk = flat[j * self.numFolds:(j + 1) * self.numFolds]
             avg.append(sum(chunk) / len(chunk))
         better = max if self.evaluator.isLargerBetter() else min
         best_idx = avg.index(better(avg))
