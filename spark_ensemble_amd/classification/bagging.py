@@ -165,6 +165,13 @@ class BaggingClassifier(ProbabilisticClassifier, _BaggingClassifierParams):
 class BaggingClassificationModel(
     ProbabilisticClassificationModel, _BaggingClassifierParams
 ):
+    @property
+    def models(self):
+        return list(self._models)
+
+    @property
+    def subspaces(self):
+        return list(self._subspaces)
     _models: List = []
     _subspaces: List[torch.Tensor] = []
 

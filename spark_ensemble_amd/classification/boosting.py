@@ -192,6 +192,14 @@ class BoostingClassifier(ProbabilisticClassifier, _BoostingClassifierParams):
 class BoostingClassificationModel(
     ProbabilisticClassificationModel, _BoostingClassifierParams
 ):
+    # public accessors (reference BoostingClassificationModel fields)
+    @property
+    def models(self):
+        return list(self._models)
+
+    @property
+    def weights(self):
+        return list(self._weights)
     _models: List = []
     _weights: List[float] = []
 

@@ -537,6 +537,15 @@ class GBMClassifier(ProbabilisticClassifier, _GBMClassifierParams):
 
 
 class GBMClassificationModel(ProbabilisticClassificationModel, _GBMClassifierParams):
+    @property
+    def models(self):
+        """Per-(stage, class) base models (reference .models)."""
+        return [list(ms) for ms in self._models]
+
+    @property
+    def weights(self):
+        """Per-(stage, class) weights (reference .weights)."""
+        return [list(ws) for ws in self._weights]
     _init = None
     _models: List[List] = []
     _weights: List[List[float]] = []

@@ -185,6 +185,14 @@ class StackingClassifier(Predictor, _StackingClassifierParams):
 
 
 class StackingClassificationModel(PredictionModel, _StackingClassifierParams):
+    @property
+    def models(self):
+        """Fitted base models (reference field)."""
+        return list(self._models)
+
+    @property
+    def stack(self):
+        return self._stack
     _models: List = []
     _stack = None
 

@@ -217,6 +217,14 @@ class BaggingRegressor(Regressor, _BaggingRegressorParams):
 
 
 class BaggingRegressionModel(RegressionModel, _BaggingRegressorParams):
+    @property
+    def models(self):
+        """Fitted member models (reference BaggingRegressionModel.models)."""
+        return list(self._models)
+
+    @property
+    def subspaces(self):
+        return list(self._subspaces)
     _models: List = []
     _subspaces: List[torch.Tensor] = []
 

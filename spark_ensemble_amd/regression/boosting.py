@@ -179,6 +179,15 @@ class BoostingRegressor(Regressor, _BoostingRegressorParams):
 
 
 class BoostingRegressionModel(RegressionModel, _BoostingRegressorParams):
+    @property
+    def models(self):
+        """Reference BoostingRegressionModel.models."""
+        return list(self._models)
+
+    @property
+    def weights(self):
+        """Per-stage estimator weights (reference .weights)."""
+        return list(self._weights)
     _models: List = []
     _weights: List[float] = []
 

@@ -443,6 +443,15 @@ class GBMRegressor(Regressor, _GBMRegressorParams):
 
 
 class GBMRegressionModel(RegressionModel, _GBMRegressorParams):
+    @property
+    def models(self):
+        """Per-stage base models (reference GBMRegressionModel.models)."""
+        return list(self._models)
+
+    @property
+    def weights(self):
+        """Per-stage weights (reference .weights)."""
+        return list(self._weights)
     _init = None
     _models: List = []
     _weights: List[float] = []

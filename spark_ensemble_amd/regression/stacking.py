@@ -188,6 +188,15 @@ class StackingRegressor(Regressor, _StackingRegressorParams):
 
 
 class StackingRegressionModel(RegressionModel, _StackingRegressorParams):
+    @property
+    def models(self):
+        """Fitted base models (reference StackingRegressionModel.models)."""
+        return list(self._models)
+
+    @property
+    def stack(self):
+        """The meta-learner (reference .stack)."""
+        return self._stack
     _models: List = []
     _stack = None
 
