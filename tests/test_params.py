@@ -103,3 +103,41 @@ def test_mllib_metadata_layout(tmp_path):
     assert (p / "data-0" / "part-00000").exists()
     row = json.loads((p / "data-0" / "part-00000").read_text().strip())
     assert "subspace" in row
+
+
+def test_model_public_accessors():
+    """Every ensemble model exposes the reference's public fields
+    (.models, .weights / .subspaces / .stack)."""
+    import torch
+    from spark_ensemble_amd import (
+        BaggingRegressor, BoostingRegressor, GBMClassifier, GBMRegressor,
+        StackingRegressor,
+    )
+    from spark_ensemble_amd.frame import TensorFrame
+    from spark_ensemble_amd.models import LinearRegression
+
+    g = torch.Generator().manual_seed(0)
+    x = torch.randn(400, 5, generator=g)
+    y = x[:, 0] - 0.5 * x[:, 1]
+    df = TensorFrame(features=x, label=y)
+
+    m = GBMRegressor().setNumBaseLearners(2).fit(df)
+    assert len(m.models) == 2 and len(m.weights) == 2
+    b = BaggingRegressor().setNumBaseLearners(2).fit(df)
+    assert len(b.models) == 2 and len(b.subspaces) == 2
+    bo = BoostingRegressor().setNumBaseLearners(2).fit(df)
+    assert len(bo.models) == len(bo.weights) > 0
+    st = (
+        StackingRegressor()
+        .setBaseLearners([LinearRegression()])
+        .setStacker(LinearRegression())
+        .setNumFolds(2)
+        .fit(df)
+    )
+    assert len(st.models) == 1 and st.stack is not None
+
+    yc = (y > 0).float()
+    dfc = TensorFrame(features=x, label=yc)
+    c = GBMClassifier().setNumBaseLearners(2).fit(dfc)
+    assert len(c.models) == 2 and isinstance(c.models[0], list)
+    assert len(c.weights) == 2
