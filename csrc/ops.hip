@@ -42,7 +42,7 @@ static inline int64_t ceil_div(int64_t a, int64_t b) { return (a + b - 1) / b; }
 // fetch) before the slot is touched again.
 static torch::Tensor h2d_async(const void* src, size_t bytes, int slot,
                                const torch::Device& dev) {
-  static thread_local torch::Tensor pin[6];
+  static thread_local torch::Tensor pin[8];
   auto& b = pin[slot];
   if (!b.defined() || (size_t)b.numel() < bytes)
     b = torch::empty({(int64_t)std::max<size_t>(bytes, 4096)},
@@ -892,6 +892,119 @@ __global__ void forest_predict2_kernel(
 // bandwidth); this is a 64x64 LDS-tiled transpose moving u32 quads on
 // both sides (full tiles) — byte path only on edge tiles.
 // ---------------------------------------------------------------------------
+// gather_ranges / leaf_scatter: the forest grower's per-level row-arena
+// maintenance.  Both replace a 3-kernel torch expansion
+// (repeat_interleave + cumsum + arange gather — the reference's per-node
+// driver loops have no on-GPU analog) with ONE pass:
+// each output position binary-searches the tiny per-segment prefix table
+// (L2-resident) and streams contiguous source ranges.
+// ---------------------------------------------------------------------------
+
+__device__ inline int seg_of(const long long* __restrict__ cum, int n_segs,
+                             long long p) {
+  int lo = 0, hi = n_segs;  // cum[0] = 0, cum[n_segs] = total
+  while (hi - lo > 1) {
+    const int mid = (lo + hi) >> 1;
+    if (cum[mid] <= p) lo = mid; else hi = mid;
+  }
+  return lo;
+}
+
+__global__ void gather_ranges_kernel(
+    int* __restrict__ out,               // [total]
+    const int* __restrict__ src,
+    const long long* __restrict__ seg,   // [n_segs+1 cum..., n_segs starts...]
+    int n_segs, long long total) {
+  const long long* cum = seg;
+  const long long* starts = seg + n_segs + 1;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long p = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       p < total; p += stride) {
+    const int s = seg_of(cum, n_segs, p);
+    out[p] = src[starts[s] + (p - cum[s])];
+  }
+}
+
+__global__ void leaf_scatter_kernel(
+    float* __restrict__ tp,              // [N, T] row-major
+    const int* __restrict__ row_idx,     // level row arena
+    const long long* __restrict__ seg,   // [n_segs+1 cum..., n_segs starts...]
+    const int* __restrict__ tree,        // [n_segs]
+    const float* __restrict__ val,       // [n_segs]
+    int T, int n_segs, long long total) {
+  const long long* cum = seg;
+  const long long* starts = seg + n_segs + 1;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long p = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       p < total; p += stride) {
+    const int s = seg_of(cum, n_segs, p);
+    const int r = row_idx[starts[s] + (p - cum[s])];
+    tp[(long long)r * T + tree[s]] = val[s];
+  }
+}
+
+void gather_ranges(torch::Tensor out, torch::Tensor src,
+                   torch::Tensor starts, torch::Tensor lens) {
+  CHECK_GPU(out); CHECK_GPU(src);
+  const int n_segs = (int)starts.numel();
+  const long long total = out.numel();
+  if (n_segs == 0 || total == 0) return;
+  auto st = starts.accessor<int64_t, 1>();
+  auto ln = lens.accessor<int64_t, 1>();
+  std::vector<long long> seg_v(2 * n_segs + 1);
+  seg_v[0] = 0;
+  for (int i = 0; i < n_segs; ++i) {
+    seg_v[i + 1] = seg_v[i] + ln[i];
+    seg_v[n_segs + 1 + i] = st[i];
+  }
+  TORCH_CHECK(seg_v[n_segs] == total, "gather_ranges: lens sum != out size");
+  auto seg_b = h2d_async(seg_v.data(), seg_v.size() * 8, 5, src.device());
+  auto stream = at::hip::getCurrentHIPStream();
+  const int blocks =
+      (int)std::min<long long>(8192, ceil_div(total, (long long)1024));
+  hipLaunchKernelGGL(gather_ranges_kernel, dim3(blocks), dim3(256), 0, stream,
+                     out.data_ptr<int>(), src.data_ptr<int>(),
+                     (const long long*)seg_b.data_ptr(), n_segs, total);
+}
+
+void leaf_scatter(torch::Tensor tp, torch::Tensor row_idx,
+                  torch::Tensor starts, torch::Tensor lens,
+                  torch::Tensor tree, torch::Tensor val) {
+  CHECK_GPU(tp); CHECK_GPU(row_idx);
+  const int n_segs = (int)starts.numel();
+  if (n_segs == 0) return;
+  const int T = (int)tp.size(1);
+  auto st = starts.accessor<int64_t, 1>();
+  auto ln = lens.accessor<int64_t, 1>();
+  std::vector<long long> seg_v(2 * n_segs + 1);
+  seg_v[0] = 0;
+  for (int i = 0; i < n_segs; ++i) {
+    seg_v[i + 1] = seg_v[i] + ln[i];
+    seg_v[n_segs + 1 + i] = st[i];
+  }
+  const long long total = seg_v[n_segs];
+  if (total == 0) return;
+  auto seg_b = h2d_async(seg_v.data(), seg_v.size() * 8, 5, tp.device());
+  std::vector<int> tv(n_segs);
+  std::vector<float> vv(n_segs);
+  auto ta = tree.accessor<int64_t, 1>();
+  auto va = val.accessor<float, 1>();
+  for (int i = 0; i < n_segs; ++i) { tv[i] = (int)ta[i]; vv[i] = va[i]; }
+  auto tree_b = h2d_async(tv.data(), tv.size() * 4, 6, tp.device());
+  std::vector<char> vb(n_segs * 4);
+  std::memcpy(vb.data(), vv.data(), n_segs * 4);
+  auto val_b = h2d_async(vb.data(), vb.size(), 7, tp.device());
+  auto stream = at::hip::getCurrentHIPStream();
+  const int blocks =
+      (int)std::min<long long>(8192, ceil_div(total, (long long)1024));
+  hipLaunchKernelGGL(leaf_scatter_kernel, dim3(blocks), dim3(256), 0, stream,
+                     tp.data_ptr<float>(), row_idx.data_ptr<int>(),
+                     (const long long*)seg_b.data_ptr(),
+                     (const int*)tree_b.data_ptr(),
+                     (const float*)val_b.data_ptr(), T, n_segs, total);
+}
+
+// ---------------------------------------------------------------------------
 
 __global__ void transpose_u8_kernel(uint8_t* __restrict__ out,  // [F, N]
                                     const uint8_t* __restrict__ in,  // [N, F]
@@ -1517,6 +1630,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("forest_predict2", &forest_predict2,
         "LDS-staged tree-group-tiled forest predict");
   m.def("transpose_u8", &transpose_u8, "tiled u8 matrix transpose");
+  m.def("gather_ranges", &gather_ranges,
+        "one-pass concat of contiguous index ranges");
+  m.def("leaf_scatter", &leaf_scatter,
+        "scatter per-segment leaf values into the [N, T] train-pred matrix");
   m.def("grad_hess", &grad_hess, "fused per-row loss gradient/hessian");
   m.def("line_search_eval", &line_search_eval, "fused loss+grad line-search payload");
 }

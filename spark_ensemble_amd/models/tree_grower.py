@@ -94,7 +94,59 @@ def grow_tree(
     to its leaf, so this is a leaf-value scatter instead of the full tree
     walk GBM would otherwise pay per round (only produced when row_mask is
     None, i.e. the tree saw every row).
+
+    Delegates to the vectorized ``grow_forest`` with T=1 whenever the
+    channel budget allows (D + hess/count channels <= 8): the arena-based
+    grower keeps all per-level bookkeeping tensorized, while this module's
+    loop implementation (``_grow_tree_seq``) spends ~5 ms/tree at depth 8
+    in per-node Python loops and many-segment ``torch.cat`` copies
+    (measured on the flagship bench: ~9.7 ms of GPU idle per 20 ms round,
+    concentrated in the deep levels and the leaf-capture finalization).
+    The loop version remains the reference for the fused-vs-sequential
+    parity tests and the fallback for wide multi-output (gini D > 6)
+    trees.
     """
+    D = grad.shape[1]
+    if hess_is_count is None:
+        hess_is_count = bool((hess == 1).all())
+    if D + (1 if hess_is_count else 2) <= 8:
+        root_rows = None
+        if row_mask is not None:
+            root_rows = [row_mask.nonzero(as_tuple=True)[0].to(torch.int32)]
+        tp: Optional[list] = (
+            [] if (train_pred_out is not None and row_mask is None and D == 1)
+            else None
+        )
+        trees = grow_forest(
+            bins, edges, grad.unsqueeze(1), hess, params, comm,
+            hess_is_count, tp, gh_max_in, root_rows,
+            root_tot_in=(
+                root_tot_in.reshape(1, -1)
+                if root_tot_in is not None and row_mask is None else None
+            ),
+        )
+        if train_pred_out is not None and tp:
+            train_pred_out.append(tp[0])  # [N, 1] == [N, D]
+        return trees[0]
+    return _grow_tree_seq(bins, edges, grad, hess, params, comm, row_mask,
+                          hess_is_count, train_pred_out, gh_max_in,
+                          root_tot_in)
+
+
+def _grow_tree_seq(
+    bins: torch.Tensor,
+    edges: torch.Tensor,
+    grad: torch.Tensor,
+    hess: torch.Tensor,
+    params: GrowParams,
+    comm: Optional[Comm] = None,
+    row_mask: Optional[torch.Tensor] = None,
+    hess_is_count: Optional[bool] = None,
+    train_pred_out: Optional[list] = None,
+    gh_max_in: Optional[torch.Tensor] = None,
+    root_tot_in: Optional[torch.Tensor] = None,
+) -> Dict[str, torch.Tensor]:
+    """Loop-bookkeeping single-tree grower (see grow_tree docstring)."""
     device = bins.device
     N, F = bins.shape
     D = grad.shape[1]
@@ -512,6 +564,7 @@ def grow_forest(
     gh_max_in: Optional[torch.Tensor] = None,
     root_rows: Optional[List[torch.Tensor]] = None,  # per-tree row sets
     feature_masks: Optional[torch.Tensor] = None,  # [T, F] 1/0 subspaces
+    root_tot_in: Optional[torch.Tensor] = None,  # [T, C] GLOBAL cpu totals
 ) -> List[Dict[str, torch.Tensor]]:
     """Grow T single-output trees LEVEL-SYNCHRONOUSLY in fused launches.
 
@@ -606,7 +659,11 @@ def grow_forest(
     fi = torch.zeros(T, F, dtype=torch.float64)
 
     # root totals: one fused reduction (+ one all-reduce) for all trees
-    if root_rows is None:
+    # (callers with a fused stats pass hand in the already-reduced values,
+    # skipping the device sync — same contract as grow_tree's root_tot_in)
+    if root_tot_in is not None:
+        totals = root_tot_in.cpu().float()
+    elif root_rows is None:
         g_sum = g3.sum(dim=0)  # [T, D]
         if h_shared:
             h_sum = hess.sum().reshape(1).expand(T)
@@ -622,13 +679,14 @@ def grow_forest(
         g_sum = torch.stack(gs)  # [T, D]
         h_sum = torch.stack(hs)
         cnt_col = torch.tensor([float(v) for v in root_lens], device=device)
-    cols = [g_sum, h_sum.unsqueeze(1)]
-    if NN == 2:
-        cols.append(cnt_col.unsqueeze(1))
-    root_tot = torch.cat(cols, dim=1)  # [T, C] device
-    if comm is not None:
-        comm.all_reduce_(root_tot)
-    totals = root_tot.cpu()
+    if root_tot_in is None:
+        cols = [g_sum, h_sum.unsqueeze(1)]
+        if NN == 2:
+            cols.append(cnt_col.unsqueeze(1))
+        root_tot = torch.cat(cols, dim=1)  # [T, C] device
+        if comm is not None:
+            comm.all_reduce_(root_tot)
+        totals = root_tot.cpu()
 
     # active level state (host tensors; sorted by tree by construction)
     node_tree = torch.arange(T, dtype=torch.long)
@@ -645,26 +703,12 @@ def grow_forest(
     idx_c = C - 1
 
     capture = train_pred_out is not None and D == 1
-    cap_rows: List[torch.Tensor] = []
-    cap_trees: List[torch.Tensor] = []
-    cap_vals: List[torch.Tensor] = []
-
-    def _ragged_positions(starts, lens):
-        """Device index positions for the concatenation of
-        [starts_i, starts_i + lens_i) ranges (cpu long inputs).  Only the
-        SMALL per-node starts/lens tensors cross to the device; the
-        [total]-sized expansion happens there (a cpu-side expansion
-        pageable-copied per level measured 16-60x slower end to end)."""
-        total = int(lens.sum())
-        if total == 0:
-            return torch.empty(0, dtype=torch.long, device=device)
-        starts_d = _to_dev_async(starts, device)
-        lens_d = _to_dev_async(lens, device)
-        base = torch.repeat_interleave(starts_d, lens_d)
-        firsts = torch.cumsum(lens_d, 0) - lens_d
-        within = torch.arange(total, dtype=torch.long, device=device) - \
-            torch.repeat_interleave(firsts, lens_d)
-        return base + within
+    # leaf capture scatters straight into the preallocated [N, T] margin
+    # matrix (one fused kernel per finalize — no index lists)
+    tp_arena = (
+        torch.zeros(N, T, dtype=torch.float32, device=device)
+        if capture else None
+    )
 
     def _finalize_and_capture(idx_t, offs, ridx):
         if idx_t.numel() == 0:
@@ -676,21 +720,13 @@ def grow_forest(
         denom = (h + params.lam).unsqueeze(1)
         vals = torch.where(denom > 0, g / denom, torch.zeros_like(g))
         leaves_t[t_ids, nids] = vals
-        vals = vals[:, 0]  # capture is D == 1 only
         if capture:
             starts = offs[idx_t]
             lens = offs[idx_t + 1] - starts
             nz = (lens > 0).nonzero(as_tuple=True)[0]
             if nz.numel():
-                pos = _ragged_positions(starts[nz], lens[nz])
-                lens_d = lens[nz].to(device)
-                cap_rows.append(ridx[pos].long())
-                cap_trees.append(
-                    torch.repeat_interleave(t_ids[nz].to(device), lens_d)
-                )
-                cap_vals.append(
-                    torch.repeat_interleave(vals[nz].to(device), lens_d)
-                )
+                ops.leaf_scatter(tp_arena, ridx, starts[nz], lens[nz],
+                                 t_ids[nz], vals[nz, 0])
 
     for depth in range(params.max_depth + 1):
         n_active = int(node_tree.numel())
@@ -731,7 +767,7 @@ def grow_forest(
                 lens = offsets[built_cpu + 1] - starts
                 b_off = torch.zeros(built_cpu.numel() + 1, dtype=torch.int64)
                 b_off[1:] = torch.cumsum(lens, 0)
-                build_rows = row_idx[_ragged_positions(starts, lens)]
+                build_rows = ops.gather_ranges(row_idx, starts, lens)
                 col0_b = (node_tree[built_cpu] * C).to(torch.int32)
                 bh = ops.hist_build_forest(
                     bins, gh, build_rows, b_off, col0_b, B, C, gh_max,
@@ -806,7 +842,7 @@ def grow_forest(
         if int(s_idx.numel()) == n_active:
             row_idx = new_rows
         else:
-            row_idx = new_rows[_ragged_positions(ls, re - ls)]
+            row_idx = ops.gather_ranges(new_rows, ls, re - ls)
         hists = hists.index_select(0, _to_dev_async(s_idx, device))
 
     # assemble per-tree arrays
@@ -826,13 +862,7 @@ def grow_forest(
         trees.append({k: _to_dev_async(v, device) for k, v in tree.items()})
 
     if capture:
-        tp = torch.zeros(N, T, dtype=torch.float32, device=device)
-        if cap_rows:
-            tp.index_put_(
-                (torch.cat(cap_rows), torch.cat(cap_trees)),
-                torch.cat(cap_vals),
-            )
-        train_pred_out.append(tp)
+        train_pred_out.append(tp_arena)
     return trees
 
 

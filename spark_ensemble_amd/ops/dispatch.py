@@ -263,6 +263,47 @@ def partition_rows_async(bins, row_idx, node_offsets, feat, thr):
     return new_rows, left_counts, offs_cpu
 
 
+def gather_ranges(src, starts, lens):
+    """Concatenation of ``src[starts_i : starts_i + lens_i)`` ranges.
+
+    src is an int32 device row arena; starts/lens are SMALL cpu int64
+    tensors (one entry per active node).  GPU: one kernel pass (binary
+    search over the L2-resident prefix table) instead of torch's
+    repeat_interleave + cumsum + arange + gather chain.  CPU: slice cat.
+    """
+    if src.is_cuda:
+        m = _require_hip("gather_ranges")
+        if m is not None:
+            out = torch.empty(int(lens.sum()), dtype=torch.int32,
+                              device=src.device)
+            m.gather_ranges(out, src, starts.to(torch.int64),
+                            lens.to(torch.int64))
+            return out
+    st, ln = starts.tolist(), lens.tolist()
+    segs = [src[s:s + l] for s, l in zip(st, ln) if l > 0]
+    if not segs:
+        return torch.empty(0, dtype=src.dtype, device=src.device)
+    return segs[0].clone() if len(segs) == 1 else torch.cat(segs)
+
+
+def leaf_scatter(tp, row_idx, starts, lens, tree, val):
+    """tp[row_idx[starts_i : starts_i+lens_i], tree_i] = val_i for every
+    segment i — the train-pred leaf capture, fused into one kernel on GPU
+    (no device-side index lists are materialized at all)."""
+    if tp.is_cuda:
+        m = _require_hip("leaf_scatter")
+        if m is not None:
+            m.leaf_scatter(tp, row_idx, starts.to(torch.int64),
+                           lens.to(torch.int64), tree.to(torch.int64),
+                           val.to(torch.float32))
+            return
+    st, ln = starts.tolist(), lens.tolist()
+    tl, vl = tree.tolist(), val.tolist()
+    for s, l, t, v in zip(st, ln, tl, vl):
+        if l > 0:
+            tp[row_idx[s:s + l].long(), int(t)] = float(v)
+
+
 def partition_rows_finish(new_rows, left_counts, offs_cpu):
     n = offs_cpu.numel() - 1
     lc = left_counts.cpu().to(torch.int64)

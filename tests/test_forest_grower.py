@@ -9,6 +9,7 @@ import torch
 
 from spark_ensemble_amd.models.tree_grower import (
     GrowParams,
+    _grow_tree_seq,
     grow_forest,
     grow_tree,
 )
@@ -50,7 +51,7 @@ def test_forest_matches_independent_trees(weighted):
 
     for t in range(T):
         single_pred = []
-        single = grow_tree(bins, edges, grads[:, t:t + 1].contiguous(),
+        single = _grow_tree_seq(bins, edges, grads[:, t:t + 1].contiguous(),
                            hess, params, train_pred_out=single_pred)
         _assert_tree_equal(forest[t], single, t)
         assert torch.allclose(forest_pred[0][:, t], single_pred[0][:, 0],
@@ -67,7 +68,7 @@ def test_forest_per_tree_hessians():
 
     forest = grow_forest(bins, edges, grads, hess, params)
     for t in range(T):
-        single = grow_tree(bins, edges, grads[:, t:t + 1].contiguous(),
+        single = _grow_tree_seq(bins, edges, grads[:, t:t + 1].contiguous(),
                            hess[:, t].contiguous(), params)
         _assert_tree_equal(forest[t], single, t)
 
@@ -115,7 +116,7 @@ def test_forest_root_rows_matches_masked_trees():
     forest = grow_forest(bins, edges, grads, hess, params,
                          train_pred_out=pred_f, root_rows=rows)
     for t in range(T):
-        single = grow_tree(bins, edges, grads[:, t:t + 1].contiguous(),
+        single = _grow_tree_seq(bins, edges, grads[:, t:t + 1].contiguous(),
                            hess, params, row_mask=(fold != t))
         _assert_tree_equal(forest[t], single, t)
         # captured training predictions are zero on held-out rows
@@ -392,3 +393,38 @@ def test_hist_build_forest_reference():
             torch.tensor([0, int(offsets[nd + 1] - offsets[nd])]), b,
         )
         assert torch.allclose(got[nd], want[0], atol=1e-4)
+
+
+@pytest.mark.parametrize("masked", [False, True])
+def test_grow_tree_delegation_matches_loop_grower(masked):
+    """grow_tree (T=1 grow_forest delegation) == the loop implementation,
+    with and without a zero-weight row mask."""
+    n, f, b = 15000, 10, 32
+    x, edges, bins, g = _data(n, f, b, seed=7)
+    params = GrowParams(max_depth=5, max_bins=b)
+    grad = torch.randn(n, 1, generator=g)
+    hess = torch.ones(n)
+    mask = (torch.rand(n, generator=g) > 0.3) if masked else None
+
+    tp_new, tp_old = [], []
+    new = grow_tree(bins, edges, grad, hess, params, row_mask=mask,
+                    train_pred_out=tp_new)
+    old = _grow_tree_seq(bins, edges, grad, hess, params, row_mask=mask,
+                         train_pred_out=tp_old)
+    _assert_tree_equal(new, old, "delegated")
+    if not masked:
+        # same tolerance as the leaf values (fused column-sum order)
+        assert torch.allclose(tp_new[0], tp_old[0], rtol=1e-4, atol=2e-5)
+
+
+def test_grow_tree_delegation_multioutput():
+    """D>1 (gini) delegation parity, plus the wide-D fallback path."""
+    n, f, b, D = 12000, 8, 32, 3
+    x, edges, bins, g = _data(n, f, b, seed=11)
+    params = GrowParams(max_depth=4, max_bins=b)
+    onehot = torch.zeros(n, D)
+    onehot[torch.arange(n), torch.randint(0, D, (n,), generator=g)] = 1.0
+    hess = torch.ones(n)
+    new = grow_tree(bins, edges, onehot, hess, params)
+    old = _grow_tree_seq(bins, edges, onehot, hess, params)
+    _assert_tree_equal(new, old, "gini")
