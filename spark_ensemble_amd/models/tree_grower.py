@@ -710,6 +710,12 @@ def grow_forest(
         if capture else None
     )
 
+    # software pipeline: the previous level's arena writes (pure host
+    # bookkeeping, read only at final assembly) are DEFERRED until after
+    # this level's kernels are queued, so the GPU is never idle behind
+    # Python (measured ~110 us of idle per level without this)
+    pending_writes = None
+
     def _finalize_and_capture(idx_t, offs, ridx):
         if idx_t.numel() == 0:
             return
@@ -784,6 +790,10 @@ def grow_forest(
         if bins.is_cuda:
             pr_async = ops.partition_rows_async(bins, row_idx, offsets, feat, b)
 
+        if pending_writes is not None:
+            pending_writes()
+            pending_writes = None
+
         gain_cpu = gain.cpu()
         feat_cpu = feat.cpu()
         b_cpu = b.cpu()
@@ -795,27 +805,35 @@ def grow_forest(
         if not bool(do_split.any()):
             break
 
-        split_feat = torch.where(do_split, feat_cpu, torch.full_like(feat_cpu, -1))
         s_idx = do_split.nonzero(as_tuple=True)[0]  # cpu long, sorted
         t_ids = node_tree[s_idx]
         nids = node_nid[s_idx]
         f_l = feat_cpu[s_idx].long()
         b_l = b_cpu[s_idx].long()
-        feats_t[t_ids, nids] = f_l.to(torch.int32)
-        thrs_t[t_ids, nids] = edges_cpu[f_l, b_l]
-        fi.index_put_((t_ids, f_l), gain_cpu[s_idx].double(), accumulate=True)
         # child allocation: split nodes are grouped by tree, so the rank
-        # within the tree's segment gives each pair's base id
+        # within the tree's segment gives each pair's base id (cur_len is
+        # read for the NEXT level's ids, so it updates immediately; the
+        # arena writes themselves are deferred — see pending_writes)
         counts = torch.bincount(t_ids, minlength=T)
         firsts = torch.cumsum(counts, 0) - counts
         rank = torch.arange(s_idx.numel(), dtype=torch.long) - firsts[t_ids]
         cid = cur_len[t_ids] + 2 * rank
-        lefts_t[t_ids, nids] = cid.to(torch.int32)
         cur_len += 2 * counts
 
+        def _writes(t_ids=t_ids, nids=nids, f_l=f_l, b_l=b_l,
+                    gains=gain_cpu[s_idx].double(), cid=cid):
+            feats_t[t_ids, nids] = f_l.to(torch.int32)
+            thrs_t[t_ids, nids] = edges_cpu[f_l, b_l]
+            fi.index_put_((t_ids, f_l), gains, accumulate=True)
+            lefts_t[t_ids, nids] = cid.to(torch.int32)
+        pending_writes = _writes
+
+        hists = hists.index_select(0, _to_dev_async(s_idx, device))
         if pr_async is not None:
             new_rows, new_offs, _ = ops.partition_rows_finish(*pr_async)
         else:
+            split_feat = torch.where(do_split, feat_cpu,
+                                     torch.full_like(feat_cpu, -1))
             new_rows, new_offs, _ = ops.partition_rows(
                 bins, row_idx, offsets,
                 split_feat.to(torch.int32), b_cpu.to(torch.int32),
@@ -843,7 +861,9 @@ def grow_forest(
             row_idx = new_rows
         else:
             row_idx = ops.gather_ranges(new_rows, ls, re - ls)
-        hists = hists.index_select(0, _to_dev_async(s_idx, device))
+
+    if pending_writes is not None:
+        pending_writes()
 
     # assemble per-tree arrays
     trees: List[Dict[str, torch.Tensor]] = []
