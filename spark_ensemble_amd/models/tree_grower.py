@@ -794,10 +794,17 @@ def grow_forest(
             pending_writes()
             pending_writes = None
 
-        gain_cpu = gain.cpu()
-        feat_cpu = feat.cpu()
-        b_cpu = b.cpu()
-        left_stats = left_stats.cpu()
+        # ONE syncing D2H for the level's split results (gain / feat / bin
+        # ids are exact in f32 — F < 2^24, B <= 256); four separate .cpu()
+        # calls cost four stream round trips
+        pack = torch.cat([
+            gain, feat.to(torch.float32), b.to(torch.float32),
+            left_stats.reshape(-1),
+        ]).cpu()
+        gain_cpu = pack[:n_active]
+        feat_cpu = pack[n_active:2 * n_active].to(torch.int32)
+        b_cpu = pack[2 * n_active:3 * n_active].to(torch.int32)
+        left_stats = pack[3 * n_active:].reshape(n_active, C)
 
         do_split = torch.isfinite(gain_cpu)
         ns_idx = (~do_split).nonzero(as_tuple=True)[0]
