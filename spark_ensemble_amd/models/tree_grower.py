@@ -786,6 +786,23 @@ def grow_forest(
                 _to_dev_async(parent_of_t[nb_cpu], device), n_active,
                 (F, B, C), device, comm, split_args, split_mask=lvl_mask,
             )
+        # ONE D2H for the level's split results (gain / feat / bin ids
+        # are exact in f32 — F < 2^24, B <= 256), queued BEFORE the
+        # partition kernel and awaited via an event: the host reads the
+        # split outcome the moment split_argmax finishes and does all its
+        # bookkeeping WHILE the partition kernel still runs
+        pack_d = torch.cat([
+            gain, feat.to(torch.float32), b.to(torch.float32),
+            left_stats.reshape(-1),
+        ])
+        ev = None
+        if bins.is_cuda:
+            pack_h = torch.empty(pack_d.numel(), dtype=torch.float32,
+                                 pin_memory=True)
+            pack_h.copy_(pack_d, non_blocking=True)
+            ev = torch.cuda.Event()
+            ev.record()
+
         pr_async = None
         if bins.is_cuda:
             pr_async = ops.partition_rows_async(bins, row_idx, offsets, feat, b)
@@ -794,13 +811,11 @@ def grow_forest(
             pending_writes()
             pending_writes = None
 
-        # ONE syncing D2H for the level's split results (gain / feat / bin
-        # ids are exact in f32 — F < 2^24, B <= 256); four separate .cpu()
-        # calls cost four stream round trips
-        pack = torch.cat([
-            gain, feat.to(torch.float32), b.to(torch.float32),
-            left_stats.reshape(-1),
-        ]).cpu()
+        if ev is not None:
+            ev.synchronize()
+            pack = pack_h
+        else:
+            pack = pack_d
         gain_cpu = pack[:n_active]
         feat_cpu = pack[n_active:2 * n_active].to(torch.int32)
         b_cpu = pack[2 * n_active:3 * n_active].to(torch.int32)
