@@ -1374,6 +1374,74 @@ __global__ void ls_reduce_kernel(float* __restrict__ out,      // [out_width]
   }
 }
 
+// Safeguarded-Newton update on the 1-D stage-weight search state —
+// the device side of the chained line search (line_search.py _newton_1d):
+// state = [a, blo, bhi, best_a, best_f, evals, done]; payload = [f, g, h]
+// from the preceding eval at coeff = &state[0].  Mirrors the host loop
+// exactly so a chain interrupted mid-way can resume on the host.
+__global__ void newton_update_kernel(float* __restrict__ st,
+                                     const float* __restrict__ p,
+                                     float tol) {
+  if (st[6] != 0.0f) return;
+  const float f = p[0], g = p[1], h = p[2];
+  if (!isfinite(f) || !isfinite(g) || !isfinite(h)) { st[6] = 2.0f; return; }
+  st[5] += 1.0f;
+  const float a = st[0];
+  if (f < st[4]) { st[4] = f; st[3] = a; }
+  if (g > 0.0f) st[2] = a; else st[1] = a;
+  if (fabsf(g) <= tol * fmaxf(1.0f, fabsf(f)) || (st[2] - st[1]) <= tol) {
+    st[6] = 1.0f;
+    return;
+  }
+  float nxt;
+  if (h > 1e-12f) {
+    nxt = a - g / h;
+    if (!(st[1] < nxt && nxt < st[2])) nxt = 0.5f * (st[1] + st[2]);
+    if (fabsf(nxt - a) <= tol * fmaxf(1.0f, fabsf(a))) {
+      st[0] = nxt;
+      st[6] = 1.0f;
+      return;
+    }
+  } else {
+    nxt = 0.5f * (st[1] + st[2]);
+  }
+  st[0] = nxt;
+}
+
+void newton_chain_1d(torch::Tensor state, torch::Tensor payloads,
+                     torch::Tensor label, torch::Tensor pred,
+                     torch::Tensor dir, torch::Tensor weight,
+                     int64_t loss_id, double param, double tol,
+                     int64_t iters) {
+  // queue `iters` x (eval -> reduce -> update) with NO host sync: the
+  // eval kernel reads the live alpha from state[0] (its coeff pointer),
+  // the update kernel advances it.  Steps after convergence evaluate at
+  // a frozen alpha and are discarded by the update's done guard — the
+  // caller sizes `iters` from the previous round's eval count, so the
+  // overshoot is ~0 in steady state.
+  CHECK_GPU(state); CHECK_GPU(pred);
+  const int64_t n = pred.size(0);
+  TORCH_CHECK(pred.size(1) == 1, "newton_chain_1d is scalar-dim only");
+  TORCH_CHECK(state.numel() >= 7 && payloads.numel() >= 3 * iters, "sizes");
+  auto stream = at::hip::getCurrentHIPStream();
+  const int blocks = (int)std::min<int64_t>(ceil_div(n, 256 * 8), 2048);
+  auto partials = torch::empty({(int64_t)blocks * 3}, state.options());
+  const int rthreads = (256 / 3) * 3;
+  for (int k = 0; k < (int)iters; ++k) {
+    float* pay = payloads.data_ptr<float>() + 3 * k;
+    hipLaunchKernelGGL(line_search_eval_kernel<1>, dim3(blocks), dim3(256), 0,
+                       stream, partials.data_ptr<float>(),
+                       label.data_ptr<float>(), pred.data_ptr<float>(),
+                       dir.data_ptr<float>(), weight.data_ptr<float>(),
+                       state.data_ptr<float>(), n, (int)loss_id,
+                       (float)param, 1);
+    hipLaunchKernelGGL(ls_reduce_kernel, dim3(1), dim3(rthreads), 0, stream,
+                       pay, partials.data_ptr<float>(), blocks, 3, 3);
+    hipLaunchKernelGGL(newton_update_kernel, dim3(1), dim3(1), 0, stream,
+                       state.data_ptr<float>(), pay, (float)tol);
+  }
+}
+
 void line_search_eval(torch::Tensor payload, torch::Tensor label,
                       torch::Tensor pred, torch::Tensor dir,
                       torch::Tensor weight, torch::Tensor coeff,
@@ -1636,4 +1704,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "scatter per-segment leaf values into the [N, T] train-pred matrix");
   m.def("grad_hess", &grad_hess, "fused per-row loss gradient/hessian");
   m.def("line_search_eval", &line_search_eval, "fused loss+grad line-search payload");
+  m.def("newton_chain_1d", &newton_chain_1d,
+        "device-chained safeguarded-Newton stage-weight search");
 }

@@ -168,6 +168,9 @@ def optimize_weight_nd(
     return res.x.astype(np.float64)
 
 
+_CHAIN_K = [4]  # adaptive device-chain length (last search's evals)
+
+
 def _newton_1d(loss, label, pred, direction, weight, comm, max_iter, tol,
                lo, hi):
     """Safeguarded Newton on phi'(a) over [lo, hi]; returns None to fall
@@ -186,7 +189,47 @@ def _newton_1d(loss, label, pred, direction, weight, comm, max_iter, tol,
     blo, bhi = lo, hi
     best_a, best_f = None, float("inf")
     LAST_EVALS = 0
-    for _ in range(max(8, min(max_iter, 20))):
+    total = max(8, min(max_iter, 20))
+
+    # GPU fast path: a device-CHAINED sequence of (eval -> safeguarded
+    # update) kernels — the update runs on device and feeds the next
+    # eval's alpha, so the whole search costs ONE host round trip
+    # instead of one per iteration.  Chain length adapts to the previous
+    # search's eval count (steady state: zero wasted evals).  Disabled
+    # when distributed (the per-eval all-reduce needs the host).
+    if (pred.is_cuda and pred.shape[1] == 1
+            and (comm is None or not comm.is_distributed)):
+        from ..ops import dispatch
+
+        m = dispatch._require_hip("newton_chain_1d")
+        if m is not None:
+            k_chain = min(8, max(2, _CHAIN_K[0]))
+            state = torch.tensor(
+                [1.0, lo, hi, 0.0, float("inf"), 0.0, 0.0],
+                dtype=torch.float32, device=pred.device,
+            )
+            payloads = torch.zeros(k_chain, 3, dtype=torch.float32,
+                                   device=pred.device)
+            m.newton_chain_1d(
+                state, payloads, label.contiguous(), pred.contiguous(),
+                direction.contiguous(), weight.contiguous(),
+                loss.loss_id, float(loss.param), float(max(tol, 1e-8)),
+                k_chain,
+            )
+            st = state.cpu()
+            evals, done = int(st[5]), int(st[6])
+            LAST_EVALS = evals
+            _CHAIN_K[0] = max(2, evals + (0 if done == 1 else 2))
+            if done == 1:
+                return float(st[0])
+            if done == 2:
+                return None  # non-finite: Brent fallback
+            # chain exhausted before convergence: resume on the host
+            a, blo, bhi = float(st[0]), float(st[1]), float(st[2])
+            if evals:
+                best_f, best_a = float(st[4]), float(st[3])
+
+    for _ in range(total - LAST_EVALS):
         f, g, h = eval_at(a)
         LAST_EVALS += 1
         if not (np.isfinite(f) and np.isfinite(g) and np.isfinite(h)):

@@ -257,10 +257,18 @@ def partition_rows_async(bins, row_idx, node_offsets, feat, thr):
         f32,
         t32,
     )
+    # prefetch the left counts: non-blocking D2H + event, so finish()
+    # wakes the host the moment the kernel ends (a blocking .cpu() there
+    # costs an extra stream round trip after the partition)
+    lc_h = torch.empty(left_counts.numel(), dtype=torch.int32,
+                       pin_memory=True)
+    lc_h.copy_(left_counts, non_blocking=True)
+    ev = torch.cuda.Event()
+    ev.record()
     if _prof:
         _t2 = _time.perf_counter()
         print(f"[part prof] prep={(_t1-_t0)*1000:.2f} native={(_t2-_t1)*1000:.2f}")
-    return new_rows, left_counts, offs_cpu
+    return new_rows, (lc_h, ev), offs_cpu
 
 
 def gather_ranges(src, starts, lens):
@@ -306,7 +314,12 @@ def leaf_scatter(tp, row_idx, starts, lens, tree, val):
 
 def partition_rows_finish(new_rows, left_counts, offs_cpu):
     n = offs_cpu.numel() - 1
-    lc = left_counts.cpu().to(torch.int64)
+    if isinstance(left_counts, tuple):
+        lc_h, ev = left_counts
+        ev.synchronize()
+        lc = lc_h.to(torch.int64)
+    else:
+        lc = left_counts.cpu().to(torch.int64)
     sizes = torch.empty(2 * n, dtype=torch.int64)
     seg = offs_cpu[1:] - offs_cpu[:-1]
     sizes[0::2] = lc
