@@ -851,17 +851,11 @@ def grow_forest(
         pending_writes = _writes
 
         hists = hists.index_select(0, _to_dev_async(s_idx, device))
-        if pr_async is not None:
-            new_rows, new_offs, _ = ops.partition_rows_finish(*pr_async)
-        else:
-            split_feat = torch.where(do_split, feat_cpu,
-                                     torch.full_like(feat_cpu, -1))
-            new_rows, new_offs, _ = ops.partition_rows(
-                bins, row_idx, offsets,
-                split_feat.to(torch.int32), b_cpu.to(torch.int32),
-            )
 
-        # ----- next level bookkeeping (all tensor ops) -------------------
+        # ----- next level bookkeeping ------------------------------------
+        # everything that depends only on the SPLIT results happens here,
+        # while the partition kernel is still running; only the offsets
+        # math below needs its left counts
         l_st = left_stats[s_idx]                       # [n_split, C]
         r_st = totals[s_idx] - l_st
         totals = torch.stack([l_st, r_st], dim=1).reshape(-1, C)
@@ -873,6 +867,16 @@ def grow_forest(
                                    dtype=torch.long).repeat_interleave(2)
         node_tree = t_ids.repeat_interleave(2)
         node_nid = torch.stack([cid, cid + 1], dim=1).reshape(-1)
+
+        if pr_async is not None:
+            new_rows, new_offs, _ = ops.partition_rows_finish(*pr_async)
+        else:
+            split_feat = torch.where(do_split, feat_cpu,
+                                     torch.full_like(feat_cpu, -1))
+            new_rows, new_offs, _ = ops.partition_rows(
+                bins, row_idx, offsets,
+                split_feat.to(torch.int32), b_cpu.to(torch.int32),
+            )
         ls = new_offs[2 * s_idx]
         le = new_offs[2 * s_idx + 1]
         re = new_offs[2 * s_idx + 2]
