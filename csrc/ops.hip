@@ -269,14 +269,13 @@ __global__ void hist_build_kernel(
   const int nh = FG >> 4;
 
   if (vec) {
-    // 4 lanes share one row, each owning a 4-feature sub-chunk: within a
-    // 16-lane LDS conflict group the atomics then target 4 DIFFERENT
-    // features (bank offsets 8 apart with the B+1 stride) x 4 rows,
-    // instead of 16 random bins of ONE feature on 32 bank-pairs
+    // one row per lane, one uint4 per 16 features (as before), but each
+    // lane walks the four 4-feature chunks in a LANE-ROTATED order: at
+    // any instant a 16-lane LDS conflict group works 4 DIFFERENT
+    // features (bank offsets 8 apart with the B+1 stride) instead of 16
+    // random bins of ONE feature squeezed onto 32 bank-pairs
     const int sub = threadIdx.x & 3;
-    const int rl = threadIdx.x >> 2;
-    const int rpb = blockDim.x >> 2;
-    for (int i = rl; i < len; i += rpb) {
+    for (int i = threadIdx.x; i < len; i += blockDim.x) {
       const int r = identity_rows ? start + i : row_idx[start + i];
       const float* g = gh + (int64_t)r * CH + col0;
       unsigned long long addend[CELLS];
@@ -292,16 +291,22 @@ __global__ void hist_build_kernel(
         const int iv = __float2int_rn(g[DC + c] * scales[DC + c]);
         addend[c] |= (unsigned)iv;
       }
-      const uint8_t* base = bins + (int64_t)r * F + f0 + 4 * sub;
       for (int hh = 0; hh < nh; ++hh) {
-        const unsigned w = *reinterpret_cast<const unsigned*>(base + 16 * hh);
+        const uint4 bv = *reinterpret_cast<const uint4*>(
+            bins + (int64_t)r * F + f0 + 16 * hh);
+        const unsigned w[4] = {bv.x, bv.y, bv.z, bv.w};
 #pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          const int b = (w >> (8 * j)) & 0xff;
-          const int f = 16 * hh + 4 * sub + j;
-          unsigned long long* cell = lds64 + ((f * BP) + b) * CELLS;
+        for (int q = 0; q < 4; ++q) {
+          const int qq = (q + sub) & 3;
+          const unsigned wq = w[qq];
 #pragma unroll
-          for (int c = 0; c < CELLS; ++c) atomicAdd(cell + c, addend[c]);
+          for (int j = 0; j < 4; ++j) {
+            const int b = (wq >> (8 * j)) & 0xff;
+            const int f = 16 * hh + 4 * qq + j;
+            unsigned long long* cell = lds64 + ((f * BP) + b) * CELLS;
+#pragma unroll
+            for (int c = 0; c < CELLS; ++c) atomicAdd(cell + c, addend[c]);
+          }
         }
       }
     }
