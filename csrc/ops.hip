@@ -238,14 +238,7 @@ __global__ void hist_build_kernel(
     int F, int B, int FG, int CH, int identity_rows) {
   constexpr int C = DC + NC;
   constexpr int CELLS = DC > NC ? DC : NC;
-  // feature stride B+1: with a B stride every LDS bank index is
-  // 2*(f*B + b) mod 64 = 2b mod 64 (B is a multiple of 32), so the 256
-  // bins land on only 32 bank-pairs and lane groups — which all work
-  // the SAME feature in the old row-per-lane layout — conflict ~67% of
-  // LDS cycles (SQ_LDS_BANK_CONFLICT, profiles/r02_summary.md).  The
-  // pad makes the bank (2f + 2b) mod 64, and the vec path below gives
-  // every 16-lane conflict group FOUR different features.
-  extern __shared__ unsigned long long lds64[];  // FG * (B+1) * CELLS
+  extern __shared__ unsigned long long lds64[];  // FG * B * CELLS
   const float* scales = reinterpret_cast<const float*>(chunks);
   const int* chk = chunks + 2 * C + blockIdx.x * 5;
   const int fg = blockIdx.y;
@@ -257,8 +250,7 @@ __global__ void hist_build_kernel(
   const int single = chk[3];
   const int col0 = chk[4];
 
-  const int BP = B + 1;
-  const int lds_cells = FG * BP * CELLS;
+  const int lds_cells = FG * B * CELLS;
   for (int i = threadIdx.x; i < lds_cells; i += blockDim.x) lds64[i] = 0ull;
   __syncthreads();
 
@@ -268,69 +260,45 @@ __global__ void hist_build_kernel(
                    ((F % FG) == 0);
   const int nh = FG >> 4;
 
-  if (vec) {
-    // one row per lane, one uint4 per 16 features (as before), but each
-    // lane walks the four 4-feature chunks in a LANE-ROTATED order: at
-    // any instant a 16-lane LDS conflict group works 4 DIFFERENT
-    // features (bank offsets 8 apart with the B+1 stride) instead of 16
-    // random bins of ONE feature squeezed onto 32 bank-pairs
-    const int sub = threadIdx.x & 3;
-    for (int i = threadIdx.x; i < len; i += blockDim.x) {
-      const int r = identity_rows ? start + i : row_idx[start + i];
-      const float* g = gh + (int64_t)r * CH + col0;
-      unsigned long long addend[CELLS];
+  for (int i = threadIdx.x; i < len; i += blockDim.x) {
+    const int r = identity_rows ? start + i : row_idx[start + i];
+    const float* g = gh + (int64_t)r * CH + col0;
+    // quantize once per row, pack one u64 addend per cell
+    unsigned long long addend[CELLS];
 #pragma unroll
-      for (int c = 0; c < CELLS; ++c) addend[c] = 0ull;
+    for (int c = 0; c < CELLS; ++c) addend[c] = 0ull;
 #pragma unroll
-      for (int d = 0; d < DC; ++d) {
-        const int iv = __float2int_rn(g[d] * scales[d]);
-        addend[d] |= ((unsigned long long)(unsigned)iv) << 32;
-      }
+    for (int d = 0; d < DC; ++d) {
+      const int iv = __float2int_rn(g[d] * scales[d]);
+      addend[d] |= ((unsigned long long)(unsigned)iv) << 32;
+    }
 #pragma unroll
-      for (int c = 0; c < NC; ++c) {
-        const int iv = __float2int_rn(g[DC + c] * scales[DC + c]);
-        addend[c] |= (unsigned)iv;
-      }
+    for (int c = 0; c < NC; ++c) {
+      const int iv = __float2int_rn(g[DC + c] * scales[DC + c]);
+      addend[c] |= (unsigned)iv;
+    }
+    if (vec) {
       for (int hh = 0; hh < nh; ++hh) {
         const uint4 bv = *reinterpret_cast<const uint4*>(
             bins + (int64_t)r * F + f0 + 16 * hh);
         const unsigned w[4] = {bv.x, bv.y, bv.z, bv.w};
 #pragma unroll
         for (int q = 0; q < 4; ++q) {
-          const int qq = (q + sub) & 3;
-          const unsigned wq = w[qq];
 #pragma unroll
           for (int j = 0; j < 4; ++j) {
-            const int b = (wq >> (8 * j)) & 0xff;
-            const int f = 16 * hh + 4 * qq + j;
-            unsigned long long* cell = lds64 + ((f * BP) + b) * CELLS;
+            const int b = (w[q] >> (8 * j)) & 0xff;
+            unsigned long long* cell =
+                lds64 + (((16 * hh + q * 4 + j) * B) + b) * CELLS;
 #pragma unroll
             for (int c = 0; c < CELLS; ++c) atomicAdd(cell + c, addend[c]);
           }
         }
       }
-    }
-  } else {
-    for (int i = threadIdx.x; i < len; i += blockDim.x) {
-      const int r = identity_rows ? start + i : row_idx[start + i];
-      const float* g = gh + (int64_t)r * CH + col0;
-      unsigned long long addend[CELLS];
-#pragma unroll
-      for (int c = 0; c < CELLS; ++c) addend[c] = 0ull;
-#pragma unroll
-      for (int d = 0; d < DC; ++d) {
-        const int iv = __float2int_rn(g[d] * scales[d]);
-        addend[d] |= ((unsigned long long)(unsigned)iv) << 32;
-      }
-#pragma unroll
-      for (int c = 0; c < NC; ++c) {
-        const int iv = __float2int_rn(g[DC + c] * scales[DC + c]);
-        addend[c] |= (unsigned)iv;
-      }
+    } else {
       const uint8_t* br = bins + (int64_t)r * F + f0;
       for (int f = 0; f < nf; ++f) {
         const int b = br[f];
-        unsigned long long* cell = lds64 + ((f * BP) + b) * CELLS;
+        unsigned long long* cell = lds64 + ((f * B) + b) * CELLS;
 #pragma unroll
         for (int c = 0; c < CELLS; ++c) atomicAdd(cell + c, addend[c]);
       }
@@ -347,7 +315,7 @@ __global__ void hist_build_kernel(
     float* dst = out + (((int64_t)node * F + f0) * B) * C;
     for (int i = threadIdx.x; i < nf * B; i += blockDim.x) {
       const int f = i / B, b = i - f * B;
-      const unsigned long long* cell = lds64 + ((f * BP) + b) * CELLS;
+      const unsigned long long* cell = lds64 + ((f * B) + b) * CELLS;
       float* o = dst + ((int64_t)f * B + b) * C;
 #pragma unroll
       for (int d = 0; d < DC; ++d)
@@ -361,7 +329,7 @@ __global__ void hist_build_kernel(
     long long* sdst = stage + (((int64_t)node * F + f0) * B) * C;
     for (int i = threadIdx.x; i < nf * B; i += blockDim.x) {
       const int f = i / B, b = i - f * B;
-      const unsigned long long* cell = lds64 + ((f * BP) + b) * CELLS;
+      const unsigned long long* cell = lds64 + ((f * B) + b) * CELLS;
       long long* o = sdst + ((int64_t)f * B + b) * C;
 #pragma unroll
       for (int d = 0; d < DC; ++d) {
@@ -439,7 +407,7 @@ void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
   // FG=64 / 1024 threads / 128 KiB; CELLS==2 -> FG=32 / 1024 / 128 KiB;
   // CELLS>=3 (wide multiclass) keeps the conservative 64-KiB config
   const int budget = CELLS <= 2 ? lds_budget : std::min(lds_budget, 65536);
-  int FG = std::max<int>(1, std::min<int>(F, budget / ((B + 1) * CELLS * 8)));
+  int FG = std::max<int>(1, std::min<int>(F, budget / (B * CELLS * 8)));
   if (CELLS == 1 && FG >= 64 && (F % 64) == 0) FG = 64;
   else if (CELLS <= 2 && FG >= 32 && (F % 32) == 0) FG = 32;
   else if (FG >= 16) FG = 16;
@@ -456,7 +424,7 @@ void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
   auto offs = node_offsets.accessor<int64_t, 1>();
   int64_t total_rows = 0;
   for (int nd = 0; nd < n_nodes; ++nd) total_rows += offs[nd + 1] - offs[nd];
-  const size_t lds_bytes_pre = (size_t)FG * (B + 1) * CELLS * 8;
+  const size_t lds_bytes_pre = (size_t)FG * B * CELLS * 8;
   const int blocks_per_cu =
       std::max<int>(1, (int)(163840 / std::max<size_t>(1, lds_bytes_pre)));
   const int64_t resident = (int64_t)256 * blocks_per_cu;
