@@ -549,11 +549,12 @@ __global__ void partition_kernel(
     int* __restrict__ new_rows,        // [M]
     int* __restrict__ cursors,         // [n_nodes, 2] = {lcur, rcur}
     const uint8_t* __restrict__ bins,  // [N, F]
+    const uint8_t* __restrict__ bins_t,  // [F, N] or null
     const int* __restrict__ row_idx,   // [M]
     const int* __restrict__ chunks,    // [n_chunks, 3]
     const int* __restrict__ feat,      // [n_nodes]
     const int* __restrict__ thr,       // [n_nodes]
-    int F) {
+    int F, long long Nrows) {
   __shared__ int base_l, base_r, loc_l, loc_r;
   __shared__ int wl[8];
   extern __shared__ unsigned long long bits[];  // ceil(chunk_rows/64)
@@ -573,7 +574,13 @@ __global__ void partition_kernel(
     bool left = false;
     if (active) {
       const int r = row_idx[start + i];
-      left = (f < 0) || (bins[(int64_t)r * F + f] <= t);
+      // the transposed matrix turns the per-row gather (1 useful byte
+      // per 64-B line with the row-major layout) into a locally-dense
+      // read: a node's rows at level L are ~2^-L dense in [0, N), so a
+      // line yields ~64/2^L useful bytes at shallow levels
+      const uint8_t bv = bins_t ? bins_t[(int64_t)f * Nrows + r]
+                                : bins[(int64_t)r * F + f];
+      left = (f < 0) || (bv <= t);
     }
     const unsigned long long m = __ballot(active && left);
     if (lane == 0) {
@@ -622,7 +629,8 @@ __global__ void partition_kernel(
 }
 
 void partition_rows(torch::Tensor new_rows, torch::Tensor left_counts,
-                    torch::Tensor bins, torch::Tensor row_idx,
+                    torch::Tensor bins, torch::Tensor bins_t,
+                    torch::Tensor row_idx,
                     torch::Tensor node_offsets, torch::Tensor feat,
                     torch::Tensor thr) {
   auto tpA = std::chrono::steady_clock::now();
@@ -660,12 +668,16 @@ void partition_rows(torch::Tensor new_rows, torch::Tensor left_counts,
     auto tp2 = std::chrono::steady_clock::now();
     const int n_chunks = (int)(chunk_v.size() / 3);
     const size_t bit_lds = (size_t)((chunk_rows + 63) / 64) * 8;
+    const uint8_t* bt_ptr =
+        bins_t.defined() && bins_t.numel() ? bins_t.data_ptr<uint8_t>()
+                                           : nullptr;
     hipLaunchKernelGGL(partition_kernel, dim3(n_chunks), dim3(256), bit_lds,
                        stream,
                        new_rows.data_ptr<int>(), cursors.data_ptr<int>(),
-                       bins.data_ptr<uint8_t>(), row_idx.data_ptr<int>(),
+                       bins.data_ptr<uint8_t>(), bt_ptr,
+                       row_idx.data_ptr<int>(),
                        chunks.data_ptr<int>(), feat.data_ptr<int>(),
-                       thr.data_ptr<int>(), F);
+                       thr.data_ptr<int>(), F, (long long)bins.size(0));
     if (part_dbg) {
       auto tp3 = std::chrono::steady_clock::now();
       auto us = [](auto a, auto b) {
