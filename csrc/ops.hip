@@ -574,13 +574,17 @@ __global__ void partition_kernel(
     bool left = false;
     if (active) {
       const int r = row_idx[start + i];
-      // the transposed matrix turns the per-row gather (1 useful byte
-      // per 64-B line with the row-major layout) into a locally-dense
-      // read: a node's rows at level L are ~2^-L dense in [0, N), so a
-      // line yields ~64/2^L useful bytes at shallow levels
-      const uint8_t bv = bins_t ? bins_t[(int64_t)f * Nrows + r]
-                                : bins[(int64_t)r * F + f];
-      left = (f < 0) || (bv <= t);
+      if (f < 0) {
+        left = true;  // unsplit node: every row goes left (no load!)
+      } else {
+        // the transposed matrix turns the per-row gather (1 useful byte
+        // per 64-B line with the row-major layout) into a locally-dense
+        // read: a node's rows at level L are ~2^-L dense in [0, N), so
+        // a line yields ~64/2^L useful bytes at shallow levels
+        const uint8_t bv = bins_t ? bins_t[(int64_t)f * Nrows + r]
+                                  : bins[(int64_t)r * F + f];
+        left = bv <= t;
+      }
     }
     const unsigned long long m = __ballot(active && left);
     if (lane == 0) {
