@@ -246,15 +246,17 @@ def _bins_transposed(bins):
     m = _load_hip()
     if m is None or not bins.is_cuda:
         return None
-    for i, (wr, bt) in enumerate(_BT_CACHE):
+    # prune entries whose bins tensor died FIRST: a stale entry pins a
+    # multi-GB transpose (25.6 GB at the 100M x 256 capacity point)
+    _BT_CACHE[:] = [(w, b) for (w, b) in _BT_CACHE if w() is not None]
+    for wr, bt in _BT_CACHE:
         if wr() is bins:
             return bt
-    _BT_CACHE[:] = [(w, b) for (w, b) in _BT_CACHE if w() is not None]
     n, f = bins.shape
     bt = torch.empty(f, n, dtype=torch.uint8, device=bins.device)
     m.transpose_u8(bt, bins)
     _BT_CACHE.insert(0, (weakref.ref(bins), bt))
-    del _BT_CACHE[2:]
+    del _BT_CACHE[1:]
     return bt
 
 
