@@ -139,3 +139,30 @@ def test_forest_predict_weighted_sum():
     x = torch.zeros(3, 1)
     out = ops.forest_predict(x, [t1, t2], torch.tensor([1.0, 0.5]))
     assert out.squeeze(1).tolist() == [4.5, 4.5, 4.5]
+
+
+def test_gather_ranges_cpu_fallback():
+    import torch as t
+    from spark_ensemble_amd.ops import dispatch
+
+    src = t.arange(100, dtype=t.int32)
+    starts = t.tensor([5, 40, 90], dtype=t.int64)
+    lens = t.tensor([3, 0, 10], dtype=t.int64)
+    out = dispatch.gather_ranges(src, starts, lens)
+    assert out.tolist() == list(range(5, 8)) + list(range(90, 100))
+
+
+def test_leaf_scatter_cpu_fallback():
+    import torch as t
+    from spark_ensemble_amd.ops import dispatch
+
+    tp = t.zeros(10, 2)
+    row_idx = t.tensor([3, 4, 5, 9, 0], dtype=t.int32)
+    dispatch.leaf_scatter(
+        tp, row_idx,
+        t.tensor([0, 3], dtype=t.int64), t.tensor([3, 2], dtype=t.int64),
+        t.tensor([0, 1], dtype=t.int64), t.tensor([2.5, -1.0]),
+    )
+    assert tp[3, 0] == 2.5 and tp[4, 0] == 2.5 and tp[5, 0] == 2.5
+    assert tp[9, 1] == -1.0 and tp[0, 1] == -1.0
+    assert tp.abs().sum() == 2.5 * 3 + 2.0
