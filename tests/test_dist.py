@@ -77,6 +77,27 @@ from spark_ensemble_amd.tuning import MulticlassClassificationEvaluator
 ev = MulticlassClassificationEvaluator("accuracy")
 out["eval_acc"] = ev.evaluate(m.transform(df))
 
+# --- BIT-EXACT sharded-vs-single parity ---------------------------------
+# categorical (identity-edge) features kill the only nondeterminism
+# source (sample-dependent quantile edges), and integer-valued labels
+# with unit weights make every histogram cell an exact small-integer f32
+# sum — order-independent, so the 2-rank all-reduced tree must equal the
+# single-process tree BITWISE.
+from spark_ensemble_amd.frame import TensorFrame
+from spark_ensemble_amd.models import DecisionTreeRegressor
+
+ge = torch.Generator().manual_seed(17)
+xi_all = torch.randint(0, 32, (40000, 8), generator=ge).float()
+yi_all = (xi_all[:, 0] > 15).float() * 4 + (xi_all[:, 1] % 3).float()
+xi = xi_all[rank::world]
+yi = yi_all[rank::world]
+dfx = TensorFrame(features=xi, label=yi)
+dfx.set_categorical({f: 32 for f in range(8)})
+mex = DecisionTreeRegressor().setMaxDepth(6).setMaxBins(32).fit(dfx)
+out["exact_tree"] = {
+    k: v.tolist() for k, v in mex._tree.items() if k != "feature_importance"
+}
+
 if rank == 0:
     with open(os.environ["SEA_OUT"], "w") as f:
         json.dump(out, f)
@@ -163,4 +184,25 @@ def test_dist_stacking_quality(world2_results):
 
 def test_dist_logreg_and_evaluator(world2_results):
     assert world2_results["logreg_acc"] > 0.5
+
+
+def test_sharded_tree_bit_exact_with_fixed_edges(world2_results):
+    """With categorical identity edges (no sample-dependent quantiles)
+    and exact-integer histogram sums, the 2-rank all-reduced tree equals
+    the single-process tree BITWISE (pin the
+    collectives with a fixed-edge variant)."""
+    from spark_ensemble_amd.frame import TensorFrame
+    from spark_ensemble_amd.models import DecisionTreeRegressor
+
+    ge = torch.Generator().manual_seed(17)
+    xi = torch.randint(0, 32, (40000, 8), generator=ge).float()
+    yi = (xi[:, 0] > 15).float() * 4 + (xi[:, 1] % 3).float()
+    dfx = TensorFrame(features=xi, label=yi)
+    dfx.set_categorical({f: 32 for f in range(8)})
+    m = DecisionTreeRegressor().setMaxDepth(6).setMaxBins(32).fit(dfx)
+    got = world2_results["exact_tree"]
+    for k in ("feature", "threshold", "left_child", "leaf_value"):
+        single = m._tree[k]
+        dist2 = torch.tensor(got[k], dtype=single.dtype).reshape(single.shape)
+        assert torch.equal(single, dist2), k
     assert 0.0 <= world2_results["eval_acc"] <= 1.0
