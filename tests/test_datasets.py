@@ -317,3 +317,51 @@ def test_letter_gbm_multiclass_learns(letter):
     )
     acc = _acc(gbm, te)
     assert acc > 0.3, acc  # prior is ~1/26 = 0.038
+
+
+# ---------------------------------------------------------------------------
+# the same quality gates on the HIP path (real data on GPU)
+# ---------------------------------------------------------------------------
+
+
+def _to_cuda(pair):
+    tr, te = pair
+    return (
+        TensorFrame({c: tr[c].cuda() for c in tr.columns}),
+        TensorFrame({c: te[c].cuda() for c in te.columns}),
+    )
+
+
+@pytest.mark.gpu
+def test_adult_gbm_quality_on_gpu(adult):
+    """GBM on real adult data trained entirely on the HIP kernels must
+    reach the same quality band the CPU suite asserts (fixed-point
+    histogram quantization must not cost accuracy)."""
+    tr, te = _to_cuda(adult)
+    m = (sea.GBMClassifier().setLoss("bernoulli").setNumBaseLearners(30)
+         .setBaseLearner(DecisionTreeRegressor().setMaxDepth(5))
+         .fit(tr))
+    base = DecisionTreeClassifier().setMaxDepth(5).fit(tr)
+    assert _acc(m, te) > _acc(base, te)
+    assert _acc(m, te) > 0.82
+
+
+@pytest.mark.gpu
+def test_letter_multiclass_quality_on_gpu(letter):
+    """26-class letter on the fused multiclass round (wide-gini + K-tree
+    forest kernels) — the hardest kernel configuration on real data."""
+    tr, te = _to_cuda(letter)
+    m = (sea.GBMClassifier().setLoss("logloss").setNumBaseLearners(12)
+         .setBaseLearner(DecisionTreeRegressor().setMaxDepth(6))
+         .fit(tr))
+    assert _acc(m, te) > 0.70
+
+
+@pytest.mark.gpu
+def test_cpusmall_bagging_quality_on_gpu(cpusmall):
+    tr, te = _to_cuda(cpusmall)
+    bag = (sea.BaggingRegressor().setNumBaseLearners(20)
+           .setBaseLearner(DecisionTreeRegressor().setMaxDepth(7))
+           .setSubspaceRatio(0.8).setSeed(3).fit(tr))
+    base = DecisionTreeRegressor().setMaxDepth(7).fit(tr)
+    assert _mse(bag, te) < _mse(base, te)
