@@ -106,3 +106,41 @@ def test_logloss_gradient_forall(seed, k):
         dp[:, d] = eps
         num = ((loss.loss(y, p + dp) - loss.loss(y, p - dp)) / (2 * eps)).reshape(-1)
         assert torch.allclose(num, ana[:, d].reshape(-1), rtol=1e-4, atol=1e-5)
+
+
+@settings(max_examples=15, deadline=None)
+@given(
+    n=st.integers(200, 4000),
+    f=st.integers(2, 12),
+    b=st.sampled_from([8, 16, 32]),
+    depth=st.integers(1, 6),
+    weighted=st.booleans(),
+    masked=st.booleans(),
+    seed=st.integers(0, 2**31 - 1),
+)
+def test_grow_tree_delegation_forall(n, f, b, depth, weighted, masked, seed):
+    """The arena grower (grow_forest T=1) and the loop grower agree on
+    ARBITRARY shapes, depths, weightings and row masks — the fuzzed
+    version of the fixed-shape delegation parity tests."""
+    from spark_ensemble_amd.models.tree_grower import (
+        GrowParams, _grow_tree_seq, grow_tree,
+    )
+    from spark_ensemble_amd.ops import reference
+
+    g = torch.Generator().manual_seed(seed)
+    x = torch.randn(n, f, generator=g)
+    edges = reference.quantile_bins(x, b)
+    bins = reference.bin_features(x, edges)
+    grad = torch.randn(n, 1, generator=g)
+    hess = (torch.rand(n, generator=g) + 0.5 if weighted
+            else torch.ones(n))
+    mask = (torch.rand(n, generator=g) > 0.25) if masked else None
+    if mask is not None and int(mask.sum()) == 0:
+        return
+    params = GrowParams(max_depth=depth, max_bins=b)
+    a = grow_tree(bins, edges, grad, hess, params, row_mask=mask)
+    c = _grow_tree_seq(bins, edges, grad, hess, params, row_mask=mask)
+    for k in ("feature", "threshold", "left_child"):
+        assert torch.equal(a[k].float(), c[k].float()), k
+    assert torch.allclose(a["leaf_value"], c["leaf_value"],
+                          rtol=1e-4, atol=2e-5)
