@@ -507,7 +507,10 @@ def _finish_level_split(
         if root:
             hists = built
         else:
-            hists = torch.zeros(n_active, F, B, C, dtype=torch.float32,
+            # every active row is either built (scattered copy) or
+            # non-built (parent - sibling) — assemble_chunk overwrites
+            # ALL of it, so no zero fill
+            hists = torch.empty(n_active, F, B, C, dtype=torch.float32,
                                 device=device)
             if built is not None and built.numel():
                 assemble_chunk(hists, built, 0, F)
