@@ -471,7 +471,10 @@ void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
       chunk_v.push_back(col0_p ? col0_p[nd] : 0);
     }
   }
-  if ((int)chunk_v.size() == 2 * C) return;
+  if ((int)chunk_v.size() == 2 * C) {
+    out.zero_();  // dispatch allocates `out` uninitialized
+    return;
+  }
   const int n_chunks = (int)((chunk_v.size() - 2 * C) / 5);
   auto chunks_b = h2d_async(chunk_v.data(), chunk_v.size() * 4, 0,
                             bins.device());

@@ -136,7 +136,9 @@ def hist_build(bins, gh, row_idx, node_offsets, num_bins, d_dims=-1, max_abs=Non
                     if ci == len(spans) - 1:
                         parts.append(part[..., w:])  # tail once
                 return torch.cat(parts, dim=-1)
-            out = torch.zeros(
+            # every (node, f, b) cell is fully written by either the
+            # single-chunk flush or the multi-chunk decode — no zero fill
+            out = torch.empty(
                 n_nodes, F, num_bins, C, dtype=torch.float32, device=bins.device
             )
             m.hist_build(
@@ -175,7 +177,7 @@ def hist_build_forest(bins, gh, row_idx, node_offsets, node_col0, num_bins,
         if m is not None:
             n_nodes = node_offsets.numel() - 1
             F = bins.shape[1]
-            out = torch.zeros(
+            out = torch.empty(
                 n_nodes, F, num_bins, c_per_node,
                 dtype=torch.float32, device=bins.device,
             )
