@@ -406,7 +406,11 @@ void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
   // probe-measured optima (profiles/r01_hist_probe3): CELLS==1 ->
   // FG=64 / 1024 threads / 128 KiB; CELLS==2 -> FG=32 / 1024 / 128 KiB;
   // CELLS>=3 (wide multiclass) keeps the conservative 64-KiB config
-  const int budget = CELLS <= 2 ? lds_budget : std::min(lds_budget, 65536);
+  // CELLS>=3 (multiclass): FG=16 fits in ~99 KiB with the B+1-free
+  // layout and keeps the vectorized uint4 row loads; the old 64-KiB cap
+  // forced FG=8 byte loads and DOUBLED the per-row gh re-reads (one per
+  // feature group)
+  const int budget = lds_budget;
   int FG = std::max<int>(1, std::min<int>(F, budget / (B * CELLS * 8)));
   if (CELLS == 1 && FG >= 64 && (F % 64) == 0) FG = 64;
   else if (CELLS <= 2 && FG >= 32 && (F % 32) == 0) FG = 32;
@@ -416,7 +420,8 @@ void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
   if (fg_env == 16 || fg_env == 32 || fg_env == 64)
     FG = std::min(FG, fg_env);
   const int threads =
-      (FG >= 64 || (CELLS == 2 && FG >= 32)) ? 1024 : (FG >= 32 ? 512 : 256);
+      (FG >= 64 || (CELLS == 2 && FG >= 32)) ? 1024
+      : ((FG >= 32 || (CELLS >= 3 && FG >= 16)) ? 512 : 256);
   const int n_groups = (int)ceil_div(F, FG);
 
   // ---- adaptive chunking: target ~resident-grid x OVERSUB blocks --------
