@@ -468,6 +468,13 @@ void hist_build(torch::Tensor out, torch::Tensor bins, torch::Tensor gh,
     int64_t s = offs[nd], e = offs[nd + 1];
     const int single = (e - s) <= chunk_rows ? 1 : 0;
     if (!single) multi_nodes.push_back(nd);
+    if (e == s) {
+      // `out` is allocated uninitialized (flush/decode fully overwrite
+      // every chunked node) — a ZERO-ROW node emits no chunk, so its
+      // slice must be zeroed here (rank-local empty nodes are real in
+      // sharded fits: this node's rows may all live on other ranks)
+      out.slice(0, nd, nd + 1).zero_();
+    }
     for (int64_t c = s; c < e; c += chunk_rows) {
       chunk_v.push_back(nd);
       chunk_v.push_back((int)c);
