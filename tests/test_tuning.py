@@ -139,3 +139,44 @@ def test_cross_validator_model_save_load(tmp_path):
     a = m.transform(df)["prediction"]
     b = m2.transform(df)["prediction"]
     assert torch.allclose(a, b, rtol=1e-6, atol=1e-7)
+
+
+def test_evaluators_match_sklearn_exactly():
+    """Metric formulas pinned against scikit-learn (the reference pins
+    its evaluators to Spark MLlib's — SURVEY §4.1 cross-library genre)."""
+    import numpy as np
+    from sklearn.metrics import (
+        accuracy_score, f1_score, mean_absolute_error, mean_squared_error,
+        precision_score, r2_score, recall_score,
+    )
+
+    from spark_ensemble_amd.frame import TensorFrame
+    from spark_ensemble_amd.tuning import (
+        MulticlassClassificationEvaluator, RegressionEvaluator,
+    )
+
+    g = torch.Generator().manual_seed(11)
+    y = torch.randint(0, 4, (2000,), generator=g).float()
+    p = torch.where(torch.rand(2000, generator=g) < 0.7, y,
+                    torch.randint(0, 4, (2000,), generator=g).float())
+    df = TensorFrame(label=y, prediction=p)
+    yn, pn = y.numpy(), p.numpy()
+    assert abs(MulticlassClassificationEvaluator("accuracy").evaluate(df)
+               - accuracy_score(yn, pn)) < 1e-6
+    assert abs(MulticlassClassificationEvaluator("f1").evaluate(df)
+               - f1_score(yn, pn, average="weighted")) < 1e-5
+    assert abs(MulticlassClassificationEvaluator("weightedPrecision").evaluate(df)
+               - precision_score(yn, pn, average="weighted")) < 1e-5
+    assert abs(MulticlassClassificationEvaluator("weightedRecall").evaluate(df)
+               - recall_score(yn, pn, average="weighted")) < 1e-5
+
+    yr = torch.randn(2000, generator=g)
+    pr = yr + 0.3 * torch.randn(2000, generator=g)
+    dfr = TensorFrame(label=yr, prediction=pr)
+    yrn, prn = yr.numpy(), pr.numpy()
+    assert abs(RegressionEvaluator("rmse").evaluate(dfr)
+               - mean_squared_error(yrn, prn) ** 0.5) < 1e-5
+    assert abs(RegressionEvaluator("mae").evaluate(dfr)
+               - mean_absolute_error(yrn, prn)) < 1e-5
+    assert abs(RegressionEvaluator("r2").evaluate(dfr)
+               - r2_score(yrn, prn)) < 1e-5
