@@ -64,7 +64,7 @@ dfr = synthetic_regression(3000, 16, seed=0)
 
 @pytest.mark.parametrize("rel", [f for f in FILES
                                  if os.path.exists(os.path.join(ROOT, f))])
-def test_doc_snippets_run(rel, tmp_path):
+def test_doc_snippets_run(rel, tmp_path, monkeypatch):
     text = open(os.path.join(ROOT, rel)).read()
     blocks = re.findall(r"```python\n(.*?)```", text, re.S)
     assert blocks, f"{rel} has no python blocks"
@@ -72,7 +72,7 @@ def test_doc_snippets_run(rel, tmp_path):
     # the per-algorithm guides are fragments continuing an ambient
     # session (the reference's mdoc does the same with a shared prelude)
     exec(compile(PRELUDE, "prelude", "exec"), env)
-    os.chdir(tmp_path)
+    monkeypatch.chdir(tmp_path)  # snippets may save to relative paths
     ran = 0
     for i, code in enumerate(blocks):
         if any(s in code for s in SKIP_MARKERS):
