@@ -43,11 +43,22 @@ static inline int64_t ceil_div(int64_t a, int64_t b) { return (a + b - 1) / b; }
 static torch::Tensor h2d_async(const void* src, size_t bytes, int slot,
                                const torch::Device& dev) {
   static thread_local torch::Tensor pin[8];
+  static thread_local hipEvent_t done[8] = {};
   auto& b = pin[slot];
-  if (!b.defined() || (size_t)b.numel() < bytes)
-    b = torch::empty({(int64_t)std::max<size_t>(bytes, 4096)},
+  // the PREVIOUS copy from this slot may still be pending on the
+  // stream: overwriting (or freeing, when the slot grows) the pinned
+  // staging before the DMA reads it feeds the kernels garbage
+  // descriptors — the depth-12 / 100M-row GPU memory faults.  The
+  // host-run-ahead pipelining makes this race real; wait on the slot's
+  // completion event before touching the buffer (normally long done).
+  if (done[slot]) (void)hipEventSynchronize(done[slot]);
+  if (!b.defined() || (size_t)b.numel() < bytes) {
+    size_t cap = 4096;
+    while (cap < bytes) cap *= 2;
+    b = torch::empty({(int64_t)cap},
                      torch::TensorOptions().dtype(torch::kByte)
                          .pinned_memory(true));
+  }
   std::memcpy(b.data_ptr(), src, bytes);
   auto d = torch::empty({(int64_t)bytes},
                         torch::TensorOptions().dtype(torch::kByte).device(dev));
@@ -56,6 +67,9 @@ static torch::Tensor h2d_async(const void* src, size_t bytes, int slot,
   auto stream = at::hip::getCurrentHIPStream();
   (void)hipMemcpyAsync(d.data_ptr(), b.data_ptr(), bytes,
                        hipMemcpyHostToDevice, stream);
+  if (!done[slot])
+    (void)hipEventCreateWithFlags(&done[slot], hipEventDisableTiming);
+  (void)hipEventRecord(done[slot], stream);
   return d;
 }
 
