@@ -42,16 +42,21 @@ static inline int64_t ceil_div(int64_t a, int64_t b) { return (a + b - 1) / b; }
 // fetch) before the slot is touched again.
 static torch::Tensor h2d_async(const void* src, size_t bytes, int slot,
                                const torch::Device& dev) {
-  static thread_local torch::Tensor pin[8];
-  static thread_local hipEvent_t done[8] = {};
-  auto& b = pin[slot];
+  static thread_local torch::Tensor pin[16];
+  static thread_local hipEvent_t done[16] = {};
+  static thread_local unsigned char flip[8] = {};
   // the PREVIOUS copy from this slot may still be pending on the
   // stream: overwriting (or freeing, when the slot grows) the pinned
   // staging before the DMA reads it feeds the kernels garbage
-  // descriptors — the depth-12 / 100M-row GPU memory faults.  The
-  // host-run-ahead pipelining makes this race real; wait on the slot's
-  // completion event before touching the buffer (normally long done).
-  if (done[slot]) (void)hipEventSynchronize(done[slot]);
+  // descriptors — the depth-12 / 100M-row GPU memory faults.  Each
+  // slot PING-PONGS between two buffers so the completion wait below
+  // lands on the copy from TWO calls ago — with the host running at
+  // most one level ahead that copy has long executed, preserving the
+  // run-ahead pipelining the level loop depends on.
+  flip[slot] ^= 1;
+  const int bi = slot * 2 + flip[slot];
+  auto& b = pin[bi];
+  if (done[bi]) (void)hipEventSynchronize(done[bi]);
   if (!b.defined() || (size_t)b.numel() < bytes) {
     size_t cap = 4096;
     while (cap < bytes) cap *= 2;
@@ -67,9 +72,9 @@ static torch::Tensor h2d_async(const void* src, size_t bytes, int slot,
   auto stream = at::hip::getCurrentHIPStream();
   (void)hipMemcpyAsync(d.data_ptr(), b.data_ptr(), bytes,
                        hipMemcpyHostToDevice, stream);
-  if (!done[slot])
-    (void)hipEventCreateWithFlags(&done[slot], hipEventDisableTiming);
-  (void)hipEventRecord(done[slot], stream);
+  if (!done[bi])
+    (void)hipEventCreateWithFlags(&done[bi], hipEventDisableTiming);
+  (void)hipEventRecord(done[bi], stream);
   return d;
 }
 
