@@ -766,3 +766,23 @@ def test_partition_transposed_matches_rowmajor():
     assert got_left == left0
     # unsplit node (-1): everything goes left, order preserved
     assert int(lc[1]) == n - n // 3
+
+
+@pytest.mark.gpu
+def test_deep_tree_gbm_large():
+    """Depth-12 GBM at 5M rows: per-level descriptor tables outgrow the
+    pinned-staging minimum, exercising the slot-growth / reuse path that
+    raced before the ping-pong fix (GPU memory fault postmortem in
+    profiles/r02_summary.md)."""
+    import spark_ensemble_amd as sea
+    from spark_ensemble_amd.models import DecisionTreeRegressor
+    from spark_ensemble_amd.utils.io import synthetic_regression
+
+    df = synthetic_regression(5_000_000, 64, seed=21, device="cuda:0")
+    m = (sea.GBMRegressor().setNumBaseLearners(2)
+         .setBaseLearner(DecisionTreeRegressor().setMaxDepth(12)
+                         .setMaxBins(256))
+         .fit(df))
+    p = m.predict(df["features"])
+    assert torch.isfinite(p).all()
+    assert float(((p - df["label"]) ** 2).mean()) < float(df["label"].var())
